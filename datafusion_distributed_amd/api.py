@@ -1,0 +1,311 @@
+"""api.py — ctypes binding to libdd_shuffle.so (the C ABI, include/dd_shuffle.h).
+
+Harness layer only: uploads/downloads numpy-described Arrow columns (the column-dict
+convention of oracle/pyref.py) to device memory and drives the C ABI. The compute path is
+entirely inside libdd_shuffle.so; there is NO CPU fallback here — calls on a GPU-less box
+raise DDError(DD_ERR_NO_DEVICE), by design.
+"""
+
+import ctypes
+import os
+
+import numpy as np
+
+_DIR = os.path.dirname(os.path.abspath(__file__))
+_SO = os.path.join(_DIR, "libdd_shuffle.so")
+
+DD_OK = 0
+STATUS_NAMES = {
+    0: "DD_OK", 1: "DD_ERR_INVALID", 2: "DD_ERR_NO_DEVICE", 3: "DD_ERR_HIP",
+    4: "DD_ERR_RCCL", 5: "DD_ERR_NOT_FOUND", 6: "DD_ERR_UNSUPPORTED",
+}
+
+DTYPE_CODE = {"u8": 1, "i16": 2, "i32": 3, "i64": 4, "f32": 5, "f64": 6,
+              "bool": 7, "utf8": 8, "dict32": 9}
+FIXED_NP = {"u8": np.uint8, "bool": np.uint8, "i16": np.int16, "i32": np.int32,
+            "f32": np.float32, "i64": np.int64, "f64": np.float64, "dict32": np.int32}
+ELEM_SIZE = {"u8": 1, "bool": 1, "i16": 2, "i32": 4, "f32": 4, "i64": 8, "f64": 8,
+             "dict32": 4}
+
+DD_MAX_COLS = 24
+UNIQUE_ID_BYTES = 128
+
+
+class DDError(RuntimeError):
+    def __init__(self, status, msg):
+        super().__init__(f"{STATUS_NAMES.get(status, status)}: {msg}")
+        self.status = status
+
+
+class ColDesc(ctypes.Structure):
+    _fields_ = [
+        ("dtype", ctypes.c_int32),
+        ("data", ctypes.c_void_p),
+        ("validity", ctypes.c_void_p),
+        ("offsets", ctypes.c_void_p),
+        ("data_len", ctypes.c_int64),
+        ("dict_bytes", ctypes.c_void_p),
+        ("dict_offsets", ctypes.c_void_p),
+        ("dict_n", ctypes.c_int64),
+    ]
+
+
+class BatchDesc(ctypes.Structure):
+    _fields_ = [
+        ("n_rows", ctypes.c_int64),
+        ("n_cols", ctypes.c_int32),
+        ("cols", ColDesc * DD_MAX_COLS),
+    ]
+
+
+class TaskKeyC(ctypes.Structure):
+    _fields_ = [
+        ("query_id_hi", ctypes.c_uint64),
+        ("query_id_lo", ctypes.c_uint64),
+        ("stage_id", ctypes.c_uint64),
+        ("task_number", ctypes.c_uint64),
+    ]
+
+
+_lib = None
+
+
+def lib():
+    global _lib
+    if _lib is None:
+        if not os.path.exists(_SO):
+            raise FileNotFoundError(
+                f"{_SO} missing — build it with __graft_entry__.build(); the product path "
+                "has no fallback")
+        L = ctypes.CDLL(_SO)
+        L.dd_last_error.restype = ctypes.c_char_p
+        L.dd_version.restype = ctypes.c_char_p
+        L.dd_device_count.restype = ctypes.c_int
+        L.dd_partitioner_pids.restype = ctypes.c_void_p
+        L.dd_partitioner_col_data.restype = ctypes.c_void_p
+        L.dd_partitioner_col_validity.restype = ctypes.c_void_p
+        L.dd_partitioner_col_lengths.restype = ctypes.c_void_p
+        L.dd_exchanged_col_data.restype = ctypes.c_void_p
+        L.dd_exchanged_col_validity.restype = ctypes.c_void_p
+        L.dd_exchanged_col_lengths.restype = ctypes.c_void_p
+        L.dd_exchanged_total_rows.restype = ctypes.c_int64
+        _lib = L
+    return _lib
+
+
+def _check(st):
+    if st != DD_OK:
+        raise DDError(st, lib().dd_last_error().decode())
+
+
+def device_count():
+    return lib().dd_device_count()
+
+
+def _dev_alloc(nbytes):
+    p = ctypes.c_void_p()
+    _check(lib().dd_dev_alloc(ctypes.c_int64(max(nbytes, 1)), ctypes.byref(p)))
+    return p
+
+
+def _h2d(dst, arr):
+    arr = np.ascontiguousarray(arr)
+    _check(lib().dd_memcpy_h2d(dst, arr.ctypes.data_as(ctypes.c_void_p),
+                               ctypes.c_int64(arr.nbytes)))
+
+
+def _d2h(src_ptr, nbytes, dtype):
+    out = np.empty(nbytes // np.dtype(dtype).itemsize, dtype=dtype)
+    _check(lib().dd_memcpy_d2h(out.ctypes.data_as(ctypes.c_void_p),
+                               ctypes.c_void_p(src_ptr), ctypes.c_int64(nbytes)))
+    return out
+
+
+class DeviceBatch:
+    """Device-resident batch uploaded from the column-dict convention."""
+
+    def __init__(self, cols):
+        self.cols = cols
+        self.n_rows = (len(cols[0]["offsets"]) - 1 if cols[0]["dtype"] == "utf8"
+                       else len(cols[0]["data"]))
+        self._bufs = []
+        self.desc = BatchDesc()
+        self.desc.n_rows = self.n_rows
+        self.desc.n_cols = len(cols)
+        for i, col in enumerate(cols):
+            cd = self.desc.cols[i]
+            cd.dtype = DTYPE_CODE[col["dtype"]]
+            if col["dtype"] == "utf8":
+                data = np.ascontiguousarray(col["data"], dtype=np.uint8)
+                off = np.ascontiguousarray(col["offsets"], dtype=np.int32)
+                b = self._up(data)
+                cd.data = b
+                cd.offsets = self._up(off)
+                cd.data_len = int(data.nbytes)
+            else:
+                npdt = FIXED_NP[col["dtype"]]
+                data = np.ascontiguousarray(col["data"], dtype=npdt)
+                cd.data = self._up(data)
+                cd.data_len = 0
+            if col.get("valid") is not None:
+                cd.validity = self._up(np.ascontiguousarray(col["valid"], np.uint8))
+            if col["dtype"] == "dict32":
+                cd.dict_bytes = self._up(np.ascontiguousarray(col["dict_bytes"], np.uint8))
+                cd.dict_offsets = self._up(np.ascontiguousarray(col["dict_offsets"], np.int32))
+                cd.dict_n = len(col["dict_offsets"]) - 1
+
+    def _up(self, arr):
+        p = _dev_alloc(arr.nbytes)
+        _h2d(p, arr)
+        self._bufs.append(p)
+        return p.value
+
+    def free(self):
+        for b in self._bufs:
+            lib().dd_dev_free(b)
+        self._bufs = []
+
+
+class Partitioner:
+    """Wraps dd_partitioner_*: the producer-head replacement."""
+
+    def __init__(self, batch: DeviceBatch, key_idx, nparts):
+        self.batch = batch
+        self.nparts = nparts
+        self.key_idx = list(key_idx)
+        keys = (ctypes.c_int32 * len(key_idx))(*key_idx)
+        self.h = ctypes.c_void_p()
+        _check(lib().dd_partitioner_create(ctypes.byref(batch.desc), keys,
+                                           len(key_idx), ctypes.c_uint32(nparts),
+                                           ctypes.byref(self.h)))
+
+    def run(self, stream=None):
+        _check(lib().dd_partitioner_run(self.h, stream))
+
+    def sync(self):
+        _check(lib().dd_device_sync())
+
+    def pids(self):
+        p = lib().dd_partitioner_pids(self.h)
+        return _d2h(p, self.batch.n_rows * 4, np.uint32)
+
+    def row_offsets(self):
+        out = np.empty(self.nparts + 1, dtype=np.int64)
+        _check(lib().dd_partitioner_row_offsets(
+            self.h, out.ctypes.data_as(ctypes.POINTER(ctypes.c_int64))))
+        return out
+
+    def byte_offsets(self, col):
+        out = np.empty(self.nparts + 1, dtype=np.int64)
+        _check(lib().dd_partitioner_byte_offsets(
+            self.h, col, out.ctypes.data_as(ctypes.POINTER(ctypes.c_int64))))
+        return out
+
+    def kernel_ms(self):
+        out = (ctypes.c_float * 3)()
+        _check(lib().dd_partitioner_kernel_ms(self.h, out))
+        return list(out)
+
+    def col_out(self, i):
+        """Download partitioned output of column i (tests/smoke only)."""
+        col = self.batch.cols[i]
+        n = self.batch.n_rows
+        res = {"dtype": col["dtype"]}
+        if col["dtype"] == "utf8":
+            dl = int(np.asarray(col["data"], dtype=np.uint8).nbytes)
+            res["data"] = _d2h(lib().dd_partitioner_col_data(self.h, i), dl, np.uint8)
+            res["lengths"] = _d2h(lib().dd_partitioner_col_lengths(self.h, i), n * 4,
+                                  np.uint32)
+        else:
+            npdt = FIXED_NP[col["dtype"]]
+            res["data"] = _d2h(lib().dd_partitioner_col_data(self.h, i),
+                               n * ELEM_SIZE[col["dtype"]], npdt)
+        vp = lib().dd_partitioner_col_validity(self.h, i)
+        if vp:
+            res["valid"] = _d2h(vp, n, np.uint8)
+        return res
+
+    def destroy(self):
+        if self.h:
+            lib().dd_partitioner_destroy(self.h)
+            self.h = None
+
+
+class Comm:
+    """RCCL communicator (dd_comm_*)."""
+
+    def __init__(self, unique_id: bytes, rank: int, nranks: int):
+        assert len(unique_id) == UNIQUE_ID_BYTES
+        self.rank, self.nranks = rank, nranks
+        self.h = ctypes.c_void_p()
+        buf = ctypes.create_string_buffer(unique_id, UNIQUE_ID_BYTES)
+        _check(lib().dd_comm_init(buf, rank, nranks, ctypes.byref(self.h)))
+
+    @staticmethod
+    def unique_id() -> bytes:
+        buf = ctypes.create_string_buffer(UNIQUE_ID_BYTES)
+        _check(lib().dd_comm_unique_id(buf))
+        return buf.raw
+
+    def exchange(self, part: Partitioner, stream=None):
+        h = ctypes.c_void_p()
+        _check(lib().dd_exchange_run(self.h, part.h, stream, ctypes.byref(h)))
+        return Exchanged(h, part, self)
+
+    def destroy(self):
+        if self.h:
+            lib().dd_comm_destroy(self.h)
+            self.h = None
+
+
+class Exchanged:
+    def __init__(self, h, part, comm):
+        self.h = h
+        self.part = part
+        self.comm = comm
+
+    @property
+    def total_rows(self):
+        return lib().dd_exchanged_total_rows(self.h)
+
+    def row_counts(self):
+        out = np.empty(self.comm.nranks * (self.part.nparts // self.comm.nranks),
+                       dtype=np.int64)
+        _check(lib().dd_exchanged_row_counts(
+            self.h, out.ctypes.data_as(ctypes.POINTER(ctypes.c_int64))))
+        return out.reshape(self.comm.nranks, -1)
+
+    def byte_counts(self, col):
+        out = np.empty(self.comm.nranks * (self.part.nparts // self.comm.nranks),
+                       dtype=np.int64)
+        _check(lib().dd_exchanged_byte_counts(
+            self.h, col, out.ctypes.data_as(ctypes.POINTER(ctypes.c_int64))))
+        return out.reshape(self.comm.nranks, -1)
+
+    def col_data(self, i):
+        col = self.part.batch.cols[i]
+        n = self.total_rows
+        if col["dtype"] == "utf8":
+            total_b = int(self.byte_counts(i).sum())
+            data = _d2h(lib().dd_exchanged_col_data(self.h, i), total_b, np.uint8)
+            lens = _d2h(lib().dd_exchanged_col_lengths(self.h, i), n * 4, np.uint32)
+            return {"dtype": "utf8", "data": data, "lengths": lens}
+        npdt = FIXED_NP[col["dtype"]]
+        return {"dtype": col["dtype"],
+                "data": _d2h(lib().dd_exchanged_col_data(self.h, i),
+                             n * ELEM_SIZE[col["dtype"]], npdt)}
+
+    def col_validity(self, i):
+        vp = lib().dd_exchanged_col_validity(self.h, i)
+        return _d2h(vp, self.total_rows, np.uint8) if vp else None
+
+    def stats(self):
+        ms = ctypes.c_float()
+        eg = ctypes.c_int64()
+        _check(lib().dd_exchanged_stats(self.h, ctypes.byref(ms), ctypes.byref(eg)))
+        return float(ms.value), int(eg.value)
+
+    def destroy(self):
+        if self.h:
+            lib().dd_exchanged_destroy(self.h)
+            self.h = None

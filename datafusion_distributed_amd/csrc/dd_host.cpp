@@ -1,0 +1,633 @@
+/* dd_host.cpp — host side of libdd_shuffle.so: the C ABI (include/dd_shuffle.h), the task
+ * cache mirroring the reference's worker execute path, and the RCCL/xGMI exchange replacing
+ * the Arrow-Flight data plane. See header citations in include/dd_shuffle.h and DESIGN.md.
+ *
+ * Product path only: no CPU fallback anywhere — a missing/failed HIP device is a hard error
+ * (DD_ERR_NO_DEVICE), by design.
+ */
+
+#include <hip/hip_runtime.h>
+#include <rccl/rccl.h>
+
+#include <cmath>
+#include <cstdio>
+#include <cstring>
+#include <map>
+#include <mutex>
+#include <string>
+#include <vector>
+
+#include "../../include/dd_shuffle.h"
+#include "dd_internal.h"
+
+/* ---------------- error plumbing ---------------- */
+
+static thread_local std::string g_last_error;
+
+static dd_status set_err(dd_status s, const std::string &msg) {
+    g_last_error = msg;
+    return s;
+}
+
+extern "C" const char *dd_last_error(void) { return g_last_error.c_str(); }
+extern "C" const char *dd_version(void) { return "dd_shuffle 0.1 (gfx950)"; }
+
+#define HIP_TRY(expr)                                                                        \
+    do {                                                                                     \
+        hipError_t _e = (expr);                                                              \
+        if (_e != hipSuccess)                                                                \
+            return set_err(DD_ERR_HIP, std::string(#expr) + ": " + hipGetErrorString(_e));   \
+    } while (0)
+
+#define NCCL_TRY(expr)                                                                       \
+    do {                                                                                     \
+        ncclResult_t _r = (expr);                                                            \
+        if (_r != ncclSuccess)                                                               \
+            return set_err(DD_ERR_RCCL, std::string(#expr) + ": " + ncclGetErrorString(_r)); \
+    } while (0)
+
+extern "C" int dd_device_count(void) {
+    int n = 0;
+    if (hipGetDeviceCount(&n) != hipSuccess) return 0;
+    return n;
+}
+
+/* ---------------- device helpers ---------------- */
+
+extern "C" dd_status dd_dev_alloc(int64_t bytes, void **out) {
+    if (dd_device_count() == 0) return set_err(DD_ERR_NO_DEVICE, "no HIP device");
+    HIP_TRY(hipMalloc(out, bytes > 0 ? (size_t)bytes : 1));
+    return DD_OK;
+}
+extern "C" dd_status dd_dev_free(void *p) {
+    HIP_TRY(hipFree(p));
+    return DD_OK;
+}
+extern "C" dd_status dd_memcpy_h2d(void *dst, const void *src, int64_t bytes) {
+    HIP_TRY(hipMemcpy(dst, src, (size_t)bytes, hipMemcpyHostToDevice));
+    return DD_OK;
+}
+extern "C" dd_status dd_memcpy_d2h(void *dst, const void *src, int64_t bytes) {
+    HIP_TRY(hipMemcpy(dst, src, (size_t)bytes, hipMemcpyDeviceToHost));
+    return DD_OK;
+}
+extern "C" dd_status dd_device_sync(void) {
+    HIP_TRY(hipDeviceSynchronize());
+    return DD_OK;
+}
+
+/* ---------------- partitioner ---------------- */
+
+static int fixed_elem_size(int dtype) {
+    switch (dtype) {
+    case DD_DT_U8:
+    case DD_DT_BOOL:
+        return 1;
+    case DD_DT_I16:
+        return 2;
+    case DD_DT_I32:
+    case DD_DT_F32:
+    case DD_DT_DICT32:
+        return 4;
+    case DD_DT_I64:
+    case DD_DT_F64:
+        return 8;
+    default:
+        return 0; /* utf8 */
+    }
+}
+
+struct dd_partitioner {
+    dd_batch_desc batch;
+    dd_kargs ka;
+    uint32_t nparts = 0;
+    int nbits = 0;
+    int64_t nchunks = 0, chunk_rows = 0;
+    size_t lds_k1 = 0, lds_k3 = 0;
+
+    /* device buffers (owned) */
+    uint32_t *pid = nullptr;
+    uint32_t *counts = nullptr;          /* [nchunks][P] */
+    uint32_t *partials = nullptr;        /* [RANGES][P] */
+    uint64_t *part_offsets = nullptr;    /* [P+1] */
+    uint32_t *bcounts = nullptr;         /* [nvar][nchunks][P] */
+    uint32_t *bpartials = nullptr;       /* [nvar][RANGES][P] */
+    uint64_t *part_boffsets = nullptr;   /* [nvar][P+1] */
+    uint64_t *dict_hashes[DD_KMAX_COLS] = {};
+    void *out_data[DD_KMAX_COLS] = {};
+    uint8_t *out_valid[DD_KMAX_COLS] = {};
+    uint32_t *out_lengths[DD_KMAX_COLS] = {};
+
+    hipEvent_t ev[4] = {};
+    bool has_run = false;
+
+    ~dd_partitioner() {
+        hipFree(pid);
+        hipFree(counts);
+        hipFree(partials);
+        hipFree(part_offsets);
+        hipFree(bcounts);
+        hipFree(bpartials);
+        hipFree(part_boffsets);
+        for (int i = 0; i < DD_KMAX_COLS; i++) {
+            hipFree(dict_hashes[i]);
+            hipFree(out_data[i]);
+            hipFree(out_valid[i]);
+            hipFree(out_lengths[i]);
+        }
+        for (auto &e : ev)
+            if (e) hipEventDestroy(e);
+    }
+};
+
+extern "C" dd_status dd_partitioner_create(const dd_batch_desc *batch, const int32_t *key_cols,
+                                           int32_t n_keys, uint32_t n_partitions,
+                                           dd_partitioner **out) {
+    if (!batch || !key_cols || !out) return set_err(DD_ERR_INVALID, "null argument");
+    if (dd_device_count() == 0)
+        return set_err(DD_ERR_NO_DEVICE,
+                       "no HIP device: the product path has no CPU fallback (DESIGN.md §2)");
+    if (batch->n_cols < 1 || batch->n_cols > DD_MAX_COLS)
+        return set_err(DD_ERR_INVALID, "n_cols out of range");
+    if (n_keys < 1 || n_keys > DD_MAX_KEYS) return set_err(DD_ERR_INVALID, "n_keys out of range");
+    if (n_partitions < 1 || n_partitions > DD_MAX_PARTITIONS)
+        return set_err(DD_ERR_UNSUPPORTED, "n_partitions outside round-1 cap");
+    if (batch->n_rows < 0 || batch->n_rows >= (int64_t)UINT32_MAX)
+        return set_err(DD_ERR_UNSUPPORTED, "n_rows >= 2^32: chunk the batch");
+
+    auto p = new dd_partitioner();
+    p->batch = *batch;
+    p->nparts = n_partitions;
+    p->nbits = 0;
+    while ((1u << p->nbits) < n_partitions) p->nbits++;
+
+    const int64_t n = batch->n_rows;
+    int64_t nchunks = (n + 4095) / 4096;
+    if (nchunks < 4) nchunks = 4;
+    if (nchunks > 8192) nchunks = 8192;
+    nchunks = (nchunks + 3) & ~3LL;
+    p->nchunks = nchunks;
+    p->chunk_rows = (n + nchunks - 1) / nchunks;
+    if (p->chunk_rows < 1) p->chunk_rows = 1;
+
+    dd_kargs &ka = p->ka;
+    memset(&ka, 0, sizeof(ka));
+    ka.n_rows = n;
+    ka.n_cols = batch->n_cols;
+    ka.n_keys = n_keys;
+
+    auto fail = [&](dd_status s, const char *m) {
+        delete p;
+        return set_err(s, m);
+    };
+
+    for (int k = 0; k < n_keys; k++) {
+        if (key_cols[k] < 0 || key_cols[k] >= batch->n_cols)
+            return fail(DD_ERR_INVALID, "key column index out of range");
+        ka.key_idx[k] = key_cols[k];
+    }
+
+    for (int c = 0; c < batch->n_cols; c++) {
+        const dd_col_desc &cd = batch->cols[c];
+        dd_kcol &kc = ka.cols[c];
+        kc.dtype = cd.dtype;
+        kc.elem = fixed_elem_size(cd.dtype);
+        kc.data = cd.data;
+        kc.valid = cd.validity;
+        kc.offsets = cd.offsets;
+        if (cd.dtype == DD_DT_UTF8) {
+            if (ka.n_var >= DD_KMAX_VAR) return fail(DD_ERR_UNSUPPORTED, "too many var columns");
+            if (cd.data_len < 0 || cd.data_len >= (int64_t)UINT32_MAX)
+                return fail(DD_ERR_UNSUPPORTED, "utf8 data >= 4 GB: chunk the batch");
+            ka.var_idx[ka.n_var++] = c;
+        }
+    }
+
+    const int64_t P = n_partitions;
+    const int nvar = ka.n_var;
+    p->lds_k1 = (size_t)WAVES_PER_BLOCK_H * P * 4 * (1 + nvar);
+    p->lds_k3 = (size_t)WAVES_PER_BLOCK_H * P * 8 * (1 + nvar) + WAVES_PER_BLOCK_H * 64 * 4;
+    if (p->lds_k1 > 163840 || p->lds_k3 > 163840)
+        return fail(DD_ERR_UNSUPPORTED, "partition count x var columns exceeds LDS budget");
+
+    auto halloc = [&](void **ptr, size_t bytes) {
+        return hipMalloc(ptr, bytes > 0 ? bytes : 1) == hipSuccess;
+    };
+    bool ok = halloc((void **)&p->pid, (size_t)n * 4) &&
+              halloc((void **)&p->counts, (size_t)nchunks * P * 4) &&
+              halloc((void **)&p->partials, (size_t)DD_SCAN_RANGES * P * 4) &&
+              halloc((void **)&p->part_offsets, (size_t)(P + 1) * 8);
+    if (ok && nvar > 0)
+        ok = halloc((void **)&p->bcounts, (size_t)nvar * nchunks * P * 4) &&
+             halloc((void **)&p->bpartials, (size_t)nvar * DD_SCAN_RANGES * P * 4) &&
+             halloc((void **)&p->part_boffsets, (size_t)nvar * (P + 1) * 8);
+    if (!ok) return fail(DD_ERR_HIP, "allocation failed (workspace)");
+
+    for (int c = 0; c < batch->n_cols && ok; c++) {
+        const dd_col_desc &cd = batch->cols[c];
+        dd_kcol &kc = ka.cols[c];
+        if (cd.dtype == DD_DT_UTF8) {
+            ok = halloc(&p->out_data[c], (size_t)cd.data_len) &&
+                 halloc((void **)&p->out_lengths[c], (size_t)n * 4);
+            kc.out_lengths = p->out_lengths[c];
+        } else {
+            ok = halloc(&p->out_data[c], (size_t)n * kc.elem);
+        }
+        kc.out_data = p->out_data[c];
+        if (ok && cd.validity) {
+            ok = halloc((void **)&p->out_valid[c], (size_t)n);
+            kc.out_valid = p->out_valid[c];
+        }
+        if (ok && cd.dtype == DD_DT_DICT32) {
+            ok = halloc((void **)&p->dict_hashes[c], (size_t)cd.dict_n * 8);
+            if (ok) {
+                hipError_t e = dd_launch_dict_hashes((const uint8_t *)cd.dict_bytes,
+                                                     cd.dict_offsets, cd.dict_n,
+                                                     p->dict_hashes[c], nullptr);
+                ok = (e == hipSuccess);
+            }
+            kc.dict_hashes = p->dict_hashes[c];
+        }
+    }
+    if (!ok) return fail(DD_ERR_HIP, "allocation failed (outputs)");
+
+    for (auto &e : p->ev)
+        if (hipEventCreate(&e) != hipSuccess) return fail(DD_ERR_HIP, "event create failed");
+
+    *out = p;
+    return DD_OK;
+}
+
+extern "C" dd_status dd_partitioner_run(dd_partitioner *p, void *stream) {
+    if (!p) return set_err(DD_ERR_INVALID, "null partitioner");
+    hipStream_t s = (hipStream_t)stream;
+    HIP_TRY(hipEventRecord(p->ev[0], s));
+    HIP_TRY(dd_launch_hash_count(&p->ka, p->nchunks, p->chunk_rows, p->nparts, p->nbits, p->pid,
+                                 p->counts, p->bcounts, p->lds_k1, s));
+    HIP_TRY(hipEventRecord(p->ev[1], s));
+    HIP_TRY(dd_launch_scan(p->counts, p->nchunks, p->nparts, DD_SCAN_RANGES, p->partials,
+                           p->part_offsets, s));
+    for (int v = 0; v < p->ka.n_var; v++) {
+        HIP_TRY(dd_launch_scan(p->bcounts + (size_t)v * p->nchunks * p->nparts, p->nchunks,
+                               p->nparts, DD_SCAN_RANGES,
+                               p->bpartials + (size_t)v * DD_SCAN_RANGES * p->nparts,
+                               p->part_boffsets + (size_t)v * (p->nparts + 1), s));
+    }
+    HIP_TRY(hipEventRecord(p->ev[2], s));
+    HIP_TRY(dd_launch_scatter(&p->ka, p->nchunks, p->chunk_rows, p->nparts, p->nbits, p->pid,
+                              p->counts, p->part_offsets, p->bcounts, p->part_boffsets,
+                              p->lds_k3, s));
+    HIP_TRY(hipEventRecord(p->ev[3], s));
+    p->has_run = true;
+    return DD_OK;
+}
+
+extern "C" void dd_partitioner_destroy(dd_partitioner *p) { delete p; }
+
+extern "C" const uint32_t *dd_partitioner_pids(const dd_partitioner *p) { return p->pid; }
+extern "C" const void *dd_partitioner_col_data(const dd_partitioner *p, int32_t c) {
+    return (c >= 0 && c < p->batch.n_cols) ? p->out_data[c] : nullptr;
+}
+extern "C" const uint8_t *dd_partitioner_col_validity(const dd_partitioner *p, int32_t c) {
+    return (c >= 0 && c < p->batch.n_cols) ? p->out_valid[c] : nullptr;
+}
+extern "C" const uint32_t *dd_partitioner_col_lengths(const dd_partitioner *p, int32_t c) {
+    return (c >= 0 && c < p->batch.n_cols) ? p->out_lengths[c] : nullptr;
+}
+
+extern "C" dd_status dd_partitioner_row_offsets(const dd_partitioner *p, int64_t *host_out) {
+    if (!p->has_run) return set_err(DD_ERR_INVALID, "partitioner has not run");
+    HIP_TRY(hipMemcpy(host_out, p->part_offsets, (size_t)(p->nparts + 1) * 8,
+                      hipMemcpyDeviceToHost));
+    return DD_OK;
+}
+
+extern "C" dd_status dd_partitioner_byte_offsets(const dd_partitioner *p, int32_t col,
+                                                 int64_t *host_out) {
+    if (!p->has_run) return set_err(DD_ERR_INVALID, "partitioner has not run");
+    for (int v = 0; v < p->ka.n_var; v++) {
+        if (p->ka.var_idx[v] == col) {
+            HIP_TRY(hipMemcpy(host_out, p->part_boffsets + (size_t)v * (p->nparts + 1),
+                              (size_t)(p->nparts + 1) * 8, hipMemcpyDeviceToHost));
+            return DD_OK;
+        }
+    }
+    return set_err(DD_ERR_INVALID, "not a var column");
+}
+
+extern "C" dd_status dd_partitioner_kernel_ms(const dd_partitioner *p, float out_ms[3]) {
+    if (!p->has_run) return set_err(DD_ERR_INVALID, "partitioner has not run");
+    HIP_TRY(hipEventSynchronize(p->ev[3]));
+    HIP_TRY(hipEventElapsedTime(&out_ms[0], p->ev[0], p->ev[1]));
+    HIP_TRY(hipEventElapsedTime(&out_ms[1], p->ev[1], p->ev[2]));
+    HIP_TRY(hipEventElapsedTime(&out_ms[2], p->ev[2], p->ev[3]));
+    return DD_OK;
+}
+
+/* ---------------- task cache (mirrors src/worker/task_data.rs, worker_service.rs:12,31) --- */
+
+struct TaskKeyCmp {
+    bool operator()(const dd_task_key &a, const dd_task_key &b) const {
+        return memcmp(&a, &b, sizeof(a)) < 0;
+    }
+};
+
+static std::mutex g_tasks_mu;
+static std::map<dd_task_key, dd_partitioner *, TaskKeyCmp> g_tasks;
+
+extern "C" dd_status dd_set_plan(const dd_task_key *key, const dd_batch_desc *batch,
+                                 const int32_t *key_cols, int32_t n_keys,
+                                 uint32_t n_partitions) {
+    if (!key) return set_err(DD_ERR_INVALID, "null key");
+    dd_partitioner *p = nullptr;
+    dd_status st = dd_partitioner_create(batch, key_cols, n_keys, n_partitions, &p);
+    if (st != DD_OK) return st;
+    std::lock_guard<std::mutex> g(g_tasks_mu);
+    auto it = g_tasks.find(*key);
+    if (it != g_tasks.end()) {
+        delete it->second;
+        it->second = p;
+    } else {
+        g_tasks[*key] = p;
+    }
+    return DD_OK;
+}
+
+extern "C" dd_status dd_execute_task(const dd_task_key *key, uint32_t part_lo, uint32_t part_hi,
+                                     void *stream, dd_partitioner **out) {
+    if (!key || !out) return set_err(DD_ERR_INVALID, "null argument");
+    dd_partitioner *p = nullptr;
+    {
+        std::lock_guard<std::mutex> g(g_tasks_mu);
+        auto it = g_tasks.find(*key);
+        if (it == g_tasks.end())
+            return set_err(DD_ERR_NOT_FOUND,
+                           "unknown TaskKey (no plan was set: cf. impl_execute_task.rs:29-34)");
+        p = it->second;
+    }
+    if (part_lo > part_hi || part_hi > p->nparts)
+        return set_err(DD_ERR_INVALID, "partition range out of bounds");
+    if (!p->has_run) {
+        dd_status st = dd_partitioner_run(p, stream);
+        if (st != DD_OK) return st;
+    }
+    *out = p;
+    return DD_OK;
+}
+
+extern "C" dd_status dd_drop_task(const dd_task_key *key) {
+    std::lock_guard<std::mutex> g(g_tasks_mu);
+    auto it = g_tasks.find(*key);
+    if (it == g_tasks.end()) return set_err(DD_ERR_NOT_FOUND, "unknown TaskKey");
+    delete it->second;
+    g_tasks.erase(it);
+    return DD_OK;
+}
+
+/* ---------------- RCCL exchange ---------------- */
+
+struct dd_comm {
+    ncclComm_t comm = nullptr;
+    int rank = 0, nranks = 1;
+    ~dd_comm() {
+        if (comm) ncclCommDestroy(comm);
+    }
+};
+
+extern "C" dd_status dd_comm_unique_id(void *bytes128) {
+    ncclUniqueId id;
+    NCCL_TRY(ncclGetUniqueId(&id));
+    memcpy(bytes128, &id, sizeof(id));
+    return DD_OK;
+}
+
+extern "C" dd_status dd_comm_init(const void *bytes128, int rank, int nranks, dd_comm **out) {
+    if (dd_device_count() == 0) return set_err(DD_ERR_NO_DEVICE, "no HIP device");
+    auto c = new dd_comm();
+    c->rank = rank;
+    c->nranks = nranks;
+    ncclUniqueId id;
+    memcpy(&id, bytes128, sizeof(id));
+    ncclResult_t r = ncclCommInitRank(&c->comm, nranks, id, rank);
+    if (r != ncclSuccess) {
+        delete c;
+        return set_err(DD_ERR_RCCL, std::string("ncclCommInitRank: ") + ncclGetErrorString(r));
+    }
+    *out = c;
+    return DD_OK;
+}
+
+extern "C" void dd_comm_destroy(dd_comm *c) { delete c; }
+
+struct dd_exchanged {
+    int32_t n_cols = 0;
+    int nranks = 1;
+    uint32_t P = 0;
+    int64_t total_rows = 0;
+    std::vector<int64_t> row_counts;                 /* [nranks][P] */
+    std::vector<std::vector<int64_t>> byte_counts;   /* per col: [nranks][P] (var only) */
+    void *data[DD_MAX_COLS] = {};
+    uint8_t *valid[DD_MAX_COLS] = {};
+    uint32_t *lengths[DD_MAX_COLS] = {};
+    float ms = 0;
+    int64_t egress = 0;
+    hipEvent_t e0 = nullptr, e1 = nullptr;
+    ~dd_exchanged() {
+        for (int i = 0; i < DD_MAX_COLS; i++) {
+            hipFree(data[i]);
+            hipFree(valid[i]);
+            hipFree(lengths[i]);
+        }
+        if (e0) hipEventDestroy(e0);
+        if (e1) hipEventDestroy(e1);
+    }
+};
+
+extern "C" dd_status dd_exchange_run(dd_comm *c, const dd_partitioner *p, void *stream,
+                                     dd_exchanged **out) {
+    if (!c || !p || !out) return set_err(DD_ERR_INVALID, "null argument");
+    if (!p->has_run) return set_err(DD_ERR_INVALID, "partitioner has not run");
+    if (p->nparts % c->nranks != 0)
+        return set_err(DD_ERR_INVALID, "P_total must be nranks * partitions-per-consumer "
+                                       "(scale_partitioning, common.rs:18-30)");
+    hipStream_t s = (hipStream_t)stream;
+    const uint32_t P = p->nparts / c->nranks;
+    const int R = c->nranks;
+    const int nvar = p->ka.n_var;
+
+    /* 1) size matrix: my per-partition row counts (+ per var col byte counts), allgathered.
+     * meta layout per rank: [P_total rows][nvar * P_total bytes] (u64) */
+    const size_t meta_n = (size_t)p->nparts * (1 + nvar);
+    std::vector<int64_t> my_meta(meta_n);
+    std::vector<int64_t> off_h(p->nparts + 1);
+    dd_status st = dd_partitioner_row_offsets(p, off_h.data());
+    if (st != DD_OK) return st;
+    for (uint32_t q = 0; q < p->nparts; q++) my_meta[q] = off_h[q + 1] - off_h[q];
+    std::vector<std::vector<int64_t>> boff_h(nvar, std::vector<int64_t>(p->nparts + 1));
+    for (int v = 0; v < nvar; v++) {
+        st = dd_partitioner_byte_offsets(p, p->ka.var_idx[v], boff_h[v].data());
+        if (st != DD_OK) return st;
+        for (uint32_t q = 0; q < p->nparts; q++)
+            my_meta[(size_t)(1 + v) * p->nparts + q] = boff_h[v][q + 1] - boff_h[v][q];
+    }
+    int64_t *d_meta_in = nullptr, *d_meta_all = nullptr;
+    HIP_TRY(hipMalloc(&d_meta_in, meta_n * 8));
+    HIP_TRY(hipMalloc(&d_meta_all, (size_t)R * meta_n * 8));
+    HIP_TRY(hipMemcpyAsync(d_meta_in, my_meta.data(), meta_n * 8, hipMemcpyHostToDevice, s));
+    NCCL_TRY(ncclAllGather(d_meta_in, d_meta_all, meta_n, ncclInt64, c->comm, s));
+    std::vector<int64_t> all_meta((size_t)R * meta_n);
+    HIP_TRY(hipStreamSynchronize(s));
+    HIP_TRY(hipMemcpy(all_meta.data(), d_meta_all, (size_t)R * meta_n * 8,
+                      hipMemcpyDeviceToHost));
+    hipFree(d_meta_in);
+    hipFree(d_meta_all);
+
+    auto e = new dd_exchanged();
+    e->n_cols = p->batch.n_cols;
+    e->nranks = R;
+    e->P = P;
+    e->row_counts.assign((size_t)R * P, 0);
+    e->byte_counts.assign(p->batch.n_cols, {});
+    /* my window = partitions [P*rank, P*(rank+1)) of every producer */
+    for (int r = 0; r < R; r++)
+        for (uint32_t q = 0; q < P; q++) {
+            e->row_counts[(size_t)r * P + q] =
+                all_meta[(size_t)r * meta_n + (size_t)P * c->rank + q];
+            e->total_rows += e->row_counts[(size_t)r * P + q];
+        }
+    std::vector<int64_t> recv_rows_per_producer(R, 0);
+    for (int r = 0; r < R; r++)
+        for (uint32_t q = 0; q < P; q++)
+            recv_rows_per_producer[r] += e->row_counts[(size_t)r * P + q];
+
+    auto fail = [&](dd_status sc, const char *m) {
+        delete e;
+        return set_err(sc, m);
+    };
+
+    /* allocate receive buffers */
+    std::vector<std::vector<int64_t>> recv_bytes_per_producer(p->batch.n_cols,
+                                                              std::vector<int64_t>(R, 0));
+    for (int ci = 0; ci < p->batch.n_cols; ci++) {
+        const dd_kcol &kc = p->ka.cols[ci];
+        if (kc.elem > 0) {
+            if (hipMalloc(&e->data[ci], (size_t)e->total_rows * kc.elem + 1) != hipSuccess)
+                return fail(DD_ERR_HIP, "recv alloc");
+        } else {
+            /* var col: find its v index, total bytes of my window */
+            int v = -1;
+            for (int vv = 0; vv < nvar; vv++)
+                if (p->ka.var_idx[vv] == ci) v = vv;
+            e->byte_counts[ci].assign((size_t)R * P, 0);
+            int64_t total_b = 0;
+            for (int r = 0; r < R; r++)
+                for (uint32_t q = 0; q < P; q++) {
+                    int64_t b = all_meta[(size_t)r * meta_n + (size_t)(1 + v) * p->nparts +
+                                         (size_t)P * c->rank + q];
+                    e->byte_counts[ci][(size_t)r * P + q] = b;
+                    recv_bytes_per_producer[ci][r] += b;
+                    total_b += b;
+                }
+            if (hipMalloc(&e->data[ci], (size_t)total_b + 1) != hipSuccess)
+                return fail(DD_ERR_HIP, "recv alloc (var)");
+            if (hipMalloc((void **)&e->lengths[ci], (size_t)e->total_rows * 4 + 1) != hipSuccess)
+                return fail(DD_ERR_HIP, "recv alloc (lengths)");
+        }
+        if (kc.valid) {
+            if (hipMalloc((void **)&e->valid[ci], (size_t)e->total_rows + 1) != hipSuccess)
+                return fail(DD_ERR_HIP, "recv alloc (validity)");
+        }
+    }
+
+    if (hipEventCreate(&e->e0) != hipSuccess || hipEventCreate(&e->e1) != hipSuccess)
+        return fail(DD_ERR_HIP, "event create");
+
+    /* 2) grouped send/recv: one contiguous slice per (column-stream, peer).
+     * send slice for peer j = partitions [P*j, P*(j+1)) of my partition-major output;
+     * recv placed producer-major. */
+    HIP_TRY(hipEventRecord(e->e0, s));
+    NCCL_TRY(ncclGroupStart());
+    for (int j = 0; j < R; j++) {
+        const int64_t srow0 = off_h[(size_t)P * j], srow1 = off_h[(size_t)P * (j + 1)];
+        int64_t rrow0 = 0;
+        for (int r = 0; r < j; r++) rrow0 += recv_rows_per_producer[r];
+        const int64_t rrows = recv_rows_per_producer[j];
+        for (int ci = 0; ci < p->batch.n_cols; ci++) {
+            const dd_kcol &kc = p->ka.cols[ci];
+            if (kc.elem > 0) {
+                int64_t sb = (srow1 - srow0) * kc.elem;
+                if (sb > 0)
+                    NCCL_TRY(ncclSend((const uint8_t *)p->out_data[ci] + srow0 * kc.elem, sb,
+                                      ncclUint8, j, c->comm, s));
+                int64_t rb = rrows * kc.elem;
+                if (rb > 0)
+                    NCCL_TRY(ncclRecv((uint8_t *)e->data[ci] + rrow0 * kc.elem, rb, ncclUint8,
+                                      j, c->comm, s));
+                if (j != c->rank) e->egress += sb;
+            } else {
+                int v = -1;
+                for (int vv = 0; vv < nvar; vv++)
+                    if (p->ka.var_idx[vv] == ci) v = vv;
+                int64_t sb0 = boff_h[v][(size_t)P * j], sb1 = boff_h[v][(size_t)P * (j + 1)];
+                if (sb1 > sb0)
+                    NCCL_TRY(ncclSend((const uint8_t *)p->out_data[ci] + sb0, sb1 - sb0,
+                                      ncclUint8, j, c->comm, s));
+                int64_t rb0 = 0;
+                for (int r = 0; r < j; r++) rb0 += recv_bytes_per_producer[ci][r];
+                int64_t rb = recv_bytes_per_producer[ci][j];
+                if (rb > 0)
+                    NCCL_TRY(ncclRecv((uint8_t *)e->data[ci] + rb0, rb, ncclUint8, j, c->comm, s));
+                if ((srow1 - srow0) > 0)
+                    NCCL_TRY(ncclSend(p->out_lengths[ci] + srow0, (srow1 - srow0) * 4,
+                                      ncclUint8, j, c->comm, s));
+                if (rrows > 0)
+                    NCCL_TRY(ncclRecv((uint8_t *)e->lengths[ci] + rrow0 * 4, rrows * 4,
+                                      ncclUint8, j, c->comm, s));
+                if (j != c->rank) e->egress += (sb1 - sb0) + (srow1 - srow0) * 4;
+            }
+            if (kc.valid) {
+                if (srow1 > srow0)
+                    NCCL_TRY(ncclSend(p->out_valid[ci] + srow0, srow1 - srow0, ncclUint8, j,
+                                      c->comm, s));
+                if (rrows > 0)
+                    NCCL_TRY(ncclRecv(e->valid[ci] + rrow0, rrows, ncclUint8, j, c->comm, s));
+                if (j != c->rank) e->egress += srow1 - srow0;
+            }
+        }
+    }
+    NCCL_TRY(ncclGroupEnd());
+    HIP_TRY(hipEventRecord(e->e1, s));
+    HIP_TRY(hipStreamSynchronize(s));
+    HIP_TRY(hipEventElapsedTime(&e->ms, e->e0, e->e1));
+    *out = e;
+    return DD_OK;
+}
+
+extern "C" void dd_exchanged_destroy(dd_exchanged *e) { delete e; }
+extern "C" int64_t dd_exchanged_total_rows(const dd_exchanged *e) { return e->total_rows; }
+extern "C" const void *dd_exchanged_col_data(const dd_exchanged *e, int32_t c) {
+    return (c >= 0 && c < e->n_cols) ? e->data[c] : nullptr;
+}
+extern "C" const uint8_t *dd_exchanged_col_validity(const dd_exchanged *e, int32_t c) {
+    return (c >= 0 && c < e->n_cols) ? e->valid[c] : nullptr;
+}
+extern "C" const uint32_t *dd_exchanged_col_lengths(const dd_exchanged *e, int32_t c) {
+    return (c >= 0 && c < e->n_cols) ? e->lengths[c] : nullptr;
+}
+extern "C" dd_status dd_exchanged_row_counts(const dd_exchanged *e, int64_t *host_out) {
+    memcpy(host_out, e->row_counts.data(), e->row_counts.size() * 8);
+    return DD_OK;
+}
+extern "C" dd_status dd_exchanged_byte_counts(const dd_exchanged *e, int32_t col,
+                                              int64_t *host_out) {
+    if (col < 0 || col >= e->n_cols || e->byte_counts[col].empty())
+        return set_err(DD_ERR_INVALID, "not a var column");
+    memcpy(host_out, e->byte_counts[col].data(), e->byte_counts[col].size() * 8);
+    return DD_OK;
+}
+extern "C" dd_status dd_exchanged_stats(const dd_exchanged *e, float *ms,
+                                        int64_t *egress_bytes) {
+    *ms = e->ms;
+    *egress_bytes = e->egress;
+    return DD_OK;
+}

@@ -1,0 +1,67 @@
+/* dd_internal.h — shared between dd_kernels.hip and dd_host.cpp (not part of the ABI). */
+
+#ifndef DD_INTERNAL_H
+#define DD_INTERNAL_H
+
+#include <hip/hip_runtime.h>
+#include <stdint.h>
+
+#define DD_MAX_P 2048
+#define WAVES_PER_BLOCK_H 4 /* must match WAVES_PER_BLOCK in dd_kernels.hip */
+#define DD_KMAX_COLS 24
+#define DD_KMAX_KEYS 8
+#define DD_KMAX_VAR 4
+#define DD_SCAN_RANGES 64
+
+/* dtype codes mirror dd_dtype in include/dd_shuffle.h */
+enum {
+    DD_KDT_U8 = 1,
+    DD_KDT_I16 = 2,
+    DD_KDT_I32 = 3,
+    DD_KDT_I64 = 4,
+    DD_KDT_F32 = 5,
+    DD_KDT_F64 = 6,
+    DD_KDT_BOOL = 7,
+    DD_KDT_UTF8 = 8,
+    DD_KDT_DICT32 = 9,
+};
+
+struct dd_kcol {
+    int32_t dtype;
+    int32_t elem;               /* fixed elem size (1/2/4/8); 0 for var (utf8) data */
+    const void *data;           /* input values / utf8 bytes */
+    const uint8_t *valid;       /* unpacked u8, or null */
+    const int32_t *offsets;     /* utf8 input offsets[n+1] */
+    const uint64_t *dict_hashes;/* dict32: per-value hash */
+    void *out_data;             /* partition-major output values / bytes */
+    uint8_t *out_valid;         /* partition-major output validity (if valid != null) */
+    uint32_t *out_lengths;      /* utf8: partition-major per-row byte lengths */
+};
+
+struct dd_kargs {
+    int64_t n_rows;
+    int32_t n_cols;
+    int32_t n_keys;
+    int32_t n_var;
+    int32_t key_idx[DD_KMAX_KEYS];
+    int32_t var_idx[DD_KMAX_VAR];
+    dd_kcol cols[DD_KMAX_COLS];
+};
+
+extern "C" {
+hipError_t dd_launch_dict_hashes(const uint8_t *bytes, const int32_t *offsets, int64_t n,
+                                 uint64_t *out, hipStream_t s);
+hipError_t dd_launch_hash_count(const dd_kargs *a, int64_t nchunks, int64_t chunk_rows,
+                                uint32_t nparts, int nbits, uint32_t *pid_out,
+                                uint32_t *counts, uint32_t *bcounts, size_t lds_bytes,
+                                hipStream_t s);
+hipError_t dd_launch_scan(uint32_t *counts, int64_t nchunks, uint32_t nparts, int nranges,
+                          uint32_t *partials, uint64_t *part_offsets, hipStream_t s);
+hipError_t dd_launch_scatter(const dd_kargs *a, int64_t nchunks, int64_t chunk_rows,
+                             uint32_t nparts, int nbits, const uint32_t *pid_in,
+                             const uint32_t *chunk_off, const uint64_t *part_offsets,
+                             const uint32_t *chunk_boff, const uint64_t *part_boffsets,
+                             size_t lds_bytes, hipStream_t s);
+}
+
+#endif

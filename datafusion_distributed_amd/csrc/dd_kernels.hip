@@ -1,0 +1,425 @@
+/* dd_kernels.hip — CDNA4 (gfx950) kernels for the hash-repartition hot path.
+ *
+ * Replaces the per-batch hot loop of DataFusion 55's RepartitionExec hash partitioning
+ * (constructed at /root/reference/src/execution_plans/network_shuffle.rs:121-127 and
+ * /root/reference/src/distributed_planner/network_boundary.rs:100-103): per row, hash the
+ * key columns, part = hash % P_total, stable gather of every column into per-partition
+ * contiguous outputs. Semantics are the normative spec of DESIGN.md §3 (bit-exact vs
+ * oracle/dd_oracle.c).
+ *
+ * Decomposition: the input is split into contiguous row "chunks", one per
+ * 64-lane wave; blocks are 256 threads = 4 independent waves (no intra-block barriers on the
+ * hot path). Stability: chunks are ordered, a wave walks its chunk's 64-row groups in order,
+ * and lane order == row order inside a group; ballot-multisplit (log2(P) ballots) gives the
+ * stable intra-group rank.
+ *
+ *   K1 hash_count : h -> pid (stored u32) + per-(chunk, partition) row/byte histograms
+ *   K2 scan       : column-major exclusive scan of the histograms (3 passes, coalesced)
+ *   K3 scatter    : stable scatter of every column into partition-major output buffers
+ *
+ * All integer/byte work — HBM-bound; no MFMA (this is indexing, not GEMM-shaped work).
+ */
+
+#include <hip/hip_runtime.h>
+#include <stdint.h>
+
+#include "dd_internal.h"
+
+#define WAVE 64
+#define WAVES_PER_BLOCK 4
+#define BLOCK_THREADS (WAVE * WAVES_PER_BLOCK)
+
+/* ---------------- normative hash (must match oracle/dd_oracle.c bit-exactly) ------------ */
+
+__device__ __forceinline__ uint64_t dd_mix64(uint64_t x) {
+    x ^= x >> 30;
+    x *= 0xbf58476d1ce4e5b9ULL;
+    x ^= x >> 27;
+    x *= 0x94d049bb133111ebULL;
+    x ^= x >> 31;
+    return x;
+}
+
+__device__ __forceinline__ uint64_t dd_hash_bytes_dev(const uint8_t *p, int64_t len) {
+    uint64_t h = 0x9e3779b97f4a7c15ULL ^ ((uint64_t)len * 0xff51afd7ed558ccdULL);
+    int64_t i = 0;
+    for (; i + 8 <= len; i += 8) {
+        uint64_t c = 0;
+        /* byte loads: string starts are unaligned; correctness first */
+        for (int b = 0; b < 8; b++) c |= (uint64_t)p[i + b] << (8 * b);
+        h = dd_mix64(h ^ c);
+    }
+    if (i < len) {
+        uint64_t c = 0;
+        for (int64_t b = 0; b < len - i; b++) c |= (uint64_t)p[i + b] << (8 * b);
+        h = dd_mix64(h ^ c);
+    }
+    return h;
+}
+
+__device__ __forceinline__ uint64_t dd_canon_f64_dev(double v) {
+    if (v == 0.0) v = 0.0;
+    uint64_t b = __double_as_longlong(v);
+    if (v != v) b = 0x7ff8000000000000ULL;
+    return b;
+}
+
+__device__ __forceinline__ uint64_t dd_canon_f32_dev(float v) {
+    if (v == 0.0f) v = 0.0f;
+    uint32_t b = __float_as_uint(v);
+    if (v != v) b = 0x7fc00000u;
+    return (uint64_t)b;
+}
+
+/* value hash of a valid row (column described by a dd_kcol) */
+__device__ __forceinline__ uint64_t dd_value_hash_dev(const dd_kcol &c, int64_t i) {
+    switch (c.dtype) {
+    case DD_KDT_U8:
+        return dd_mix64((uint64_t)((const uint8_t *)c.data)[i]);
+    case DD_KDT_I16:
+        return dd_mix64((uint64_t)((const uint16_t *)c.data)[i]);
+    case DD_KDT_I32:
+        return dd_mix64((uint64_t)((const uint32_t *)c.data)[i]);
+    case DD_KDT_I64:
+        return dd_mix64(((const uint64_t *)c.data)[i]);
+    case DD_KDT_F32:
+        return dd_mix64(dd_canon_f32_dev(((const float *)c.data)[i]));
+    case DD_KDT_F64:
+        return dd_mix64(dd_canon_f64_dev(((const double *)c.data)[i]));
+    case DD_KDT_UTF8: {
+        int32_t o0 = c.offsets[i], o1 = c.offsets[i + 1];
+        return dd_hash_bytes_dev((const uint8_t *)c.data + o0, (int64_t)(o1 - o0));
+    }
+    case DD_KDT_DICT32: {
+        int32_t k = ((const int32_t *)c.data)[i];
+        return c.dict_hashes[k]; /* precomputed by k_dict_hashes */
+    }
+    default:
+        return 0;
+    }
+}
+
+__global__ void k_dict_hashes(const uint8_t *bytes, const int32_t *offsets, int64_t n,
+                              uint64_t *out) {
+    int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    if (i < n) out[i] = dd_hash_bytes_dev(bytes + offsets[i], (int64_t)(offsets[i + 1] - offsets[i]));
+}
+
+/* row hash over the key columns; create_hashes restatement */
+__device__ __forceinline__ uint64_t dd_row_hash(const dd_kargs &a, int64_t i) {
+    uint64_t h = 0;
+    for (int k = 0; k < a.n_keys; k++) {
+        const dd_kcol &c = a.cols[a.key_idx[k]];
+        if (c.valid && !c.valid[i]) continue;
+        uint64_t vh = dd_value_hash_dev(c, i);
+        h = h ^ (vh + 0x9e3779b97f4a7c15ULL + (h << 6) + (h >> 2));
+    }
+    return h;
+}
+
+/* ballot-multisplit: lanes with equal pid (among `act`); returns the equal-mask */
+__device__ __forceinline__ uint64_t dd_eq_mask(uint32_t pid, uint64_t act, int nbits) {
+    uint64_t eq = act;
+    for (int b = 0; b < nbits; b++) {
+        uint64_t bal = __ballot((pid >> b) & 1u);
+        eq &= ((pid >> b) & 1u) ? bal : ~bal;
+    }
+    return eq;
+}
+
+/* ---------------- K1: hash + per-chunk histogram ---------------- */
+
+__global__ __launch_bounds__(BLOCK_THREADS) void k_hash_count(
+    dd_kargs a, int64_t chunk_rows, uint32_t nparts, int nbits, uint32_t *pid_out,
+    uint32_t *counts /* [nchunks][P] */, uint32_t *bcounts /* [nvar][nchunks][P] or null */) {
+    extern __shared__ __attribute__((aligned(16))) char smem[];
+    const int wid = threadIdx.x / WAVE;
+    const int lane = threadIdx.x % WAVE;
+    const int chunk = blockIdx.x * WAVES_PER_BLOCK + wid;
+
+    /* LDS carve: hist[WPB][P] u32, then bhist[nvar][WPB][P] u32 */
+    uint32_t *hist = (uint32_t *)smem + (size_t)wid * nparts;
+    uint32_t *bhist0 = (uint32_t *)smem + (size_t)WAVES_PER_BLOCK * nparts;
+
+    for (uint32_t p = lane; p < nparts; p += WAVE) {
+        hist[p] = 0;
+        for (int v = 0; v < a.n_var; v++)
+            bhist0[((size_t)v * WAVES_PER_BLOCK + wid) * nparts + p] = 0;
+    }
+    /* single wave: LDS ops are program-ordered; no barrier needed */
+
+    const int64_t start = (int64_t)chunk * chunk_rows;
+    const int64_t end = (start + chunk_rows < a.n_rows) ? (start + chunk_rows) : a.n_rows;
+
+    for (int64_t base = start; base < end; base += WAVE) {
+        const int64_t row = base + lane;
+        const bool active = row < end;
+        uint32_t pid = 0;
+        if (active) {
+            uint64_t h = dd_row_hash(a, row);
+            pid = (uint32_t)(h % (uint64_t)nparts);
+            pid_out[row] = pid;
+        }
+        uint64_t act = __ballot(active);
+        if (active) {
+            uint64_t eq = dd_eq_mask(pid, act, nbits);
+            int leader = __ffsll((unsigned long long)eq) - 1;
+            if (lane == leader) hist[pid] += (uint32_t)__popcll((unsigned long long)eq);
+            /* var-col byte sums: LDS atomic add (order-free) */
+            for (int v = 0; v < a.n_var; v++) {
+                const dd_kcol &c = a.cols[a.var_idx[v]];
+                uint32_t len = (uint32_t)(c.offsets[row + 1] - c.offsets[row]);
+                atomicAdd(&bhist0[((size_t)v * WAVES_PER_BLOCK + wid) * nparts + pid], len);
+            }
+        }
+    }
+
+    for (uint32_t p = lane; p < nparts; p += WAVE) {
+        counts[(size_t)chunk * nparts + p] = hist[p];
+        for (int v = 0; v < a.n_var; v++)
+            bcounts[((size_t)v * gridDim.x * WAVES_PER_BLOCK + chunk) * nparts + p] =
+                bhist0[((size_t)v * WAVES_PER_BLOCK + wid) * nparts + p];
+    }
+}
+
+/* ---------------- K2: column-major exclusive scan of counts ----------------
+ * counts is [nchunks][P] row-major; we scan down each partition column. 3 coalesced passes:
+ *   K2a: each (range r, lane-partition p) sums its chunk range        -> partials[R][P]
+ *   K2b: one block: scan partials per partition; scan totals across P -> part_offsets[P+1]
+ *   K2c: each (r, p) re-walks, rewriting counts to global-exclusive (within partition)
+ * For byte matrices the same kernels run with their own buffers. DD_SCAN_RANGES ranges. */
+
+__global__ void k_scan_partial(const uint32_t *counts, int64_t nchunks, uint32_t nparts,
+                               int nranges, uint32_t *partials /* [nranges][P] */) {
+    const int64_t tid = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    const int64_t total = (int64_t)nranges * nparts;
+    if (tid >= total) return;
+    const int r = (int)(tid / nparts);
+    const uint32_t p = (uint32_t)(tid % nparts);
+    const int64_t c0 = nchunks * r / nranges, c1 = nchunks * (r + 1) / nranges;
+    uint32_t s = 0;
+    for (int64_t c = c0; c < c1; c++) s += counts[c * nparts + p];
+    partials[(size_t)r * nparts + p] = s;
+}
+
+/* one block; scans partials in place (exclusive within partition), then exclusive-scans the
+ * per-partition totals across partitions into part_offsets[P+1] (u64). */
+__global__ void k_scan_combine(uint32_t *partials, int nranges, uint32_t nparts,
+                               uint64_t *part_offsets) {
+    __shared__ uint64_t totals[DD_MAX_P];
+    for (uint32_t p = threadIdx.x; p < nparts; p += blockDim.x) {
+        uint32_t run = 0;
+        for (int r = 0; r < nranges; r++) {
+            uint32_t v = partials[(size_t)r * nparts + p];
+            partials[(size_t)r * nparts + p] = run;
+            run += v;
+        }
+        totals[p] = run;
+    }
+    __syncthreads();
+    if (threadIdx.x == 0) { /* P <= 2048: serial scan is microseconds */
+        uint64_t run = 0;
+        for (uint32_t p = 0; p < nparts; p++) {
+            uint64_t v = totals[p];
+            part_offsets[p] = run;
+            run += v;
+        }
+        part_offsets[nparts] = run;
+    }
+}
+
+__global__ void k_scan_rewrite(uint32_t *counts, int64_t nchunks, uint32_t nparts,
+                               int nranges, const uint32_t *partials) {
+    const int64_t tid = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    const int64_t total = (int64_t)nranges * nparts;
+    if (tid >= total) return;
+    const int r = (int)(tid / nparts);
+    const uint32_t p = (uint32_t)(tid % nparts);
+    const int64_t c0 = nchunks * r / nranges, c1 = nchunks * (r + 1) / nranges;
+    uint32_t run = partials[(size_t)r * nparts + p];
+    for (int64_t c = c0; c < c1; c++) {
+        uint32_t v = counts[c * nparts + p];
+        counts[c * nparts + p] = run;
+        run += v;
+    }
+}
+
+/* ---------------- K3: stable scatter ---------------- */
+
+__global__ __launch_bounds__(BLOCK_THREADS) void k_scatter(
+    dd_kargs a, int64_t chunk_rows, uint32_t nparts, int nbits, const uint32_t *pid_in,
+    const uint32_t *chunk_off /* [nchunks][P] exclusive within partition */,
+    const uint64_t *part_offsets /* [P+1] */,
+    const uint32_t *chunk_boff /* [nvar][nchunks][P] or null */,
+    const uint64_t *part_boffsets /* [nvar][P+1] or null */) {
+    extern __shared__ __attribute__((aligned(16))) char smem[];
+    const int wid = threadIdx.x / WAVE;
+    const int lane = threadIdx.x % WAVE;
+    const int chunk = blockIdx.x * WAVES_PER_BLOCK + wid;
+    const int64_t nchunksx = (int64_t)gridDim.x * WAVES_PER_BLOCK;
+
+    /* LDS carve (per wave): row_base u64[P]; per var: byte_base u64[P], brun u32[P];
+     * then lens u32[WPB][64] staging for intra-group byte prefix */
+    char *ws = smem;
+    uint64_t *row_base = (uint64_t *)ws + (size_t)wid * nparts;
+    ws += sizeof(uint64_t) * WAVES_PER_BLOCK * nparts;
+    uint64_t *byte_base0 = (uint64_t *)ws; /* [nvar][WPB][P] */
+    ws += sizeof(uint64_t) * (size_t)a.n_var * WAVES_PER_BLOCK * nparts;
+    uint32_t *lens_stage = (uint32_t *)ws + (size_t)wid * WAVE; /* [WPB][64] */
+
+    const int64_t start = (int64_t)chunk * chunk_rows;
+    const int64_t end = (start + chunk_rows < a.n_rows) ? (start + chunk_rows) : a.n_rows;
+
+    /* stage bases: global row slot base per partition for this chunk */
+    for (uint32_t p = lane; p < nparts; p += WAVE) {
+        row_base[p] = part_offsets[p] + chunk_off[(size_t)chunk * nparts + p];
+        for (int v = 0; v < a.n_var; v++)
+            byte_base0[((size_t)v * WAVES_PER_BLOCK + wid) * nparts + p] =
+                part_boffsets[(size_t)v * (nparts + 1) + p] +
+                chunk_boff[((size_t)v * nchunksx + chunk) * nparts + p];
+    }
+
+    for (int64_t base = start; base < end; base += WAVE) {
+        const int64_t row = base + lane;
+        const bool active = row < end;
+        const uint32_t pid = active ? pid_in[row] : 0u;
+        const uint64_t act = __ballot(active);
+        const uint64_t eq = active ? dd_eq_mask(pid, act, nbits) : 0;
+        const uint64_t lt = ((uint64_t)1 << lane) - 1;
+        const int rank = active ? (int)__popcll((unsigned long long)(eq & lt)) : 0;
+        const int leader = active ? __ffsll((unsigned long long)eq) - 1 : 0;
+        const int gsize = active ? (int)__popcll((unsigned long long)eq) : 0;
+
+        uint64_t dst = 0;
+        if (active) {
+            /* leader reads + advances the base; others get it via shuffle */
+            uint64_t b = 0;
+            if (lane == leader) {
+                b = row_base[pid];
+                row_base[pid] = b + (uint64_t)gsize;
+            }
+            b = (uint64_t)__shfl((long long)b, leader);
+            dst = b + (uint64_t)rank;
+
+            /* fixed-width columns + validity */
+            for (int c = 0; c < a.n_cols; c++) {
+                const dd_kcol &col = a.cols[c];
+                switch (col.elem) {
+                case 1:
+                    ((uint8_t *)col.out_data)[dst] = ((const uint8_t *)col.data)[row];
+                    break;
+                case 2:
+                    ((uint16_t *)col.out_data)[dst] = ((const uint16_t *)col.data)[row];
+                    break;
+                case 4:
+                    ((uint32_t *)col.out_data)[dst] = ((const uint32_t *)col.data)[row];
+                    break;
+                case 8:
+                    ((uint64_t *)col.out_data)[dst] = ((const uint64_t *)col.data)[row];
+                    break;
+                default:
+                    break; /* var col: handled below */
+                }
+                if (col.valid) col.out_valid[dst] = col.valid[row];
+            }
+        }
+
+        /* var columns: intra-group byte prefix via LDS-staged lengths */
+        for (int v = 0; v < a.n_var; v++) {
+            const dd_kcol &col = a.cols[a.var_idx[v]];
+            uint32_t len = 0;
+            int32_t o0 = 0;
+            if (active) {
+                o0 = col.offsets[row];
+                len = (uint32_t)(col.offsets[row + 1] - o0);
+            }
+            lens_stage[lane] = len; /* single wave: program-ordered LDS */
+            uint32_t bpre = 0, bsum = 0;
+            if (active) {
+                uint64_t m = eq;
+                while (m) {
+                    int j = __ffsll((unsigned long long)m) - 1;
+                    uint32_t lj = lens_stage[j];
+                    if (j < lane) bpre += lj;
+                    bsum += lj;
+                    m &= m - 1;
+                }
+                uint64_t *bb = &byte_base0[((size_t)v * WAVES_PER_BLOCK + wid) * nparts + pid];
+                uint64_t bbase = 0;
+                if (lane == leader) {
+                    bbase = *bb;
+                    *bb = bbase + bsum;
+                }
+                bbase = (uint64_t)__shfl((long long)bbase, leader);
+                uint64_t bdst = bbase + bpre;
+                col.out_lengths[dst] = len;
+                const uint8_t *src = (const uint8_t *)col.data + o0;
+                uint8_t *d = (uint8_t *)col.out_data + bdst;
+                for (uint32_t b = 0; b < len; b++) d[b] = src[b];
+            }
+        }
+    }
+}
+
+/* ---------------- launchers (called from dd_host.cpp) ---------------- */
+
+extern "C" {
+
+hipError_t dd_launch_dict_hashes(const uint8_t *bytes, const int32_t *offsets, int64_t n,
+                                 uint64_t *out, hipStream_t s) {
+    int threads = 256;
+    int blocks = (int)((n + threads - 1) / threads);
+    if (blocks == 0) blocks = 1;
+    hipLaunchKernelGGL(k_dict_hashes, dim3(blocks), dim3(threads), 0, s, bytes, offsets, n, out);
+    return hipGetLastError();
+}
+
+hipError_t dd_launch_hash_count(const dd_kargs *a, int64_t nchunks, int64_t chunk_rows,
+                                uint32_t nparts, int nbits, uint32_t *pid_out,
+                                uint32_t *counts, uint32_t *bcounts, size_t lds_bytes,
+                                hipStream_t s) {
+    dim3 grid((unsigned)(nchunks / WAVES_PER_BLOCK));
+    if (lds_bytes > 65536) {
+        hipError_t e = hipFuncSetAttribute((const void *)k_hash_count,
+                                           hipFuncAttributeMaxDynamicSharedMemorySize,
+                                           (int)lds_bytes);
+        if (e != hipSuccess) return e;
+    }
+    hipLaunchKernelGGL(k_hash_count, grid, dim3(BLOCK_THREADS), lds_bytes, s, *a, chunk_rows,
+                       nparts, nbits, pid_out, counts, bcounts);
+    return hipGetLastError();
+}
+
+hipError_t dd_launch_scan(uint32_t *counts, int64_t nchunks, uint32_t nparts, int nranges,
+                          uint32_t *partials, uint64_t *part_offsets, hipStream_t s) {
+    int threads = 256;
+    int64_t total = (int64_t)nranges * nparts;
+    int blocks = (int)((total + threads - 1) / threads);
+    hipLaunchKernelGGL(k_scan_partial, dim3(blocks), dim3(threads), 0, s, counts, nchunks,
+                       nparts, nranges, partials);
+    hipLaunchKernelGGL(k_scan_combine, dim3(1), dim3(256), 0, s, partials, nranges, nparts,
+                       part_offsets);
+    hipLaunchKernelGGL(k_scan_rewrite, dim3(blocks), dim3(threads), 0, s, counts, nchunks,
+                       nparts, nranges, partials);
+    return hipGetLastError();
+}
+
+hipError_t dd_launch_scatter(const dd_kargs *a, int64_t nchunks, int64_t chunk_rows,
+                             uint32_t nparts, int nbits, const uint32_t *pid_in,
+                             const uint32_t *chunk_off, const uint64_t *part_offsets,
+                             const uint32_t *chunk_boff, const uint64_t *part_boffsets,
+                             size_t lds_bytes, hipStream_t s) {
+    dim3 grid((unsigned)(nchunks / WAVES_PER_BLOCK));
+    if (lds_bytes > 65536) {
+        hipError_t e = hipFuncSetAttribute((const void *)k_scatter,
+                                           hipFuncAttributeMaxDynamicSharedMemorySize,
+                                           (int)lds_bytes);
+        if (e != hipSuccess) return e;
+    }
+    hipLaunchKernelGGL(k_scatter, grid, dim3(BLOCK_THREADS), lds_bytes, s, *a, chunk_rows,
+                       nparts, nbits, pid_in, chunk_off, part_offsets, chunk_boff,
+                       part_boffsets);
+    return hipGetLastError();
+}
+
+} /* extern "C" */

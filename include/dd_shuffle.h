@@ -1,0 +1,184 @@
+/* include/dd_shuffle.h — C ABI of the MI355X-native shuffle/exchange path.
+ *
+ * This is the drop-in boundary: the exports mirror, 1:1, the reference's
+ * transport/operator seam for the hash-shuffle hot path so that a one-file Rust shim
+ * implementing `WorkerChannel` (reference: src/protocol/worker_channel.rs:19-46, resolved via
+ * src/protocol/channel_resolver.rs:27-41) can delegate to these functions wherever cargo
+ * exists. The reference-side binding a maintainer would add is shown in INTEGRATION.md.
+ *
+ * No torch types anywhere; plain pointers and sizes. All device pointers are HIP device
+ * memory on the current device. All entry points return dd_status; dd_last_error() gives a
+ * human-readable message (mirrors the reference's DataFusionError <-> tonic Status mapping,
+ * src/protocol/grpc/errors/).
+ */
+
+#ifndef DD_SHUFFLE_H
+#define DD_SHUFFLE_H
+
+#include <stddef.h>
+#include <stdint.h>
+
+#ifdef __cplusplus
+extern "C" {
+#endif
+
+/* ---------------- status / errors ---------------- */
+
+typedef enum dd_status {
+    DD_OK = 0,
+    DD_ERR_INVALID = 1,     /* bad arguments / unsupported shape */
+    DD_ERR_NO_DEVICE = 2,   /* no HIP device — the product path FAILS here, no CPU fallback */
+    DD_ERR_HIP = 3,         /* HIP runtime error */
+    DD_ERR_RCCL = 4,        /* RCCL error */
+    DD_ERR_NOT_FOUND = 5,   /* unknown TaskKey (mirrors worker "plan not found" timeout,
+                               src/worker/impl_execute_task.rs:29-34) */
+    DD_ERR_UNSUPPORTED = 6  /* dtype/partition-count combination outside round-1 coverage */
+} dd_status;
+
+const char *dd_last_error(void);
+int dd_device_count(void); /* number of visible HIP devices (0 on a GPU-less box) */
+const char *dd_version(void);
+
+/* ---------------- data model ----------------
+ * Device-resident Arrow-layout columns. Validity is UNPACKED u8 (1=valid).
+ * dtype codes shared with oracle/dd_oracle.c. */
+
+typedef enum dd_dtype {
+    DD_DT_U8 = 1,
+    DD_DT_I16 = 2,
+    DD_DT_I32 = 3,
+    DD_DT_I64 = 4,
+    DD_DT_F32 = 5,
+    DD_DT_F64 = 6,
+    DD_DT_BOOL = 7,  /* unpacked u8 0/1 */
+    DD_DT_UTF8 = 8,  /* i32 offsets[n+1] + byte buffer */
+    DD_DT_DICT32 = 9 /* i32 indices; dictionary values as utf8 (offsets+bytes) */
+} dd_dtype;
+
+#define DD_MAX_COLS 24
+#define DD_MAX_KEYS 8
+#define DD_MAX_PARTITIONS 2048u /* round-1 cap (LDS budget), DESIGN.md §5 */
+
+typedef struct dd_col_desc {
+    int32_t dtype;            /* dd_dtype */
+    const void *data;         /* device: values, or utf8 bytes */
+    const uint8_t *validity;  /* device, unpacked u8, NULL => all valid */
+    const int32_t *offsets;   /* device, UTF8 only: offsets[n_rows+1] */
+    int64_t data_len;         /* UTF8: byte length of `data`; else 0 */
+    const void *dict_bytes;   /* DICT32: device value bytes */
+    const int32_t *dict_offsets; /* DICT32: device offsets[dict_n+1] */
+    int64_t dict_n;           /* DICT32: number of dictionary values */
+} dd_col_desc;
+
+typedef struct dd_batch_desc {
+    int64_t n_rows;
+    int32_t n_cols;
+    dd_col_desc cols[DD_MAX_COLS];
+} dd_batch_desc;
+
+/* ---------------- hash-repartition (the producer head) ----------------
+ * Replaces the `RepartitionExec(Hash(keys, P_total))` head the reference inserts on the
+ * producer: ProducerHead::insert (src/distributed_planner/network_boundary.rs:86-106),
+ * required input shape at src/execution_plans/network_shuffle.rs:121-127, lazily re-created
+ * per task at src/worker/task_data.rs:104-116. Semantics: DESIGN.md §3 (normative; stable).
+ *
+ * Two-phase (prepare/execute) so repeated executions reuse workspace + output buffers —
+ * mirroring the worker's lazy plan-head caching. All output buffers are partition-major
+ * contiguous. */
+
+typedef struct dd_partitioner dd_partitioner; /* opaque */
+
+dd_status dd_partitioner_create(const dd_batch_desc *batch, const int32_t *key_cols,
+                                int32_t n_keys, uint32_t n_partitions,
+                                dd_partitioner **out);
+/* Runs K1 (hash+count), K2 (scan), K3 (stable scatter) on `stream` (a hipStream_t, or NULL
+ * for the default stream). Asynchronous; results valid after stream sync. */
+dd_status dd_partitioner_run(dd_partitioner *p, void *stream);
+void dd_partitioner_destroy(dd_partitioner *p);
+
+/* result accessors (pointers are device memory owned by the partitioner) */
+const uint32_t *dd_partitioner_pids(const dd_partitioner *p);      /* u32[n_rows] */
+const void *dd_partitioner_col_data(const dd_partitioner *p, int32_t col); /* partition-major */
+const uint8_t *dd_partitioner_col_validity(const dd_partitioner *p, int32_t col);
+const uint32_t *dd_partitioner_col_lengths(const dd_partitioner *p, int32_t col); /* utf8 */
+/* copies part_row_offsets[P+1] (rows) to host */
+dd_status dd_partitioner_row_offsets(const dd_partitioner *p, int64_t *host_out);
+/* copies per-var-col part_byte_offsets[P+1] to host */
+dd_status dd_partitioner_byte_offsets(const dd_partitioner *p, int32_t col, int64_t *host_out);
+/* per-kernel last-run durations in ms, measured with hipEvents on the run stream:
+ * [0]=K1 hash+count, [1]=K2 scan, [2]=K3 scatter. Valid after a synced run. */
+dd_status dd_partitioner_kernel_ms(const dd_partitioner *p, float out_ms[3]);
+
+/* ---------------- exchange (the transport data plane) ----------------
+ * Replaces the Arrow-Flight-over-gRPC data plane (client demux src/protocol/grpc/
+ * worker_client.rs:89-320; server encode+tagging src/protocol/grpc/worker_service.rs:133-184,
+ * 363-433) with an RCCL all-to-all-v over xGMI: one rank per GPU, rank = task
+ * (TaskKey.task_number, src/stage.rs:134-139). Consumer rank r owns partition window
+ * [P*r, P*(r+1)) of P_total = P*nranks (src/execution_plans/network_shuffle.rs:232-244,
+ * scale_partitioning src/execution_plans/common.rs:18-30). Received data is concatenated
+ * in producer-rank order (deterministic refinement of the reference's select_all merge). */
+
+typedef struct dd_comm dd_comm; /* opaque: RCCL communicator + streams */
+
+#define DD_UNIQUE_ID_BYTES 128 /* == NCCL_UNIQUE_ID_BYTES */
+dd_status dd_comm_unique_id(void *bytes128); /* rank 0 calls; share out-of-band */
+dd_status dd_comm_init(const void *bytes128, int rank, int nranks, dd_comm **out);
+void dd_comm_destroy(dd_comm *c);
+
+typedef struct dd_exchanged dd_exchanged; /* opaque: received window buffers */
+
+/* Exchange the partitioned result: every rank sends partition window [P*j, P*(j+1)) of its
+ * local result to rank j and receives its own window from every rank. P (= partitions per
+ * consumer) is inferred: partitioner P_total must equal P*nranks. Synchronous on `stream`. */
+dd_status dd_exchange_run(dd_comm *c, const dd_partitioner *p, void *stream,
+                          dd_exchanged **out);
+void dd_exchanged_destroy(dd_exchanged *e);
+
+/* accessors: my window, producer-major then partition-major per producer */
+int64_t dd_exchanged_total_rows(const dd_exchanged *e);
+const void *dd_exchanged_col_data(const dd_exchanged *e, int32_t col);
+const uint8_t *dd_exchanged_col_validity(const dd_exchanged *e, int32_t col);
+const uint32_t *dd_exchanged_col_lengths(const dd_exchanged *e, int32_t col);
+/* row counts per (producer, local partition): host_out[nranks*P] */
+dd_status dd_exchanged_row_counts(const dd_exchanged *e, int64_t *host_out);
+/* bytes per (producer, local partition) for a var col: host_out[nranks*P] */
+dd_status dd_exchanged_byte_counts(const dd_exchanged *e, int32_t col, int64_t *host_out);
+/* last exchange wall time (ms, hipEvent on the exchange stream) and payload bytes sent by
+ * this rank to OTHER ranks (xGMI egress) */
+dd_status dd_exchanged_stats(const dd_exchanged *e, float *ms, int64_t *egress_bytes);
+
+/* ---------------- task cache (worker execute path) ----------------
+ * Mirrors SetPlanRequest / ExecuteTaskRequest (src/protocol/worker_channel.rs:74-93,163-176)
+ * and the worker's TaskData cache (src/worker/task_data.rs:16-29,104-116;
+ * src/worker/worker_service.rs:12,31 — the moka TTI cache becomes an explicit
+ * set/execute/drop lifecycle). The "plan" payload here is the shuffle-path plan descriptor:
+ * a device batch + key columns + partitioning, i.e. exactly the producer head the reference
+ * ships for this path. */
+
+typedef struct dd_task_key {
+    uint64_t query_id_hi, query_id_lo; /* Uuid (src/protocol/worker_channel.rs:57-65) */
+    uint64_t stage_id;
+    uint64_t task_number;
+} dd_task_key;
+
+dd_status dd_set_plan(const dd_task_key *key, const dd_batch_desc *batch,
+                      const int32_t *key_cols, int32_t n_keys, uint32_t n_partitions);
+/* Executes (or reuses) the cached task's partitioner and returns it; partition range
+ * [part_lo, part_hi) mirrors ExecuteTaskRequest.target_partition_start/end. The returned
+ * partitioner is owned by the cache; do not destroy. */
+dd_status dd_execute_task(const dd_task_key *key, uint32_t part_lo, uint32_t part_hi,
+                          void *stream, dd_partitioner **out);
+dd_status dd_drop_task(const dd_task_key *key); /* task cleanup (stateful_data_cleanup) */
+
+/* ---------------- device helpers (harness convenience; not part of the seam) ------- */
+dd_status dd_dev_alloc(int64_t bytes, void **out);
+dd_status dd_dev_free(void *p);
+dd_status dd_memcpy_h2d(void *dst, const void *src, int64_t bytes);
+dd_status dd_memcpy_d2h(void *dst, const void *src, int64_t bytes);
+dd_status dd_device_sync(void);
+
+#ifdef __cplusplus
+}
+#endif
+
+#endif /* DD_SHUFFLE_H */

@@ -1,0 +1,55 @@
+"""RCCL exchange on the GPU box (marked gpu). gpurun provides one GPU, so the collective is
+exercised at nranks=1 (self-exchange through RCCL send/recv): the received window must equal
+the partitioner's full output. N>1 placement logic is pinned by tests/test_exchange_gloo.py's
+model and by the driver's multi-GPU scale run."""
+
+import numpy as np
+import pytest
+
+import oracle
+from datafusion_distributed_amd import api
+
+pytestmark = pytest.mark.gpu
+
+
+def test_self_exchange_roundtrip():
+    rng = np.random.default_rng(31)
+    n, P = 100000, 16
+    rows = [b"s" * int(l) for l in rng.integers(0, 20, n)]
+    off = np.zeros(n + 1, dtype=np.int32)
+    off[1:] = np.cumsum([len(r) for r in rows])
+    cols = [
+        {"dtype": "i64", "data": rng.integers(0, 10**12, n, dtype=np.int64), "valid": None},
+        {"dtype": "f64", "data": rng.normal(size=n), "valid": None},
+        {"dtype": "utf8", "data": np.frombuffer(b"".join(rows), dtype=np.uint8),
+         "offsets": off, "valid": (rng.random(n) > 0.1).astype(np.uint8)},
+    ]
+    batch = api.DeviceBatch(cols)
+    part = api.Partitioner(batch, [0], P)
+    part.run()
+    part.sync()
+    ref = oracle.repartition(cols, [0], P)
+
+    comm = api.Comm(api.Comm.unique_id(), 0, 1)
+    ex = comm.exchange(part)
+    assert ex.total_rows == n
+    rc = ex.row_counts()
+    assert rc.shape == (1, P)
+    assert (rc[0] == ref["part_offsets"][1:] - ref["part_offsets"][:-1]).all()
+
+    got0 = ex.col_data(0)
+    assert (got0["data"] == ref["cols"][0]["data"]).all()
+    got1 = ex.col_data(1)
+    assert np.array_equal(got1["data"], ref["cols"][1]["data"], equal_nan=True)
+    got2 = ex.col_data(2)
+    assert (got2["lengths"] == ref["cols"][2]["lengths"]).all()
+    assert got2["data"].tobytes() == ref["cols"][2]["data"].tobytes()
+    v = ex.col_validity(2)
+    assert (v == ref["cols"][2]["valid"]).all()
+    ms, egress = ex.stats()
+    assert ms >= 0 and egress == 0  # single rank: no xGMI egress
+
+    ex.destroy()
+    comm.destroy()
+    part.destroy()
+    batch.free()
