@@ -1,0 +1,228 @@
+"""bench.py — measures the BASELINE.json metric: TPC-H SF10 shuffle GB/s + rows/s on
+1/2/4/8 MI355X, with the HBM roofline of the dominant kernel (K3 scatter).
+
+Workload: synthetic TPC-H SF10 lineitem shuffle (q3/q5 shape) —
+59,986,052 rows per rank (weak scaling), hash key l_orderkey (int64), payload = q3
+projection (l_orderkey i64, l_extendedprice f64, l_discount f64, l_shipdate date32) =
+28 B/row. One "step" = one full hash-repartition pass over the HBM-resident input
+(+ the RCCL all-to-all-v exchange at N>1). P_total = 128 * n_gpus.
+
+value = whole-job GB/s of input bytes (rows x 28 x n_gpus / time). dtype "int64" is the
+arithmetic type of the hash/scatter path, not a precision claim. Inputs are resident in
+HBM when the timed region starts; the PCIe-inclusive staging rate is reported in
+config.h2d_GBps (never as value).
+
+cpu_baseline: the C oracle (oracle/dd_oracle.c — same algorithm; kind "port"; the
+reference itself is Rust and unbuildable here) timed on this box's host
+cores on a bounded sample. Rank 0, N=1 only.
+"""
+
+import argparse
+import json
+import os
+import sys
+import time
+
+import numpy as np
+
+REPO = os.path.dirname(os.path.abspath(__file__))
+sys.path.insert(0, REPO)
+
+SF10_ROWS = 59_986_052
+ROW_BYTES = 28  # i64 + f64 + f64 + i32
+PARTS_PER_RANK = 128
+# K3 algorithmic bytes/row: read pid (4) + read cols (28) + write cols (28)  
+K3_BYTES_PER_ROW = 60
+HBM_PEAK = 8.0e12
+
+
+def make_lineitem(rows, seed):
+    rng = np.random.default_rng(seed)
+    return [
+        {"dtype": "i64", "data": rng.integers(1, 60_000_000, rows, dtype=np.int64),
+         "valid": None},
+        {"dtype": "f64", "data": rng.uniform(900.0, 105000.0, rows), "valid": None},
+        {"dtype": "f64", "data": rng.uniform(0.0, 0.1, rows), "valid": None},
+        {"dtype": "i32", "data": rng.integers(8000, 11000, rows, dtype=np.int64)
+         .astype(np.int32), "valid": None},
+    ]
+
+
+def cpu_baseline_leg(rows_sample, key, cols_np, min_seconds=8.0, max_seconds=30.0):
+    """Time the C oracle on a bounded sample of the same workload (kind: port)."""
+    import oracle
+
+    sample = [{k: (v[:rows_sample] if isinstance(v, np.ndarray) else v) for k, v in c.items()}
+              for c in cols_np]
+    reps = 0
+    t0 = time.perf_counter()
+    while True:
+        oracle.repartition(sample, key, PARTS_PER_RANK)
+        reps += 1
+        el = time.perf_counter() - t0
+        if el >= min_seconds or el >= max_seconds:
+            break
+    el = time.perf_counter() - t0
+    gbps = rows_sample * ROW_BYTES * reps / el / 1e9
+    return {
+        "value": round(gbps, 3),
+        "unit": "GB/s",
+        "cores": oracle.lib().dd_oracle_num_threads(),
+        "kind": "port",
+        "sample": f"{rows_sample} rows x {reps} reps of the same hash-repartition "
+                  f"(28 B/row, P={PARTS_PER_RANK}), {el:.1f}s on host cores",
+    }
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=10)
+    ap.add_argument("--warmup", type=int, default=3)
+    ap.add_argument("--rows", type=int, default=SF10_ROWS, help="rows per rank")
+    ap.add_argument("--parts-per-rank", type=int, default=PARTS_PER_RANK)
+    ap.add_argument("--traffic-bytes", type=float, default=None,
+                    help="measured HBM bytes per K3 launch from a separate rocprofv3 "
+                         "--pmc run (profiles/); null if not provided")
+    ap.add_argument("--skip-cpu-baseline", action="store_true")
+    args = ap.parse_args()
+
+    rank = int(os.environ.get("RANK", "0"))
+    world = int(os.environ.get("WORLD_SIZE", str(args.gpus)))
+    local_rank = int(os.environ.get("LOCAL_RANK", "0"))
+
+    import torch
+
+    if world > 1:
+        import torch.distributed as dist
+
+        os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+        dist.init_process_group("nccl", rank=rank, world_size=world)
+        torch.cuda.set_device(local_rank)
+    elif torch.cuda.is_available():
+        torch.cuda.set_device(0)
+
+    from datafusion_distributed_amd import api
+    from datafusion_distributed_amd.exchange import create_comm
+
+    if api.device_count() == 0:
+        print(json.dumps({"error": "no HIP device"}))
+        sys.exit(1)
+
+    p_total = args.parts_per_rank * world
+    cols_np = make_lineitem(args.rows, seed=42 + rank)
+
+    th2d0 = time.perf_counter()
+    batch = api.DeviceBatch(cols_np)
+    api.lib().dd_device_sync()
+    th2d = time.perf_counter() - th2d0
+    h2d_gbps = args.rows * ROW_BYTES / th2d / 1e9
+
+    part = api.Partitioner(batch, [0], p_total)
+    comm = create_comm(rank, world) if world > 1 else None
+
+    def step():
+        part.run()
+        if comm is not None:
+            ex = comm.exchange(part)
+            ex.destroy()
+        else:
+            part.sync()
+
+    def barrier_sync():
+        if world > 1:
+            import torch.distributed as dist
+
+            torch.cuda.synchronize()
+            dist.barrier()
+        torch.cuda.synchronize() if torch.cuda.is_available() else api.lib().dd_device_sync()
+
+    for _ in range(args.warmup):
+        step()
+    barrier_sync()
+
+    k3_ms_total = 0.0
+    t0 = time.perf_counter()
+    for _ in range(args.steps):
+        step()
+        k3_ms_total += part.kernel_ms()[2]
+    barrier_sync()
+    elapsed = time.perf_counter() - t0
+
+    if world > 1:
+        import torch.distributed as dist
+
+        t = torch.tensor([elapsed], dtype=torch.float64,
+                         device="cuda" if torch.cuda.is_available() else "cpu")
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed = float(t.item())
+
+    total_bytes = args.rows * ROW_BYTES * world * args.steps
+    value = total_bytes / elapsed / 1e9
+    rows_per_s = args.rows * world * args.steps / elapsed
+
+    k1_ms, k2_ms, k3_ms = part.kernel_ms()
+    k3_mean_ms = k3_ms_total / args.steps
+    achieved = args.rows * K3_BYTES_PER_ROW / (k3_mean_ms / 1e3)
+    roofline = {
+        "bound": "hbm",
+        "achieved": round(achieved / 1e9, 1),
+        "peak": round(HBM_PEAK / 1e9, 1),
+        "unit": "GB/s",
+        "frac": round(achieved / HBM_PEAK, 4),
+        "traffic": args.traffic_bytes,
+    }
+
+    cpu_baseline = None
+    if rank == 0 and world == 1 and not args.skip_cpu_baseline:
+        cpu_baseline = cpu_baseline_leg(min(args.rows, 8_000_000), [0], cols_np)
+
+    config = {
+        "workload": "tpch_sf10_lineitem_shuffle",
+        "rows_per_rank": args.rows,
+        "row_bytes": ROW_BYTES,
+        "p_total": p_total,
+        "parts_per_rank": args.parts_per_rank,
+        "key": "l_orderkey(i64)",
+        "rows_per_s": round(rows_per_s, 0),
+        "kernel_ms_last": {"k1_hash": round(k1_ms, 3), "k2_scan": round(k2_ms, 3),
+                           "k3_scatter": round(k3_ms, 3)},
+        "h2d_GBps": round(h2d_gbps, 2),
+    }
+    if comm is not None:
+        ex = comm.exchange(part)
+        ms, egress = ex.stats()
+        config["xgmi_egress_GBps"] = round(egress / (ms / 1e3) / 1e9, 2) if ms > 0 else None
+        ex.destroy()
+
+    line = {
+        "metric": "tpch_sf10_shuffle_GBps",
+        "value": round(value, 3),
+        "unit": "GB/s",
+        "n_gpus": world,
+        "steps": args.steps,
+        "warmup": args.warmup,
+        "ms_per_step": round(elapsed / args.steps * 1e3, 3),
+        "higher_is_better": True,
+        "scaling": "weak",
+        "vs_baseline": None,  # no published number for this metric
+        "dtype": "int64",
+        "data": "synthetic",
+        "config": config,
+        "roofline": roofline,
+        "cpu_baseline": cpu_baseline,
+    }
+    if rank == 0:
+        print(json.dumps(line))
+
+    part.destroy()
+    batch.free()
+    if comm is not None:
+        comm.destroy()
+        import torch.distributed as dist
+
+        dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()
