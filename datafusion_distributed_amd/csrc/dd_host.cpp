@@ -104,6 +104,8 @@ struct dd_partitioner {
     int nbits = 0;
     int64_t nchunks = 0, chunk_rows = 0;
     size_t lds_k1 = 0, lds_k3 = 0;
+    bool staged = false; /* v2 path: block-tile + LDS-staged scatter (fixed-width only) */
+    int gmax = 0;        /* v2 rows-per-round = gmax * 256 */
 
     /* device buffers (owned) */
     uint32_t *pid = nullptr;
@@ -162,14 +164,6 @@ extern "C" dd_status dd_partitioner_create(const dd_batch_desc *batch, const int
     while ((1u << p->nbits) < n_partitions) p->nbits++;
 
     const int64_t n = batch->n_rows;
-    int64_t nchunks = (n + 4095) / 4096;
-    if (nchunks < 4) nchunks = 4;
-    if (nchunks > 8192) nchunks = 8192;
-    nchunks = (nchunks + 3) & ~3LL;
-    p->nchunks = nchunks;
-    p->chunk_rows = (n + nchunks - 1) / nchunks;
-    if (p->chunk_rows < 1) p->chunk_rows = 1;
-
     dd_kargs &ka = p->ka;
     memset(&ka, 0, sizeof(ka));
     ka.n_rows = n;
@@ -205,10 +199,50 @@ extern "C" dd_status dd_partitioner_create(const dd_batch_desc *batch, const int
 
     const int64_t P = n_partitions;
     const int nvar = ka.n_var;
-    p->lds_k1 = (size_t)WAVES_PER_BLOCK_H * P * 4 * (1 + nvar);
-    p->lds_k3 = (size_t)WAVES_PER_BLOCK_H * P * 8 * (1 + nvar) + WAVES_PER_BLOCK_H * 64 * 4;
-    if (p->lds_k1 > 163840 || p->lds_k3 > 163840)
-        return fail(DD_ERR_UNSUPPORTED, "partition count x var columns exceeds LDS budget");
+
+    /* v2 (staged) eligibility: fixed-width batch whose staging fits LDS */
+    if (nvar == 0) {
+        size_t row_stage = 4; /* dstg */
+        int nvalid = 0;
+        for (int c = 0; c < batch->n_cols; c++) {
+            row_stage += fixed_elem_size(batch->cols[c].dtype);
+            if (batch->cols[c].validity) nvalid++;
+        }
+        row_stage += nvalid;
+        const size_t part_lds = (size_t)P * (8 + 4 * WAVES_PER_BLOCK_H + 4 + 4) + 256 * 4;
+        for (int g = 8; g >= 2; g /= 2) {
+            size_t lds = part_lds + (size_t)g * 256 * row_stage;
+            if (lds <= 163840) {
+                p->staged = true;
+                p->gmax = g;
+                p->lds_k3 = lds;
+                break;
+            }
+        }
+    }
+    if (p->staged) {
+        p->lds_k1 = (size_t)WAVES_PER_BLOCK_H * P * 4;
+        int64_t nblocks = (n + 16383) / 16384;
+        if (nblocks < 8) nblocks = 8;
+        if (nblocks > 2048) nblocks = 2048;
+        p->nchunks = nblocks;
+        p->chunk_rows = (n + nblocks - 1) / nblocks;
+        if (p->chunk_rows < 1) p->chunk_rows = 1;
+    } else {
+        p->lds_k1 = (size_t)WAVES_PER_BLOCK_H * P * 4 * (1 + nvar);
+        p->lds_k3 =
+            (size_t)WAVES_PER_BLOCK_H * P * 8 * (1 + nvar) + WAVES_PER_BLOCK_H * 64 * 4;
+        if (p->lds_k1 > 163840 || p->lds_k3 > 163840)
+            return fail(DD_ERR_UNSUPPORTED, "partition count x var columns exceeds LDS budget");
+        int64_t nchunks = (n + 4095) / 4096;
+        if (nchunks < 4) nchunks = 4;
+        if (nchunks > 8192) nchunks = 8192;
+        nchunks = (nchunks + 3) & ~3LL;
+        p->nchunks = nchunks;
+        p->chunk_rows = (n + nchunks - 1) / nchunks;
+        if (p->chunk_rows < 1) p->chunk_rows = 1;
+    }
+    const int64_t nchunks = p->nchunks;
 
     auto halloc = [&](void **ptr, size_t bytes) {
         return hipMalloc(ptr, bytes > 0 ? bytes : 1) == hipSuccess;
@@ -262,8 +296,13 @@ extern "C" dd_status dd_partitioner_run(dd_partitioner *p, void *stream) {
     if (!p) return set_err(DD_ERR_INVALID, "null partitioner");
     hipStream_t s = (hipStream_t)stream;
     HIP_TRY(hipEventRecord(p->ev[0], s));
-    HIP_TRY(dd_launch_hash_count(&p->ka, p->nchunks, p->chunk_rows, p->nparts, p->nbits, p->pid,
-                                 p->counts, p->bcounts, p->lds_k1, s));
+    if (p->staged) {
+        HIP_TRY(dd_launch_hash_count_tile(&p->ka, p->nchunks, p->chunk_rows, p->nparts,
+                                          p->nbits, p->pid, p->counts, p->lds_k1, s));
+    } else {
+        HIP_TRY(dd_launch_hash_count(&p->ka, p->nchunks, p->chunk_rows, p->nparts, p->nbits,
+                                     p->pid, p->counts, p->bcounts, p->lds_k1, s));
+    }
     HIP_TRY(hipEventRecord(p->ev[1], s));
     HIP_TRY(dd_launch_scan(p->counts, p->nchunks, p->nparts, DD_SCAN_RANGES, p->partials,
                            p->part_offsets, s));
@@ -274,9 +313,15 @@ extern "C" dd_status dd_partitioner_run(dd_partitioner *p, void *stream) {
                                p->part_boffsets + (size_t)v * (p->nparts + 1), s));
     }
     HIP_TRY(hipEventRecord(p->ev[2], s));
-    HIP_TRY(dd_launch_scatter(&p->ka, p->nchunks, p->chunk_rows, p->nparts, p->nbits, p->pid,
-                              p->counts, p->part_offsets, p->bcounts, p->part_boffsets,
-                              p->lds_k3, s));
+    if (p->staged) {
+        HIP_TRY(dd_launch_scatter_staged(&p->ka, p->nchunks, p->chunk_rows, p->nparts,
+                                         p->nbits, p->pid, p->counts, p->part_offsets,
+                                         p->gmax, p->lds_k3, s));
+    } else {
+        HIP_TRY(dd_launch_scatter(&p->ka, p->nchunks, p->chunk_rows, p->nparts, p->nbits,
+                                  p->pid, p->counts, p->part_offsets, p->bcounts,
+                                  p->part_boffsets, p->lds_k3, s));
+    }
     HIP_TRY(hipEventRecord(p->ev[3], s));
     p->has_run = true;
     return DD_OK;

@@ -57,6 +57,13 @@ hipError_t dd_launch_hash_count(const dd_kargs *a, int64_t nchunks, int64_t chun
                                 hipStream_t s);
 hipError_t dd_launch_scan(uint32_t *counts, int64_t nchunks, uint32_t nparts, int nranges,
                           uint32_t *partials, uint64_t *part_offsets, hipStream_t s);
+hipError_t dd_launch_hash_count_tile(const dd_kargs *a, int64_t nblocks, int64_t tile_rows,
+                                     uint32_t nparts, int nbits, uint32_t *pid_out,
+                                     uint32_t *counts, size_t lds_bytes, hipStream_t s);
+hipError_t dd_launch_scatter_staged(const dd_kargs *a, int64_t nblocks, int64_t tile_rows,
+                                    uint32_t nparts, int nbits, const uint32_t *pid_in,
+                                    const uint32_t *tile_off, const uint64_t *part_offsets,
+                                    int gmax, size_t lds_bytes, hipStream_t s);
 hipError_t dd_launch_scatter(const dd_kargs *a, int64_t nchunks, int64_t chunk_rows,
                              uint32_t nparts, int nbits, const uint32_t *pid_in,
                              const uint32_t *chunk_off, const uint64_t *part_offsets,

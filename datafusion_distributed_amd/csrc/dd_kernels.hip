@@ -361,6 +361,249 @@ __global__ __launch_bounds__(BLOCK_THREADS) void k_scatter(
     }
 }
 
+/* ================= v2: block-tile, LDS-staged scatter (fixed-width batches) =========
+ * Motivation (profiles/r01_k3v1_pmc_summary.json): v1's direct scatter writes 3.8x its
+ * algorithmic bytes — scattered 8 B stores dirty 128 B lines that are evicted from the
+ * XCD L2 before they fill. v2 stages each round of R rows in partition-major order in
+ * LDS, then flushes with run-granular coalesced stores (run = R/P rows per partition),
+ * so stores leave the CU in >=line-sized pieces at the bench shape (R=2048, P=128:
+ * 16-row runs = 128 B per 8 B column).
+ *
+ * Work decomposition: one contiguous row tile per BLOCK (nchunks == gridDim.x); within a
+ * tile, rounds of R rows; within a round, wave w owns rows [w*R/4, (w+1)*R/4) (so row
+ * order == (wave, group, lane) order — stability preserved vs the oracle). Fixed-width
+ * columns only; batches with var-width columns take the v1 path. */
+
+__global__ __launch_bounds__(BLOCK_THREADS) void k_hash_count_tile(
+    dd_kargs a, int64_t tile_rows, uint32_t nparts, int nbits, uint32_t *pid_out,
+    uint32_t *counts /* [nblocks][P] */) {
+    extern __shared__ __attribute__((aligned(16))) char smem[];
+    uint32_t *hist = (uint32_t *)smem; /* [4][P] per-wave */
+    const int tid = threadIdx.x;
+    const int wid = tid / WAVE;
+    const int lane = tid % WAVE;
+    uint32_t *myhist = hist + (size_t)wid * nparts;
+
+    for (uint32_t p = tid; p < WAVES_PER_BLOCK * nparts; p += BLOCK_THREADS) hist[p] = 0;
+    __syncthreads();
+
+    const int64_t tstart = (int64_t)blockIdx.x * tile_rows;
+    const int64_t tend = (tstart + tile_rows < a.n_rows) ? (tstart + tile_rows) : a.n_rows;
+
+    for (int64_t base = tstart; base < tend; base += BLOCK_THREADS) {
+        const int64_t row = base + tid;
+        const bool active = row < tend;
+        uint32_t pid = 0;
+        if (active) {
+            uint64_t h = dd_row_hash(a, row);
+            pid = (uint32_t)(h % (uint64_t)nparts);
+            pid_out[row] = pid;
+        }
+        uint64_t act = __ballot(active);
+        if (active) {
+            uint64_t eq = dd_eq_mask(pid, act, nbits);
+            int leader = __ffsll((unsigned long long)eq) - 1;
+            if (lane == leader) myhist[pid] += (uint32_t)__popcll((unsigned long long)eq);
+        }
+    }
+    __syncthreads();
+    for (uint32_t p = tid; p < nparts; p += BLOCK_THREADS) {
+        uint32_t s = 0;
+        for (int w = 0; w < WAVES_PER_BLOCK; w++) s += hist[(size_t)w * nparts + p];
+        counts[(size_t)blockIdx.x * nparts + p] = s;
+    }
+}
+
+/* exclusive scan of vals[0..P) into out[0..P); tmp is u32[BLOCK_THREADS]; barriers inside */
+__device__ __forceinline__ void dd_block_excl_scan(const uint32_t *vals, uint32_t *out,
+                                                   uint32_t P, uint32_t *tmp) {
+    const int tid = threadIdx.x;
+    const uint32_t span = (P + BLOCK_THREADS - 1) / BLOCK_THREADS;
+    const uint32_t lo = tid * span;
+    const uint32_t hi = (lo + span < P) ? lo + span : P;
+    uint32_t s = 0;
+    for (uint32_t i = lo; i < hi; i++) s += vals[i];
+    tmp[tid] = s;
+    __syncthreads();
+    for (int step = 1; step < BLOCK_THREADS; step <<= 1) {
+        uint32_t v = tmp[tid];
+        uint32_t add = (tid >= step) ? tmp[tid - step] : 0;
+        __syncthreads();
+        tmp[tid] = v + add;
+        __syncthreads();
+    }
+    uint32_t run = (tid > 0) ? tmp[tid - 1] : 0;
+    for (uint32_t i = lo; i < hi; i++) {
+        out[i] = run;
+        run += vals[i];
+    }
+    __syncthreads();
+}
+
+template <int GMAX>
+__global__ __launch_bounds__(BLOCK_THREADS) void k_scatter_staged(
+    dd_kargs a, int64_t tile_rows, uint32_t nparts, int nbits, const uint32_t *pid_in,
+    const uint32_t *tile_off /* [nblocks][P] excl within partition */,
+    const uint64_t *part_offsets /* [P+1] */) {
+    constexpr int R = GMAX * BLOCK_THREADS;
+    constexpr int SEG = R / WAVES_PER_BLOCK;
+    extern __shared__ __attribute__((aligned(16))) char smem[];
+    /* carve: dstbase u64[P] | seghist u32[4][P] | roundcnt u32[P] | round_off u32[P] |
+     * scan_tmp u32[256] | dstg u32[R] | per-col staging (16B aligned) | valid staging u8 */
+    char *ws = smem;
+    uint64_t *dstbase = (uint64_t *)ws;
+    ws += sizeof(uint64_t) * nparts;
+    uint32_t *seghist = (uint32_t *)ws;
+    ws += sizeof(uint32_t) * WAVES_PER_BLOCK * nparts;
+    uint32_t *roundcnt = (uint32_t *)ws;
+    ws += sizeof(uint32_t) * nparts;
+    uint32_t *round_off = (uint32_t *)ws;
+    ws += sizeof(uint32_t) * nparts;
+    uint32_t *scan_tmp = (uint32_t *)ws;
+    ws += sizeof(uint32_t) * BLOCK_THREADS;
+    uint32_t *dstg = (uint32_t *)ws;
+    ws += sizeof(uint32_t) * R;
+    char *col_stage[DD_KMAX_COLS];
+    for (int c = 0; c < a.n_cols; c++) {
+        col_stage[c] = ws;
+        ws += (size_t)R * a.cols[c].elem;
+    }
+    char *val_stage[DD_KMAX_COLS];
+    for (int c = 0; c < a.n_cols; c++) {
+        if (a.cols[c].valid) {
+            val_stage[c] = ws;
+            ws += R;
+        } else {
+            val_stage[c] = nullptr;
+        }
+    }
+
+    const int tid = threadIdx.x;
+    const int wid = tid / WAVE;
+    const int lane = tid % WAVE;
+    uint32_t *myseg = seghist + (size_t)wid * nparts;
+    const uint64_t lt = ((uint64_t)1 << lane) - 1;
+
+    const int64_t tstart = (int64_t)blockIdx.x * tile_rows;
+    const int64_t tend = (tstart + tile_rows < a.n_rows) ? (tstart + tile_rows) : a.n_rows;
+
+    for (uint32_t p = tid; p < nparts; p += BLOCK_THREADS)
+        dstbase[p] = part_offsets[p] + tile_off[(size_t)blockIdx.x * nparts + p];
+    __syncthreads();
+
+    for (int64_t rstart = tstart; rstart < tend; rstart += R) {
+        const int64_t rend = (rstart + R < tend) ? (rstart + R) : tend;
+        const int round_rows = (int)(rend - rstart);
+        for (uint32_t i = tid; i < WAVES_PER_BLOCK * nparts; i += BLOCK_THREADS)
+            seghist[i] = 0;
+        __syncthreads();
+
+        /* pass 1: per-wave stable ranking over its contiguous segment */
+        const int64_t segstart = rstart + (int64_t)wid * SEG;
+        uint32_t pidr[GMAX], rankr[GMAX];
+        bool actr[GMAX];
+#pragma unroll
+        for (int g = 0; g < GMAX; g++) {
+            const int64_t row = segstart + g * WAVE + lane;
+            const bool active = row < rend;
+            actr[g] = active;
+            uint32_t pid = 0;
+            if (active) pid = pid_in[row];
+            pidr[g] = pid;
+            uint64_t act = __ballot(active);
+            uint32_t rk = 0;
+            if (active) {
+                uint64_t eq = dd_eq_mask(pid, act, nbits);
+                int leader = __ffsll((unsigned long long)eq) - 1;
+                uint32_t base = 0;
+                if (lane == leader) {
+                    base = myseg[pid];
+                    myseg[pid] = base + (uint32_t)__popcll((unsigned long long)eq);
+                }
+                base = (uint32_t)__shfl((int)base, leader);
+                rk = base + (uint32_t)__popcll((unsigned long long)(eq & lt));
+            }
+            rankr[g] = rk;
+        }
+        __syncthreads();
+
+        /* pass 2: exclusive scan across waves per partition + round totals */
+        for (uint32_t p = tid; p < nparts; p += BLOCK_THREADS) {
+            uint32_t s0 = seghist[p];
+            uint32_t s1 = seghist[nparts + p];
+            uint32_t s2 = seghist[2 * (size_t)nparts + p];
+            uint32_t s3 = seghist[3 * (size_t)nparts + p];
+            seghist[p] = 0;
+            seghist[nparts + p] = s0;
+            seghist[2 * (size_t)nparts + p] = s0 + s1;
+            seghist[3 * (size_t)nparts + p] = s0 + s1 + s2;
+            roundcnt[p] = s0 + s1 + s2 + s3;
+        }
+        __syncthreads();
+
+        /* pass 2b: partition-major layout of the round */
+        dd_block_excl_scan(roundcnt, round_off, nparts, scan_tmp);
+
+        /* pass 3: place rows into the partition-major LDS image */
+#pragma unroll
+        for (int g = 0; g < GMAX; g++) {
+            if (!actr[g]) continue;
+            const int64_t row = segstart + g * WAVE + lane;
+            const uint32_t pid = pidr[g];
+            const uint32_t rank_r = myseg[pid] + rankr[g];
+            const uint32_t slot = round_off[pid] + rank_r;
+            dstg[slot] = (uint32_t)(dstbase[pid] + rank_r);
+            for (int c = 0; c < a.n_cols; c++) {
+                const dd_kcol &col = a.cols[c];
+                switch (col.elem) {
+                case 1:
+                    ((uint8_t *)col_stage[c])[slot] = ((const uint8_t *)col.data)[row];
+                    break;
+                case 2:
+                    ((uint16_t *)col_stage[c])[slot] = ((const uint16_t *)col.data)[row];
+                    break;
+                case 4:
+                    ((uint32_t *)col_stage[c])[slot] = ((const uint32_t *)col.data)[row];
+                    break;
+                case 8:
+                    ((uint64_t *)col_stage[c])[slot] = ((const uint64_t *)col.data)[row];
+                    break;
+                }
+                if (col.valid) ((uint8_t *)val_stage[c])[slot] = col.valid[row];
+            }
+        }
+        __syncthreads();
+
+        /* pass 4: flush — consecutive LDS slots map to consecutive global rows within a
+         * partition run, so these stores coalesce into >=run-sized segments */
+        for (int i = tid; i < round_rows; i += BLOCK_THREADS) {
+            const uint64_t dst = dstg[i];
+            for (int c = 0; c < a.n_cols; c++) {
+                const dd_kcol &col = a.cols[c];
+                switch (col.elem) {
+                case 1:
+                    ((uint8_t *)col.out_data)[dst] = ((const uint8_t *)col_stage[c])[i];
+                    break;
+                case 2:
+                    ((uint16_t *)col.out_data)[dst] = ((const uint16_t *)col_stage[c])[i];
+                    break;
+                case 4:
+                    ((uint32_t *)col.out_data)[dst] = ((const uint32_t *)col_stage[c])[i];
+                    break;
+                case 8:
+                    ((uint64_t *)col.out_data)[dst] = ((const uint64_t *)col_stage[c])[i];
+                    break;
+                }
+                if (col.valid) col.out_valid[dst] = ((const uint8_t *)val_stage[c])[i];
+            }
+        }
+        /* advance per-partition bases past this round */
+        __syncthreads();
+        for (uint32_t p = tid; p < nparts; p += BLOCK_THREADS) dstbase[p] += roundcnt[p];
+        __syncthreads();
+    }
+}
+
 /* ---------------- launchers (called from dd_host.cpp) ---------------- */
 
 extern "C" {
@@ -401,6 +644,54 @@ hipError_t dd_launch_scan(uint32_t *counts, int64_t nchunks, uint32_t nparts, in
                        part_offsets);
     hipLaunchKernelGGL(k_scan_rewrite, dim3(blocks), dim3(threads), 0, s, counts, nchunks,
                        nparts, nranges, partials);
+    return hipGetLastError();
+}
+
+hipError_t dd_launch_hash_count_tile(const dd_kargs *a, int64_t nblocks, int64_t tile_rows,
+                                     uint32_t nparts, int nbits, uint32_t *pid_out,
+                                     uint32_t *counts, size_t lds_bytes, hipStream_t s) {
+    if (lds_bytes > 65536) {
+        hipError_t e = hipFuncSetAttribute((const void *)k_hash_count_tile,
+                                           hipFuncAttributeMaxDynamicSharedMemorySize,
+                                           (int)lds_bytes);
+        if (e != hipSuccess) return e;
+    }
+    hipLaunchKernelGGL(k_hash_count_tile, dim3((unsigned)nblocks), dim3(BLOCK_THREADS),
+                       lds_bytes, s, *a, tile_rows, nparts, nbits, pid_out, counts);
+    return hipGetLastError();
+}
+
+hipError_t dd_launch_scatter_staged(const dd_kargs *a, int64_t nblocks, int64_t tile_rows,
+                                    uint32_t nparts, int nbits, const uint32_t *pid_in,
+                                    const uint32_t *tile_off, const uint64_t *part_offsets,
+                                    int gmax, size_t lds_bytes, hipStream_t s) {
+    const void *fn;
+    switch (gmax) {
+    case 2: fn = (const void *)k_scatter_staged<2>; break;
+    case 4: fn = (const void *)k_scatter_staged<4>; break;
+    case 8: fn = (const void *)k_scatter_staged<8>; break;
+    default: return hipErrorInvalidValue;
+    }
+    if (lds_bytes > 65536) {
+        hipError_t e = hipFuncSetAttribute(fn, hipFuncAttributeMaxDynamicSharedMemorySize,
+                                           (int)lds_bytes);
+        if (e != hipSuccess) return e;
+    }
+    dim3 grid((unsigned)nblocks);
+    switch (gmax) {
+    case 2:
+        hipLaunchKernelGGL(k_scatter_staged<2>, grid, dim3(BLOCK_THREADS), lds_bytes, s, *a,
+                           tile_rows, nparts, nbits, pid_in, tile_off, part_offsets);
+        break;
+    case 4:
+        hipLaunchKernelGGL(k_scatter_staged<4>, grid, dim3(BLOCK_THREADS), lds_bytes, s, *a,
+                           tile_rows, nparts, nbits, pid_in, tile_off, part_offsets);
+        break;
+    case 8:
+        hipLaunchKernelGGL(k_scatter_staged<8>, grid, dim3(BLOCK_THREADS), lds_bytes, s, *a,
+                           tile_rows, nparts, nbits, pid_in, tile_off, part_offsets);
+        break;
+    }
     return hipGetLastError();
 }
 
