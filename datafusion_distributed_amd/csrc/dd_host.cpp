@@ -10,6 +10,7 @@
 #include <rccl/rccl.h>
 
 #include <cmath>
+#include <cstdlib>
 #include <cstdio>
 #include <cstring>
 #include <map>
@@ -210,7 +211,12 @@ extern "C" dd_status dd_partitioner_create(const dd_batch_desc *batch, const int
         }
         row_stage += nvalid;
         const size_t part_lds = (size_t)P * (8 + 4 * WAVES_PER_BLOCK_H + 4 + 4) + 256 * 4;
-        for (int g = 8; g >= 2; g /= 2) {
+        int gtop = 8;
+        if (const char *e = getenv("DD_V2_GMAX")) {
+            int v = atoi(e); /* perf experiment knob; 0/unset = auto */
+            if (v == 2 || v == 4 || v == 8) gtop = v;
+        }
+        for (int g = gtop; g >= 2; g /= 2) {
             size_t lds = part_lds + (size_t)g * 256 * row_stage;
             if (lds <= 163840) {
                 p->staged = true;

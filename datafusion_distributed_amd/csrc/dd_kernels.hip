@@ -157,7 +157,9 @@ __global__ __launch_bounds__(BLOCK_THREADS) void k_hash_count(
         uint32_t pid = 0;
         if (active) {
             uint64_t h = dd_row_hash(a, row);
-            pid = (uint32_t)(h % (uint64_t)nparts);
+            /* mask == mod for power-of-two P (the common case); u64 division otherwise */
+            pid = ((nparts & (nparts - 1)) == 0) ? (uint32_t)(h & (uint64_t)(nparts - 1))
+                                                 : (uint32_t)(h % (uint64_t)nparts);
             pid_out[row] = pid;
         }
         uint64_t act = __ballot(active);
@@ -396,7 +398,9 @@ __global__ __launch_bounds__(BLOCK_THREADS) void k_hash_count_tile(
         uint32_t pid = 0;
         if (active) {
             uint64_t h = dd_row_hash(a, row);
-            pid = (uint32_t)(h % (uint64_t)nparts);
+            /* mask == mod for power-of-two P (the common case); u64 division otherwise */
+            pid = ((nparts & (nparts - 1)) == 0) ? (uint32_t)(h & (uint64_t)(nparts - 1))
+                                                 : (uint32_t)(h % (uint64_t)nparts);
             pid_out[row] = pid;
         }
         uint64_t act = __ballot(active);
@@ -414,7 +418,8 @@ __global__ __launch_bounds__(BLOCK_THREADS) void k_hash_count_tile(
     }
 }
 
-/* exclusive scan of vals[0..P) into out[0..P); tmp is u32[BLOCK_THREADS]; barriers inside */
+/* exclusive scan of vals[0..P) into out[0..P); tmp is u32[BLOCK_THREADS]; 3 barriers:
+ * per-thread span sums -> ONE wave shfl-scans the 256 span totals -> spans rewritten */
 __device__ __forceinline__ void dd_block_excl_scan(const uint32_t *vals, uint32_t *out,
                                                    uint32_t P, uint32_t *tmp) {
     const int tid = threadIdx.x;
@@ -425,13 +430,21 @@ __device__ __forceinline__ void dd_block_excl_scan(const uint32_t *vals, uint32_
     for (uint32_t i = lo; i < hi; i++) s += vals[i];
     tmp[tid] = s;
     __syncthreads();
-    for (int step = 1; step < BLOCK_THREADS; step <<= 1) {
-        uint32_t v = tmp[tid];
-        uint32_t add = (tid >= step) ? tmp[tid - step] : 0;
-        __syncthreads();
-        tmp[tid] = v + add;
-        __syncthreads();
+    if (tid < WAVE) {
+        uint32_t carry = 0;
+        for (int k = 0; k < BLOCK_THREADS / WAVE; k++) {
+            uint32_t v = tmp[k * WAVE + tid];
+#pragma unroll
+            for (int d = 1; d < WAVE; d <<= 1) {
+                uint32_t u = (uint32_t)__shfl_up((int)v, d);
+                if (tid >= d) v += u;
+            }
+            v += carry;
+            tmp[k * WAVE + tid] = v; /* inclusive */
+            carry = (uint32_t)__shfl((int)v, WAVE - 1);
+        }
     }
+    __syncthreads();
     uint32_t run = (tid > 0) ? tmp[tid - 1] : 0;
     for (uint32_t i = lo; i < hi; i++) {
         out[i] = run;
