@@ -106,7 +106,8 @@ struct dd_partitioner {
     int64_t nchunks = 0, chunk_rows = 0;
     size_t lds_k1 = 0, lds_k3 = 0;
     bool staged = false; /* v2 path: block-tile + LDS-staged scatter (fixed-width only) */
-    int gmax = 0;        /* v2 rows-per-round = gmax * 256 */
+    int gmax = 0;        /* v2 groups per wave per round */
+    int wpb = 4;         /* v2 waves per block; rows per round R = gmax * wpb * 64 */
 
     /* device buffers (owned) */
     uint32_t *pid = nullptr;
@@ -210,17 +211,24 @@ extern "C" dd_status dd_partitioner_create(const dd_batch_desc *batch, const int
             if (batch->cols[c].validity) nvalid++;
         }
         row_stage += nvalid;
-        const size_t part_lds = (size_t)P * (8 + 4 * WAVES_PER_BLOCK_H + 4 + 4) + 256 * 4;
-        int gtop = 8;
+        /* auto: 8 waves x 4 groups (R=2048, 2 blocks/CU at the bench shape) measured best;
+         * DD_V2_GMAX / DD_V2_WPB override for experiments */
+        int wpb = 8, gtop = 4;
+        if (const char *e = getenv("DD_V2_WPB")) {
+            int v = atoi(e);
+            if (v == 4 || v == 8) wpb = v;
+        }
         if (const char *e = getenv("DD_V2_GMAX")) {
-            int v = atoi(e); /* perf experiment knob; 0/unset = auto */
+            int v = atoi(e);
             if (v == 2 || v == 4 || v == 8) gtop = v;
         }
         for (int g = gtop; g >= 2; g /= 2) {
-            size_t lds = part_lds + (size_t)g * 256 * row_stage;
+            const size_t part_lds = (size_t)P * (8 + 4 * wpb + 4 + 4) + (size_t)wpb * 64 * 4;
+            size_t lds = part_lds + (size_t)g * wpb * 64 * row_stage;
             if (lds <= 163840) {
                 p->staged = true;
                 p->gmax = g;
+                p->wpb = wpb;
                 p->lds_k3 = lds;
                 break;
             }
@@ -322,7 +330,7 @@ extern "C" dd_status dd_partitioner_run(dd_partitioner *p, void *stream) {
     if (p->staged) {
         HIP_TRY(dd_launch_scatter_staged(&p->ka, p->nchunks, p->chunk_rows, p->nparts,
                                          p->nbits, p->pid, p->counts, p->part_offsets,
-                                         p->gmax, p->lds_k3, s));
+                                         p->gmax, p->wpb, p->lds_k3, s));
     } else {
         HIP_TRY(dd_launch_scatter(&p->ka, p->nchunks, p->chunk_rows, p->nparts, p->nbits,
                                   p->pid, p->counts, p->part_offsets, p->bcounts,

@@ -63,7 +63,7 @@ hipError_t dd_launch_hash_count_tile(const dd_kargs *a, int64_t nblocks, int64_t
 hipError_t dd_launch_scatter_staged(const dd_kargs *a, int64_t nblocks, int64_t tile_rows,
                                     uint32_t nparts, int nbits, const uint32_t *pid_in,
                                     const uint32_t *tile_off, const uint64_t *part_offsets,
-                                    int gmax, size_t lds_bytes, hipStream_t s);
+                                    int gmax, int wpb, size_t lds_bytes, hipStream_t s);
 hipError_t dd_launch_scatter(const dd_kargs *a, int64_t nchunks, int64_t chunk_rows,
                              uint32_t nparts, int nbits, const uint32_t *pid_in,
                              const uint32_t *chunk_off, const uint64_t *part_offsets,

@@ -420,10 +420,11 @@ __global__ __launch_bounds__(BLOCK_THREADS) void k_hash_count_tile(
 
 /* exclusive scan of vals[0..P) into out[0..P); tmp is u32[BLOCK_THREADS]; 3 barriers:
  * per-thread span sums -> ONE wave shfl-scans the 256 span totals -> spans rewritten */
+template <int BT>
 __device__ __forceinline__ void dd_block_excl_scan(const uint32_t *vals, uint32_t *out,
                                                    uint32_t P, uint32_t *tmp) {
     const int tid = threadIdx.x;
-    const uint32_t span = (P + BLOCK_THREADS - 1) / BLOCK_THREADS;
+    const uint32_t span = (P + BT - 1) / BT;
     const uint32_t lo = tid * span;
     const uint32_t hi = (lo + span < P) ? lo + span : P;
     uint32_t s = 0;
@@ -432,7 +433,7 @@ __device__ __forceinline__ void dd_block_excl_scan(const uint32_t *vals, uint32_
     __syncthreads();
     if (tid < WAVE) {
         uint32_t carry = 0;
-        for (int k = 0; k < BLOCK_THREADS / WAVE; k++) {
+        for (int k = 0; k < BT / WAVE; k++) {
             uint32_t v = tmp[k * WAVE + tid];
 #pragma unroll
             for (int d = 1; d < WAVE; d <<= 1) {
@@ -453,13 +454,14 @@ __device__ __forceinline__ void dd_block_excl_scan(const uint32_t *vals, uint32_
     __syncthreads();
 }
 
-template <int GMAX>
-__global__ __launch_bounds__(BLOCK_THREADS) void k_scatter_staged(
+template <int GMAX, int WPB>
+__global__ __launch_bounds__(WPB * WAVE) void k_scatter_staged(
     dd_kargs a, int64_t tile_rows, uint32_t nparts, int nbits, const uint32_t *pid_in,
     const uint32_t *tile_off /* [nblocks][P] excl within partition */,
     const uint64_t *part_offsets /* [P+1] */) {
-    constexpr int R = GMAX * BLOCK_THREADS;
-    constexpr int SEG = R / WAVES_PER_BLOCK;
+    constexpr int BT = WPB * WAVE;
+    constexpr int R = GMAX * BT;
+    constexpr int SEG = R / WPB;
     extern __shared__ __attribute__((aligned(16))) char smem[];
     /* carve: dstbase u64[P] | seghist u32[4][P] | roundcnt u32[P] | round_off u32[P] |
      * scan_tmp u32[256] | dstg u32[R] | per-col staging (16B aligned) | valid staging u8 */
@@ -467,13 +469,13 @@ __global__ __launch_bounds__(BLOCK_THREADS) void k_scatter_staged(
     uint64_t *dstbase = (uint64_t *)ws;
     ws += sizeof(uint64_t) * nparts;
     uint32_t *seghist = (uint32_t *)ws;
-    ws += sizeof(uint32_t) * WAVES_PER_BLOCK * nparts;
+    ws += sizeof(uint32_t) * WPB * nparts;
     uint32_t *roundcnt = (uint32_t *)ws;
     ws += sizeof(uint32_t) * nparts;
     uint32_t *round_off = (uint32_t *)ws;
     ws += sizeof(uint32_t) * nparts;
     uint32_t *scan_tmp = (uint32_t *)ws;
-    ws += sizeof(uint32_t) * BLOCK_THREADS;
+    ws += sizeof(uint32_t) * BT;
     uint32_t *dstg = (uint32_t *)ws;
     ws += sizeof(uint32_t) * R;
     char *col_stage[DD_KMAX_COLS];
@@ -500,14 +502,14 @@ __global__ __launch_bounds__(BLOCK_THREADS) void k_scatter_staged(
     const int64_t tstart = (int64_t)blockIdx.x * tile_rows;
     const int64_t tend = (tstart + tile_rows < a.n_rows) ? (tstart + tile_rows) : a.n_rows;
 
-    for (uint32_t p = tid; p < nparts; p += BLOCK_THREADS)
+    for (uint32_t p = tid; p < nparts; p += BT)
         dstbase[p] = part_offsets[p] + tile_off[(size_t)blockIdx.x * nparts + p];
     __syncthreads();
 
     for (int64_t rstart = tstart; rstart < tend; rstart += R) {
         const int64_t rend = (rstart + R < tend) ? (rstart + R) : tend;
         const int round_rows = (int)(rend - rstart);
-        for (uint32_t i = tid; i < WAVES_PER_BLOCK * nparts; i += BLOCK_THREADS)
+        for (uint32_t i = tid; i < WPB * nparts; i += BT)
             seghist[i] = 0;
         __syncthreads();
 
@@ -541,21 +543,20 @@ __global__ __launch_bounds__(BLOCK_THREADS) void k_scatter_staged(
         __syncthreads();
 
         /* pass 2: exclusive scan across waves per partition + round totals */
-        for (uint32_t p = tid; p < nparts; p += BLOCK_THREADS) {
-            uint32_t s0 = seghist[p];
-            uint32_t s1 = seghist[nparts + p];
-            uint32_t s2 = seghist[2 * (size_t)nparts + p];
-            uint32_t s3 = seghist[3 * (size_t)nparts + p];
-            seghist[p] = 0;
-            seghist[nparts + p] = s0;
-            seghist[2 * (size_t)nparts + p] = s0 + s1;
-            seghist[3 * (size_t)nparts + p] = s0 + s1 + s2;
-            roundcnt[p] = s0 + s1 + s2 + s3;
+        for (uint32_t p = tid; p < nparts; p += BT) {
+            uint32_t run = 0;
+#pragma unroll
+            for (int w = 0; w < WPB; w++) {
+                uint32_t v = seghist[(size_t)w * nparts + p];
+                seghist[(size_t)w * nparts + p] = run;
+                run += v;
+            }
+            roundcnt[p] = run;
         }
         __syncthreads();
 
         /* pass 2b: partition-major layout of the round */
-        dd_block_excl_scan(roundcnt, round_off, nparts, scan_tmp);
+        dd_block_excl_scan<BT>(roundcnt, round_off, nparts, scan_tmp);
 
         /* pass 3: place rows into the partition-major LDS image */
 #pragma unroll
@@ -589,7 +590,7 @@ __global__ __launch_bounds__(BLOCK_THREADS) void k_scatter_staged(
 
         /* pass 4: flush — consecutive LDS slots map to consecutive global rows within a
          * partition run, so these stores coalesce into >=run-sized segments */
-        for (int i = tid; i < round_rows; i += BLOCK_THREADS) {
+        for (int i = tid; i < round_rows; i += BT) {
             const uint64_t dst = dstg[i];
             for (int c = 0; c < a.n_cols; c++) {
                 const dd_kcol &col = a.cols[c];
@@ -612,7 +613,7 @@ __global__ __launch_bounds__(BLOCK_THREADS) void k_scatter_staged(
         }
         /* advance per-partition bases past this round */
         __syncthreads();
-        for (uint32_t p = tid; p < nparts; p += BLOCK_THREADS) dstbase[p] += roundcnt[p];
+        for (uint32_t p = tid; p < nparts; p += BT) dstbase[p] += roundcnt[p];
         __syncthreads();
     }
 }
@@ -677,35 +678,28 @@ hipError_t dd_launch_hash_count_tile(const dd_kargs *a, int64_t nblocks, int64_t
 hipError_t dd_launch_scatter_staged(const dd_kargs *a, int64_t nblocks, int64_t tile_rows,
                                     uint32_t nparts, int nbits, const uint32_t *pid_in,
                                     const uint32_t *tile_off, const uint64_t *part_offsets,
-                                    int gmax, size_t lds_bytes, hipStream_t s) {
-    const void *fn;
-    switch (gmax) {
-    case 2: fn = (const void *)k_scatter_staged<2>; break;
-    case 4: fn = (const void *)k_scatter_staged<4>; break;
-    case 8: fn = (const void *)k_scatter_staged<8>; break;
-    default: return hipErrorInvalidValue;
-    }
-    if (lds_bytes > 65536) {
-        hipError_t e = hipFuncSetAttribute(fn, hipFuncAttributeMaxDynamicSharedMemorySize,
-                                           (int)lds_bytes);
-        if (e != hipSuccess) return e;
-    }
+                                    int gmax, int wpb, size_t lds_bytes, hipStream_t s) {
     dim3 grid((unsigned)nblocks);
-    switch (gmax) {
-    case 2:
-        hipLaunchKernelGGL(k_scatter_staged<2>, grid, dim3(BLOCK_THREADS), lds_bytes, s, *a,
-                           tile_rows, nparts, nbits, pid_in, tile_off, part_offsets);
-        break;
-    case 4:
-        hipLaunchKernelGGL(k_scatter_staged<4>, grid, dim3(BLOCK_THREADS), lds_bytes, s, *a,
-                           tile_rows, nparts, nbits, pid_in, tile_off, part_offsets);
-        break;
-    case 8:
-        hipLaunchKernelGGL(k_scatter_staged<8>, grid, dim3(BLOCK_THREADS), lds_bytes, s, *a,
-                           tile_rows, nparts, nbits, pid_in, tile_off, part_offsets);
-        break;
+#define DD_CASE(G, W)                                                                        \
+    if (gmax == G && wpb == W) {                                                             \
+        if (lds_bytes > 65536) {                                                             \
+            hipError_t e = hipFuncSetAttribute((const void *)k_scatter_staged<G, W>,         \
+                                               hipFuncAttributeMaxDynamicSharedMemorySize,   \
+                                               (int)lds_bytes);                              \
+            if (e != hipSuccess) return e;                                                   \
+        }                                                                                    \
+        hipLaunchKernelGGL((k_scatter_staged<G, W>), grid, dim3(W * WAVE), lds_bytes, s, *a, \
+                           tile_rows, nparts, nbits, pid_in, tile_off, part_offsets);        \
+        return hipGetLastError();                                                            \
     }
-    return hipGetLastError();
+    DD_CASE(2, 4)
+    DD_CASE(4, 4)
+    DD_CASE(8, 4)
+    DD_CASE(2, 8)
+    DD_CASE(4, 8)
+    DD_CASE(8, 8)
+#undef DD_CASE
+    return hipErrorInvalidValue;
 }
 
 hipError_t dd_launch_scatter(const dd_kargs *a, int64_t nchunks, int64_t chunk_rows,
