@@ -478,20 +478,9 @@ __global__ __launch_bounds__(WPB * WAVE) void k_scatter_staged(
     ws += sizeof(uint32_t) * BT;
     uint32_t *dstg = (uint32_t *)ws;
     ws += sizeof(uint32_t) * R;
-    char *col_stage[DD_KMAX_COLS];
-    for (int c = 0; c < a.n_cols; c++) {
-        col_stage[c] = ws;
-        ws += (size_t)R * a.cols[c].elem;
-    }
-    char *val_stage[DD_KMAX_COLS];
-    for (int c = 0; c < a.n_cols; c++) {
-        if (a.cols[c].valid) {
-            val_stage[c] = ws;
-            ws += R;
-        } else {
-            val_stage[c] = nullptr;
-        }
-    }
+    /* column staging regions follow; walked with INCREMENTAL pointers in the column loops
+     * below (a pointer array here would live in scratch: 400 B/lane of global traffic) */
+    char *const stage0 = ws;
 
     const int tid = threadIdx.x;
     const int wid = tid / WAVE;
@@ -567,23 +556,28 @@ __global__ __launch_bounds__(WPB * WAVE) void k_scatter_staged(
             const uint32_t rank_r = myseg[pid] + rankr[g];
             const uint32_t slot = round_off[pid] + rank_r;
             dstg[slot] = (uint32_t)(dstbase[pid] + rank_r);
+            char *stage = stage0;
             for (int c = 0; c < a.n_cols; c++) {
                 const dd_kcol &col = a.cols[c];
                 switch (col.elem) {
                 case 1:
-                    ((uint8_t *)col_stage[c])[slot] = ((const uint8_t *)col.data)[row];
+                    ((uint8_t *)stage)[slot] = ((const uint8_t *)col.data)[row];
                     break;
                 case 2:
-                    ((uint16_t *)col_stage[c])[slot] = ((const uint16_t *)col.data)[row];
+                    ((uint16_t *)stage)[slot] = ((const uint16_t *)col.data)[row];
                     break;
                 case 4:
-                    ((uint32_t *)col_stage[c])[slot] = ((const uint32_t *)col.data)[row];
+                    ((uint32_t *)stage)[slot] = ((const uint32_t *)col.data)[row];
                     break;
                 case 8:
-                    ((uint64_t *)col_stage[c])[slot] = ((const uint64_t *)col.data)[row];
+                    ((uint64_t *)stage)[slot] = ((const uint64_t *)col.data)[row];
                     break;
                 }
-                if (col.valid) ((uint8_t *)val_stage[c])[slot] = col.valid[row];
+                stage += (size_t)R * col.elem;
+                if (col.valid) {
+                    ((uint8_t *)stage)[slot] = col.valid[row];
+                    stage += R;
+                }
             }
         }
         __syncthreads();
@@ -592,23 +586,28 @@ __global__ __launch_bounds__(WPB * WAVE) void k_scatter_staged(
          * partition run, so these stores coalesce into >=run-sized segments */
         for (int i = tid; i < round_rows; i += BT) {
             const uint64_t dst = dstg[i];
+            char *stage = stage0;
             for (int c = 0; c < a.n_cols; c++) {
                 const dd_kcol &col = a.cols[c];
                 switch (col.elem) {
                 case 1:
-                    ((uint8_t *)col.out_data)[dst] = ((const uint8_t *)col_stage[c])[i];
+                    ((uint8_t *)col.out_data)[dst] = ((const uint8_t *)stage)[i];
                     break;
                 case 2:
-                    ((uint16_t *)col.out_data)[dst] = ((const uint16_t *)col_stage[c])[i];
+                    ((uint16_t *)col.out_data)[dst] = ((const uint16_t *)stage)[i];
                     break;
                 case 4:
-                    ((uint32_t *)col.out_data)[dst] = ((const uint32_t *)col_stage[c])[i];
+                    ((uint32_t *)col.out_data)[dst] = ((const uint32_t *)stage)[i];
                     break;
                 case 8:
-                    ((uint64_t *)col.out_data)[dst] = ((const uint64_t *)col_stage[c])[i];
+                    ((uint64_t *)col.out_data)[dst] = ((const uint64_t *)stage)[i];
                     break;
                 }
-                if (col.valid) col.out_valid[dst] = ((const uint8_t *)val_stage[c])[i];
+                stage += (size_t)R * col.elem;
+                if (col.valid) {
+                    col.out_valid[dst] = ((const uint8_t *)stage)[i];
+                    stage += R;
+                }
             }
         }
         /* advance per-partition bases past this round */
