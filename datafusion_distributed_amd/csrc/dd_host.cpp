@@ -202,8 +202,8 @@ extern "C" dd_status dd_partitioner_create(const dd_batch_desc *batch, const int
     const int64_t P = n_partitions;
     const int nvar = ka.n_var;
 
-    /* v2 (staged) eligibility: fixed-width batch whose staging fits LDS */
-    if (nvar == 0) {
+    /* v2 (staged) eligibility: fixed-width batch, <= DD_STAGE_MAXC cols, staging fits LDS */
+    if (nvar == 0 && batch->n_cols <= DD_STAGE_MAXC) {
         size_t row_stage = 4; /* dstg */
         int nvalid = 0;
         for (int c = 0; c < batch->n_cols; c++) {

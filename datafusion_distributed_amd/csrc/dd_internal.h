@@ -12,6 +12,7 @@
 #define DD_KMAX_KEYS 8
 #define DD_KMAX_VAR 4
 #define DD_SCAN_RANGES 64
+#define DD_STAGE_MAXC 8 /* staged-path column cap (register prefetch arrays) */
 
 /* dtype codes mirror dd_dtype in include/dd_shuffle.h */
 enum {

@@ -502,17 +502,36 @@ __global__ __launch_bounds__(WPB * WAVE) void k_scatter_staged(
             seghist[i] = 0;
         __syncthreads();
 
-        /* pass 1: per-wave stable ranking over its contiguous segment */
+        /* pass 1: per-wave stable ranking over its contiguous segment. Column values are
+         * prefetched into registers here so their HBM latency hides under the ranking and
+         * the cross-wave scan (pass 2) instead of being exposed in pass 3. DD_STAGE_MAXC
+         * bounds the unrolled register arrays (host falls back to v1 beyond it). */
         const int64_t segstart = rstart + (int64_t)wid * SEG;
         uint32_t pidr[GMAX], rankr[GMAX];
         bool actr[GMAX];
+        uint64_t colv[GMAX][DD_STAGE_MAXC];
+        uint8_t valv[GMAX][DD_STAGE_MAXC];
 #pragma unroll
         for (int g = 0; g < GMAX; g++) {
             const int64_t row = segstart + g * WAVE + lane;
             const bool active = row < rend;
             actr[g] = active;
             uint32_t pid = 0;
-            if (active) pid = pid_in[row];
+            if (active) {
+                pid = pid_in[row];
+#pragma unroll
+                for (int c = 0; c < DD_STAGE_MAXC; c++) {
+                    if (c >= a.n_cols) break;
+                    const dd_kcol &col = a.cols[c];
+                    switch (col.elem) {
+                    case 1: colv[g][c] = ((const uint8_t *)col.data)[row]; break;
+                    case 2: colv[g][c] = ((const uint16_t *)col.data)[row]; break;
+                    case 4: colv[g][c] = ((const uint32_t *)col.data)[row]; break;
+                    case 8: colv[g][c] = ((const uint64_t *)col.data)[row]; break;
+                    }
+                    if (col.valid) valv[g][c] = col.valid[row];
+                }
+            }
             pidr[g] = pid;
             uint64_t act = __ballot(active);
             uint32_t rk = 0;
@@ -551,31 +570,24 @@ __global__ __launch_bounds__(WPB * WAVE) void k_scatter_staged(
 #pragma unroll
         for (int g = 0; g < GMAX; g++) {
             if (!actr[g]) continue;
-            const int64_t row = segstart + g * WAVE + lane;
             const uint32_t pid = pidr[g];
             const uint32_t rank_r = myseg[pid] + rankr[g];
             const uint32_t slot = round_off[pid] + rank_r;
             dstg[slot] = (uint32_t)(dstbase[pid] + rank_r);
             char *stage = stage0;
-            for (int c = 0; c < a.n_cols; c++) {
+#pragma unroll
+            for (int c = 0; c < DD_STAGE_MAXC; c++) {
+                if (c >= a.n_cols) break;
                 const dd_kcol &col = a.cols[c];
                 switch (col.elem) {
-                case 1:
-                    ((uint8_t *)stage)[slot] = ((const uint8_t *)col.data)[row];
-                    break;
-                case 2:
-                    ((uint16_t *)stage)[slot] = ((const uint16_t *)col.data)[row];
-                    break;
-                case 4:
-                    ((uint32_t *)stage)[slot] = ((const uint32_t *)col.data)[row];
-                    break;
-                case 8:
-                    ((uint64_t *)stage)[slot] = ((const uint64_t *)col.data)[row];
-                    break;
+                case 1: ((uint8_t *)stage)[slot] = (uint8_t)colv[g][c]; break;
+                case 2: ((uint16_t *)stage)[slot] = (uint16_t)colv[g][c]; break;
+                case 4: ((uint32_t *)stage)[slot] = (uint32_t)colv[g][c]; break;
+                case 8: ((uint64_t *)stage)[slot] = colv[g][c]; break;
                 }
                 stage += (size_t)R * col.elem;
                 if (col.valid) {
-                    ((uint8_t *)stage)[slot] = col.valid[row];
+                    ((uint8_t *)stage)[slot] = valv[g][c];
                     stage += R;
                 }
             }
