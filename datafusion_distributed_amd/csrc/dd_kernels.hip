@@ -28,9 +28,6 @@
 #define WAVE 64
 #define WAVES_PER_BLOCK 4
 #define BLOCK_THREADS (WAVE * WAVES_PER_BLOCK)
-#ifndef DD_SCATTER_MINWAVES
-#define DD_SCATTER_MINWAVES 4 /* min waves/SIMD for the staged scatter: caps VGPRs at 128 */
-#endif
 
 /* ---------------- normative hash (must match oracle/dd_oracle.c bit-exactly) ------------ */
 
@@ -458,7 +455,7 @@ __device__ __forceinline__ void dd_block_excl_scan(const uint32_t *vals, uint32_
 }
 
 template <int GMAX, int WPB>
-__global__ __launch_bounds__(WPB * WAVE, GMAX <= 2 ? 4 : 2) void k_scatter_staged(
+__global__ __launch_bounds__(WPB * WAVE) void k_scatter_staged(
     dd_kargs a, int64_t tile_rows, uint32_t nparts, int nbits, const uint32_t *pid_in,
     const uint32_t *tile_off /* [nblocks][P] excl within partition */,
     const uint64_t *part_offsets /* [P+1] */) {
