@@ -213,11 +213,15 @@ extern "C" dd_status dd_partitioner_create(const dd_batch_desc *batch, const int
         row_stage += nvalid;
         /* auto: 8 waves x 4 groups (R=2048, 2 blocks/CU at the bench shape) measured best;
          * DD_V2_GMAX / DD_V2_WPB override for experiments */
-        int wpb = 8, gtop = 4;
+        int wpb = 8;
         if (const char *e = getenv("DD_V2_WPB")) {
             int v = atoi(e);
             if (v == 4 || v == 8) wpb = v;
         }
+        /* aim for flush runs of >= 8 rows per partition (R >= 8P), floor g=2 (measured
+         * best at the bench shape: wpb=8, g=2, P=128) */
+        int gtop = 2;
+        while (gtop < 8 && (size_t)gtop * wpb * 64 < 8 * (size_t)P) gtop *= 2;
         if (const char *e = getenv("DD_V2_GMAX")) {
             int v = atoi(e);
             if (v == 2 || v == 4 || v == 8) gtop = v;

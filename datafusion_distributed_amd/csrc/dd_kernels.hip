@@ -454,7 +454,7 @@ __device__ __forceinline__ void dd_block_excl_scan(const uint32_t *vals, uint32_
     __syncthreads();
 }
 
-template <int GMAX, int WPB>
+template <int GMAX, int WPB, int MAXC>
 __global__ __launch_bounds__(WPB * WAVE) void k_scatter_staged(
     dd_kargs a, int64_t tile_rows, uint32_t nparts, int nbits, const uint32_t *pid_in,
     const uint32_t *tile_off /* [nblocks][P] excl within partition */,
@@ -463,8 +463,8 @@ __global__ __launch_bounds__(WPB * WAVE) void k_scatter_staged(
     constexpr int R = GMAX * BT;
     constexpr int SEG = R / WPB;
     extern __shared__ __attribute__((aligned(16))) char smem[];
-    /* carve: dstbase u64[P] | seghist u32[4][P] | roundcnt u32[P] | round_off u32[P] |
-     * scan_tmp u32[256] | dstg u32[R] | per-col staging (16B aligned) | valid staging u8 */
+    /* carve: dstbase u64[P] | seghist u32[WPB][P] | roundcnt u32[P] | round_off u32[P] |
+     * scan_tmp u32[BT] | dstg u32[R] | per-col staging (walked incrementally) */
     char *ws = smem;
     uint64_t *dstbase = (uint64_t *)ws;
     ws += sizeof(uint64_t) * nparts;
@@ -478,8 +478,6 @@ __global__ __launch_bounds__(WPB * WAVE) void k_scatter_staged(
     ws += sizeof(uint32_t) * BT;
     uint32_t *dstg = (uint32_t *)ws;
     ws += sizeof(uint32_t) * R;
-    /* column staging regions follow; walked with INCREMENTAL pointers in the column loops
-     * below (a pointer array here would live in scratch: 400 B/lane of global traffic) */
     char *const stage0 = ws;
 
     const int tid = threadIdx.x;
@@ -495,44 +493,56 @@ __global__ __launch_bounds__(WPB * WAVE) void k_scatter_staged(
         dstbase[p] = part_offsets[p] + tile_off[(size_t)blockIdx.x * nparts + p];
     __syncthreads();
 
-    for (int64_t rstart = tstart; rstart < tend; rstart += R) {
-        const int64_t rend = (rstart + R < tend) ? (rstart + R) : tend;
-        const int round_rows = (int)(rend - rstart);
-        for (uint32_t i = tid; i < WPB * nparts; i += BT)
-            seghist[i] = 0;
-        __syncthreads();
+    /* per-round register state; (re)loaded by `preload` one round AHEAD so the global
+     * loads overlap the previous round's flush */
+    uint32_t pidr[GMAX], rankr[GMAX];
+    bool actr[GMAX];
+    uint64_t colv[GMAX][MAXC];
+    uint8_t valv[GMAX][MAXC];
 
-        /* pass 1: per-wave stable ranking over its contiguous segment. Column values are
-         * prefetched into registers here so their HBM latency hides under the ranking and
-         * the cross-wave scan (pass 2) instead of being exposed in pass 3. DD_STAGE_MAXC
-         * bounds the unrolled register arrays (host falls back to v1 beyond it). */
+    auto preload = [&](int64_t rstart, int64_t rend) {
         const int64_t segstart = rstart + (int64_t)wid * SEG;
-        uint32_t pidr[GMAX], rankr[GMAX];
-        bool actr[GMAX];
-        uint64_t colv[GMAX][DD_STAGE_MAXC];
-        uint8_t valv[GMAX][DD_STAGE_MAXC];
 #pragma unroll
         for (int g = 0; g < GMAX; g++) {
             const int64_t row = segstart + g * WAVE + lane;
             const bool active = row < rend;
             actr[g] = active;
-            uint32_t pid = 0;
-            if (active) {
-                pid = pid_in[row];
-#pragma unroll
-                for (int c = 0; c < DD_STAGE_MAXC; c++) {
-                    if (c >= a.n_cols) break;
-                    const dd_kcol &col = a.cols[c];
-                    switch (col.elem) {
-                    case 1: colv[g][c] = ((const uint8_t *)col.data)[row]; break;
-                    case 2: colv[g][c] = ((const uint16_t *)col.data)[row]; break;
-                    case 4: colv[g][c] = ((const uint32_t *)col.data)[row]; break;
-                    case 8: colv[g][c] = ((const uint64_t *)col.data)[row]; break;
-                    }
-                    if (col.valid) valv[g][c] = col.valid[row];
-                }
+            if (!active) {
+                pidr[g] = 0;
+                continue;
             }
-            pidr[g] = pid;
+            pidr[g] = pid_in[row];
+#pragma unroll
+            for (int c = 0; c < MAXC; c++) {
+                if (c >= a.n_cols) break;
+                const dd_kcol &col = a.cols[c];
+                switch (col.elem) {
+                case 1: colv[g][c] = ((const uint8_t *)col.data)[row]; break;
+                case 2: colv[g][c] = ((const uint16_t *)col.data)[row]; break;
+                case 4: colv[g][c] = ((const uint32_t *)col.data)[row]; break;
+                case 8: colv[g][c] = ((const uint64_t *)col.data)[row]; break;
+                }
+                if (col.valid) valv[g][c] = col.valid[row];
+            }
+        }
+    };
+
+    {
+        const int64_t rend0 = (tstart + R < tend) ? (tstart + R) : tend;
+        if (tstart < tend) preload(tstart, rend0);
+    }
+
+    for (int64_t rstart = tstart; rstart < tend; rstart += R) {
+        const int64_t rend = (rstart + R < tend) ? (rstart + R) : tend;
+        const int round_rows = (int)(rend - rstart);
+
+        /* rank (LDS-only; rows already in registers). Each wave zeroes and owns its own
+         * seghist row — no barrier needed before ranking. */
+        for (uint32_t p = lane; p < nparts; p += WAVE) myseg[p] = 0;
+#pragma unroll
+        for (int g = 0; g < GMAX; g++) {
+            const bool active = actr[g];
+            const uint32_t pid = pidr[g];
             uint64_t act = __ballot(active);
             uint32_t rk = 0;
             if (active) {
@@ -550,7 +560,7 @@ __global__ __launch_bounds__(WPB * WAVE) void k_scatter_staged(
         }
         __syncthreads();
 
-        /* pass 2: exclusive scan across waves per partition + round totals */
+        /* cross-wave exclusive scan per partition + round totals */
         for (uint32_t p = tid; p < nparts; p += BT) {
             uint32_t run = 0;
 #pragma unroll
@@ -562,11 +572,9 @@ __global__ __launch_bounds__(WPB * WAVE) void k_scatter_staged(
             roundcnt[p] = run;
         }
         __syncthreads();
-
-        /* pass 2b: partition-major layout of the round */
         dd_block_excl_scan<BT>(roundcnt, round_off, nparts, scan_tmp);
 
-        /* pass 3: place rows into the partition-major LDS image */
+        /* place rows into the partition-major LDS image (registers -> LDS) */
 #pragma unroll
         for (int g = 0; g < GMAX; g++) {
             if (!actr[g]) continue;
@@ -576,7 +584,7 @@ __global__ __launch_bounds__(WPB * WAVE) void k_scatter_staged(
             dstg[slot] = (uint32_t)(dstbase[pid] + rank_r);
             char *stage = stage0;
 #pragma unroll
-            for (int c = 0; c < DD_STAGE_MAXC; c++) {
+            for (int c = 0; c < MAXC; c++) {
                 if (c >= a.n_cols) break;
                 const dd_kcol &col = a.cols[c];
                 switch (col.elem) {
@@ -594,17 +602,23 @@ __global__ __launch_bounds__(WPB * WAVE) void k_scatter_staged(
         }
         __syncthreads();
 
-        /* pass 4: flush — consecutive LDS slots map to consecutive global rows within a
-         * partition run, so these stores coalesce into >=run-sized segments */
+        /* issue next round's loads BEFORE the flush so they overlap the stores */
+        if (rstart + R < tend) {
+            const int64_t nrend = (rstart + 2 * R < tend) ? (rstart + 2 * R) : tend;
+            preload(rstart + R, nrend);
+        }
+
+        /* flush: consecutive LDS slots -> consecutive global rows within each partition
+         * run; stores coalesce into run-sized segments */
         for (int i = tid; i < round_rows; i += BT) {
             const uint64_t dst = dstg[i];
             char *stage = stage0;
-            for (int c = 0; c < a.n_cols; c++) {
+#pragma unroll
+            for (int c = 0; c < MAXC; c++) {
+                if (c >= a.n_cols) break;
                 const dd_kcol &col = a.cols[c];
                 switch (col.elem) {
-                case 1:
-                    ((uint8_t *)col.out_data)[dst] = ((const uint8_t *)stage)[i];
-                    break;
+                case 1: ((uint8_t *)col.out_data)[dst] = ((const uint8_t *)stage)[i]; break;
                 case 2:
                     ((uint16_t *)col.out_data)[dst] = ((const uint16_t *)stage)[i];
                     break;
@@ -622,8 +636,7 @@ __global__ __launch_bounds__(WPB * WAVE) void k_scatter_staged(
                 }
             }
         }
-        /* advance per-partition bases past this round */
-        __syncthreads();
+        /* advance per-partition bases (disjoint from the flush's LDS regions) */
         for (uint32_t p = tid; p < nparts; p += BT) dstbase[p] += roundcnt[p];
         __syncthreads();
     }
@@ -691,24 +704,31 @@ hipError_t dd_launch_scatter_staged(const dd_kargs *a, int64_t nblocks, int64_t 
                                     const uint32_t *tile_off, const uint64_t *part_offsets,
                                     int gmax, int wpb, size_t lds_bytes, hipStream_t s) {
     dim3 grid((unsigned)nblocks);
-#define DD_CASE(G, W)                                                                        \
-    if (gmax == G && wpb == W) {                                                             \
+    const int maxc = (a->n_cols <= 4) ? 4 : 8;
+#define DD_CASE(G, W, C)                                                                     \
+    if (gmax == G && wpb == W && maxc == C) {                                                \
         if (lds_bytes > 65536) {                                                             \
-            hipError_t e = hipFuncSetAttribute((const void *)k_scatter_staged<G, W>,         \
+            hipError_t e = hipFuncSetAttribute((const void *)k_scatter_staged<G, W, C>,      \
                                                hipFuncAttributeMaxDynamicSharedMemorySize,   \
                                                (int)lds_bytes);                              \
             if (e != hipSuccess) return e;                                                   \
         }                                                                                    \
-        hipLaunchKernelGGL((k_scatter_staged<G, W>), grid, dim3(W * WAVE), lds_bytes, s, *a, \
-                           tile_rows, nparts, nbits, pid_in, tile_off, part_offsets);        \
+        hipLaunchKernelGGL((k_scatter_staged<G, W, C>), grid, dim3(W * WAVE), lds_bytes, s,  \
+                           *a, tile_rows, nparts, nbits, pid_in, tile_off, part_offsets);    \
         return hipGetLastError();                                                            \
     }
-    DD_CASE(2, 4)
-    DD_CASE(4, 4)
-    DD_CASE(8, 4)
-    DD_CASE(2, 8)
-    DD_CASE(4, 8)
-    DD_CASE(8, 8)
+    DD_CASE(2, 4, 4)
+    DD_CASE(4, 4, 4)
+    DD_CASE(8, 4, 4)
+    DD_CASE(2, 8, 4)
+    DD_CASE(4, 8, 4)
+    DD_CASE(8, 8, 4)
+    DD_CASE(2, 4, 8)
+    DD_CASE(4, 4, 8)
+    DD_CASE(8, 4, 8)
+    DD_CASE(2, 8, 8)
+    DD_CASE(4, 8, 8)
+    DD_CASE(8, 8, 8)
 #undef DD_CASE
     return hipErrorInvalidValue;
 }
