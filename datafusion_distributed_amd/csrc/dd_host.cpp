@@ -218,10 +218,10 @@ extern "C" dd_status dd_partitioner_create(const dd_batch_desc *batch, const int
             int v = atoi(e);
             if (v == 4 || v == 8) wpb = v;
         }
-        /* aim for flush runs of >= 8 rows per partition (R >= 8P), floor g=2 (measured
-         * best at the bench shape: wpb=8, g=2, P=128) */
+        /* aim for flush runs of >= 16 rows per partition (R >= 16P), floor g=2 (measured
+         * best at the bench shape: wpb=8, g=4, P=128) */
         int gtop = 2;
-        while (gtop < 8 && (size_t)gtop * wpb * 64 < 8 * (size_t)P) gtop *= 2;
+        while (gtop < 8 && (size_t)gtop * wpb * 64 < 16 * (size_t)P) gtop *= 2;
         if (const char *e = getenv("DD_V2_GMAX")) {
             int v = atoi(e);
             if (v == 2 || v == 4 || v == 8) gtop = v;
