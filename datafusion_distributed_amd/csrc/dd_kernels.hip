@@ -392,22 +392,34 @@ __global__ __launch_bounds__(BLOCK_THREADS) void k_hash_count_tile(
     const int64_t tstart = (int64_t)blockIdx.x * tile_rows;
     const int64_t tend = (tstart + tile_rows < a.n_rows) ? (tstart + tile_rows) : a.n_rows;
 
-    for (int64_t base = tstart; base < tend; base += BLOCK_THREADS) {
-        const int64_t row = base + tid;
-        const bool active = row < tend;
-        uint32_t pid = 0;
-        if (active) {
-            uint64_t h = dd_row_hash(a, row);
-            /* mask == mod for power-of-two P (the common case); u64 division otherwise */
-            pid = ((nparts & (nparts - 1)) == 0) ? (uint32_t)(h & (uint64_t)(nparts - 1))
-                                                 : (uint32_t)(h % (uint64_t)nparts);
-            pid_out[row] = pid;
+    /* 2 independent row-groups per iteration: doubles loads in flight per wave and
+     * halves the serialized LDS-histogram chain per row */
+    for (int64_t base = tstart; base < tend; base += 2 * BLOCK_THREADS) {
+        uint32_t pidu[2];
+        bool actu[2];
+#pragma unroll
+        for (int u = 0; u < 2; u++) {
+            const int64_t row = base + u * BLOCK_THREADS + tid;
+            actu[u] = row < tend;
+            pidu[u] = 0;
+            if (actu[u]) {
+                uint64_t h = dd_row_hash(a, row);
+                /* mask == mod for power-of-two P; u64 division otherwise */
+                pidu[u] = ((nparts & (nparts - 1)) == 0)
+                              ? (uint32_t)(h & (uint64_t)(nparts - 1))
+                              : (uint32_t)(h % (uint64_t)nparts);
+                pid_out[row] = pidu[u];
+            }
         }
-        uint64_t act = __ballot(active);
-        if (active) {
-            uint64_t eq = dd_eq_mask(pid, act, nbits);
-            int leader = __ffsll((unsigned long long)eq) - 1;
-            if (lane == leader) myhist[pid] += (uint32_t)__popcll((unsigned long long)eq);
+#pragma unroll
+        for (int u = 0; u < 2; u++) {
+            uint64_t act = __ballot(actu[u]);
+            if (actu[u]) {
+                uint64_t eq = dd_eq_mask(pidu[u], act, nbits);
+                int leader = __ffsll((unsigned long long)eq) - 1;
+                if (lane == leader)
+                    myhist[pidu[u]] += (uint32_t)__popcll((unsigned long long)eq);
+            }
         }
     }
     __syncthreads();
