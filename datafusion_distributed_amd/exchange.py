@@ -42,5 +42,8 @@ def create_comm(rank, world):
         t = torch.tensor(list(uid), dtype=torch.uint8)
     else:
         t = torch.zeros(UNIQUE_ID_BYTES, dtype=torch.uint8)
+    # the nccl backend broadcasts device tensors only; gloo wants host tensors
+    if dist.get_backend() == "nccl":
+        t = t.cuda()
     dist.broadcast(t, src=0)
-    return Comm(bytes(t.tolist()), rank, world)
+    return Comm(bytes(t.cpu().tolist()), rank, world)
