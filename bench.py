@@ -15,6 +15,10 @@ config.h2d_GBps (never as value).
 cpu_baseline: the C oracle (oracle/dd_oracle.c — same algorithm; kind "port"; the
 reference itself is Rust and unbuildable here) timed on this box's host
 cores on a bounded sample. Rank 0, N=1 only.
+
+--workload selects additional BASELINE config shapes for extra evidence lines
+(tpch_sf1_q1_repartition = config[1]; clickbench_userid_shuffle = config[4] stand-in);
+the driver's contract lines always use the default tpch_sf10_lineitem_shuffle.
 """
 
 import argparse
@@ -29,14 +33,15 @@ REPO = os.path.dirname(os.path.abspath(__file__))
 sys.path.insert(0, REPO)
 
 SF10_ROWS = 59_986_052
-ROW_BYTES = 28  # i64 + f64 + f64 + i32
 PARTS_PER_RANK = 128
-# K3 algorithmic bytes/row: read pid (4) + read cols (28) + write cols (28)  
-K3_BYTES_PER_ROW = 60
 HBM_PEAK = 8.0e12
 
+FIXED_SIZE = {"u8": 1, "bool": 1, "i16": 2, "i32": 4, "f32": 4, "i64": 8, "f64": 8,
+              "dict32": 4}
 
-def make_lineitem(rows, seed):
+
+def make_lineitem_q3(rows, seed):
+    """Default workload: TPC-H SF10 lineitem q3/q5 shuffle shape (28 B/row, i64 key)."""
     rng = np.random.default_rng(seed)
     return [
         {"dtype": "i64", "data": rng.integers(1, 60_000_000, rows, dtype=np.int64),
@@ -45,15 +50,93 @@ def make_lineitem(rows, seed):
         {"dtype": "f64", "data": rng.uniform(0.0, 0.1, rows), "valid": None},
         {"dtype": "i32", "data": rng.integers(8000, 11000, rows, dtype=np.int64)
          .astype(np.int32), "valid": None},
+    ], [0]
+
+
+def make_lineitem_q1(rows, seed):
+    """BASELINE config[1] shape: TPC-H SF1 q1 — repartition on the low-cardinality
+    (l_returnflag, l_linestatus) pair; q1 aggregate projection payload."""
+    rng = np.random.default_rng(seed)
+    cols = [
+        {"dtype": "u8", "data": rng.integers(0, 3, rows, dtype=np.int64).astype(np.uint8),
+         "valid": None},  # l_returnflag in {A,N,R}
+        {"dtype": "u8", "data": rng.integers(0, 2, rows, dtype=np.int64).astype(np.uint8),
+         "valid": None},  # l_linestatus in {F,O}
+        {"dtype": "f64", "data": rng.uniform(1.0, 50.0, rows), "valid": None},
+        {"dtype": "f64", "data": rng.uniform(900.0, 105000.0, rows), "valid": None},
+        {"dtype": "f64", "data": rng.uniform(0.0, 0.1, rows), "valid": None},
+        {"dtype": "f64", "data": rng.uniform(0.0, 0.08, rows), "valid": None},
+        {"dtype": "i32", "data": rng.integers(8000, 11000, rows, dtype=np.int64)
+         .astype(np.int32), "valid": None},
     ]
+    return cols, [0, 1]
+
+
+def make_clickbench_userid(rows, seed):
+    """BASELINE config[4] stand-in: GROUP BY UserID — Zipf(1.1)-skewed i64 key + wide
+    var-width URL column (exercises the v1 var-width path)."""
+    rng = np.random.default_rng(seed)
+    uid = (rng.zipf(1.1, rows) % 100_000).astype(np.int64)
+    lens = rng.integers(16, 112, rows)
+    off = np.zeros(rows + 1, dtype=np.int32)
+    off[1:] = np.cumsum(lens)
+    url = rng.integers(33, 127, int(off[-1]), dtype=np.int64).astype(np.uint8)
+    return [
+        {"dtype": "i64", "data": uid, "valid": None},
+        {"dtype": "utf8", "data": url, "offsets": off, "valid": None},
+        {"dtype": "i64", "data": rng.integers(0, 2**41, rows, dtype=np.int64),
+         "valid": None},
+        {"dtype": "i32", "data": rng.integers(0, 10**6, rows, dtype=np.int64)
+         .astype(np.int32), "valid": None},
+    ], [0]
+
+
+WORKLOADS = {
+    "tpch_sf10_lineitem_shuffle": (make_lineitem_q3, SF10_ROWS),
+    "tpch_sf1_q1_repartition": (make_lineitem_q1, 6_001_215),
+    "clickbench_userid_shuffle": (make_clickbench_userid, 20_000_000),
+}
+
+
+def workload_bytes(cols):
+    """(total input bytes, K3 algorithmic bytes) for the roofline formula.
+    K3: read pid (4/row) + read+write every column; var cols add offsets read (4/row) and
+    lengths write (4/row)."""
+    n = (len(cols[0]["offsets"]) - 1 if cols[0]["dtype"] == "utf8"
+         else len(cols[0]["data"]))
+    total = 0
+    k3 = 4 * n
+    for c in cols:
+        if c["dtype"] == "utf8":
+            b = int(np.asarray(c["data"]).nbytes)
+            total += b + 4 * n
+            k3 += 2 * b + 4 * n + 4 * n
+        else:
+            b = FIXED_SIZE[c["dtype"]] * n
+            total += b
+            k3 += 2 * b
+        if c.get("valid") is not None:
+            total += n
+            k3 += 2 * n
+    return total, k3
 
 
 def cpu_baseline_leg(rows_sample, key, cols_np, min_seconds=8.0, max_seconds=30.0):
     """Time the C oracle on a bounded sample of the same workload (kind: port)."""
     import oracle
 
-    sample = [{k: (v[:rows_sample] if isinstance(v, np.ndarray) else v) for k, v in c.items()}
-              for c in cols_np]
+    sample = []
+    for c in cols_np:
+        sc = dict(c)
+        if c["dtype"] == "utf8":
+            off = c["offsets"][: rows_sample + 1]
+            sc["offsets"] = off
+            sc["data"] = c["data"][: int(off[-1])]
+        else:
+            sc["data"] = c["data"][:rows_sample]
+        if c.get("valid") is not None:
+            sc["valid"] = c["valid"][:rows_sample]
+        sample.append(sc)
     reps = 0
     t0 = time.perf_counter()
     while True:
@@ -63,14 +146,16 @@ def cpu_baseline_leg(rows_sample, key, cols_np, min_seconds=8.0, max_seconds=30.
         if el >= min_seconds or el >= max_seconds:
             break
     el = time.perf_counter() - t0
-    gbps = rows_sample * ROW_BYTES * reps / el / 1e9
+    sample_bytes, _ = workload_bytes(sample)
+    gbps = sample_bytes * reps / el / 1e9
     return {
         "value": round(gbps, 3),
         "unit": "GB/s",
         "cores": oracle.lib().dd_oracle_num_threads(),
         "kind": "port",
         "sample": f"{rows_sample} rows x {reps} reps of the same hash-repartition "
-                  f"(28 B/row, P={PARTS_PER_RANK}), {el:.1f}s on host cores",
+                  f"({sample_bytes // max(rows_sample, 1)} B/row, P={PARTS_PER_RANK}), "
+                  f"{el:.1f}s on host cores",
     }
 
 
@@ -79,7 +164,9 @@ def main():
     ap.add_argument("--gpus", type=int, default=1)
     ap.add_argument("--steps", type=int, default=10)
     ap.add_argument("--warmup", type=int, default=3)
-    ap.add_argument("--rows", type=int, default=SF10_ROWS, help="rows per rank")
+    ap.add_argument("--rows", type=int, default=None, help="rows per rank")
+    ap.add_argument("--workload", choices=sorted(WORKLOADS), 
+                    default="tpch_sf10_lineitem_shuffle")
     ap.add_argument("--parts-per-rank", type=int, default=PARTS_PER_RANK)
     ap.add_argument("--traffic-bytes", type=float, default=None,
                     help="measured HBM bytes per K3 launch from a separate rocprofv3 "
@@ -110,15 +197,18 @@ def main():
         sys.exit(1)
 
     p_total = args.parts_per_rank * world
-    cols_np = make_lineitem(args.rows, seed=42 + rank)
+    gen, default_rows = WORKLOADS[args.workload]
+    rows = args.rows or default_rows
+    cols_np, key_idx = gen(rows, seed=42 + rank)
+    total_bytes_in, k3_bytes = workload_bytes(cols_np)
 
     th2d0 = time.perf_counter()
     batch = api.DeviceBatch(cols_np)
     api.lib().dd_device_sync()
     th2d = time.perf_counter() - th2d0
-    h2d_gbps = args.rows * ROW_BYTES / th2d / 1e9
+    h2d_gbps = total_bytes_in / th2d / 1e9
 
-    part = api.Partitioner(batch, [0], p_total)
+    part = api.Partitioner(batch, key_idx, p_total)
     comm = create_comm(rank, world) if world > 1 else None
 
     def step():
@@ -157,13 +247,13 @@ def main():
         dist.all_reduce(t, op=dist.ReduceOp.MAX)
         elapsed = float(t.item())
 
-    total_bytes = args.rows * ROW_BYTES * world * args.steps
+    total_bytes = total_bytes_in * world * args.steps
     value = total_bytes / elapsed / 1e9
-    rows_per_s = args.rows * world * args.steps / elapsed
+    rows_per_s = rows * world * args.steps / elapsed
 
     k1_ms, k2_ms, k3_ms = part.kernel_ms()
     k3_mean_ms = k3_ms_total / args.steps
-    achieved = args.rows * K3_BYTES_PER_ROW / (k3_mean_ms / 1e3)
+    achieved = k3_bytes / (k3_mean_ms / 1e3)
     roofline = {
         "bound": "hbm",
         "achieved": round(achieved / 1e9, 1),
@@ -175,15 +265,15 @@ def main():
 
     cpu_baseline = None
     if rank == 0 and world == 1 and not args.skip_cpu_baseline:
-        cpu_baseline = cpu_baseline_leg(min(args.rows, 8_000_000), [0], cols_np)
+        cpu_baseline = cpu_baseline_leg(min(rows, 8_000_000), key_idx, cols_np)
 
     config = {
-        "workload": "tpch_sf10_lineitem_shuffle",
-        "rows_per_rank": args.rows,
-        "row_bytes": ROW_BYTES,
+        "workload": args.workload,
+        "rows_per_rank": rows,
+        "row_bytes": round(total_bytes_in / max(rows, 1), 1),
         "p_total": p_total,
         "parts_per_rank": args.parts_per_rank,
-        "key": "l_orderkey(i64)",
+        "key_cols": key_idx,
         "rows_per_s": round(rows_per_s, 0),
         "kernel_ms_last": {"k1_hash": round(k1_ms, 3), "k2_scan": round(k2_ms, 3),
                            "k3_scatter": round(k3_ms, 3)},
@@ -196,7 +286,8 @@ def main():
         ex.destroy()
 
     line = {
-        "metric": "tpch_sf10_shuffle_GBps",
+        "metric": ("tpch_sf10_shuffle_GBps" if args.workload == "tpch_sf10_lineitem_shuffle"
+                   else args.workload + "_GBps"),
         "value": round(value, 3),
         "unit": "GB/s",
         "n_gpus": world,
