@@ -44,9 +44,9 @@ __device__ __forceinline__ uint64_t dd_hash_bytes_dev(const uint8_t *p, int64_t 
     uint64_t h = 0x9e3779b97f4a7c15ULL ^ ((uint64_t)len * 0xff51afd7ed558ccdULL);
     int64_t i = 0;
     for (; i + 8 <= len; i += 8) {
-        uint64_t c = 0;
-        /* byte loads: string starts are unaligned; correctness first */
-        for (int b = 0; b < 8; b++) c |= (uint64_t)p[i + b] << (8 * b);
+        uint64_t c;
+        /* gfx950 supports unaligned wide loads: this lowers to global_load_dwordx2 */
+        __builtin_memcpy(&c, p + i, 8);
         h = dd_mix64(h ^ c);
     }
     if (i < len) {
@@ -357,7 +357,15 @@ __global__ __launch_bounds__(BLOCK_THREADS) void k_scatter(
                 col.out_lengths[dst] = len;
                 const uint8_t *src = (const uint8_t *)col.data + o0;
                 uint8_t *d = (uint8_t *)col.out_data + bdst;
-                for (uint32_t b = 0; b < len; b++) d[b] = src[b];
+                /* 8-byte unaligned chunks (native on gfx950); destinations of different
+                 * rows are disjoint so wide stores cannot race */
+                uint32_t b = 0;
+                for (; b + 8 <= len; b += 8) {
+                    uint64_t t;
+                    __builtin_memcpy(&t, src + b, 8);
+                    __builtin_memcpy(d + b, &t, 8);
+                }
+                for (; b < len; b++) d[b] = src[b];
             }
         }
     }
