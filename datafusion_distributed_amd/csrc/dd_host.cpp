@@ -114,9 +114,12 @@ struct dd_partitioner {
     uint32_t *counts = nullptr;          /* [nchunks][P] */
     uint32_t *partials = nullptr;        /* [RANGES][P] */
     uint64_t *part_offsets = nullptr;    /* [P+1] */
-    uint32_t *bcounts = nullptr;         /* [nvar][nchunks][P] */
-    uint32_t *bpartials = nullptr;       /* [nvar][RANGES][P] */
+    uint32_t *bcounts = nullptr;         /* v1: [nvar][nchunks][P] */
+    uint32_t *bpartials = nullptr;       /* v1: [nvar][RANGES][P] */
     uint64_t *part_boffsets = nullptr;   /* [nvar][P+1] */
+    uint32_t *src_row = nullptr;         /* staged-var: permutation out[slot] = input row */
+    uint64_t *out_off[DD_KMAX_VAR] = {}; /* staged-var: Arrow byte offsets [n+1] per var */
+    uint64_t *k4_partials = nullptr;     /* staged-var scan scratch */
     uint64_t *dict_hashes[DD_KMAX_COLS] = {};
     void *out_data[DD_KMAX_COLS] = {};
     uint8_t *out_valid[DD_KMAX_COLS] = {};
@@ -133,6 +136,9 @@ struct dd_partitioner {
         hipFree(bcounts);
         hipFree(bpartials);
         hipFree(part_boffsets);
+        hipFree(src_row);
+        for (auto &o : out_off) hipFree(o);
+        hipFree(k4_partials);
         for (int i = 0; i < DD_KMAX_COLS; i++) {
             hipFree(dict_hashes[i]);
             hipFree(out_data[i]);
@@ -202,8 +208,11 @@ extern "C" dd_status dd_partitioner_create(const dd_batch_desc *batch, const int
     const int64_t P = n_partitions;
     const int nvar = ka.n_var;
 
-    /* v2 (staged) eligibility: fixed-width batch, <= DD_STAGE_MAXC cols, staging fits LDS */
-    if (nvar == 0 && batch->n_cols <= DD_STAGE_MAXC) {
+    /* v2 (staged) eligibility: <= DD_STAGE_MAXC columns including the synthetic staged-var
+     * entries (per var col a VARLEN length column + one ROWID permutation column; the var
+     * BYTES go through K4), staging fits LDS */
+    const int aug_cols = batch->n_cols + nvar + (nvar > 0 ? 1 : 0);
+    if (aug_cols <= DD_STAGE_MAXC) {
         size_t row_stage = 4; /* dstg */
         int nvalid = 0;
         for (int c = 0; c < batch->n_cols; c++) {
@@ -211,6 +220,7 @@ extern "C" dd_status dd_partitioner_create(const dd_batch_desc *batch, const int
             if (batch->cols[c].validity) nvalid++;
         }
         row_stage += nvalid;
+        row_stage += 4 * nvar + (nvar > 0 ? 4 : 0); /* VARLEN cols + ROWID col */
         /* auto: 8 waves x 4 groups (R=2048, 2 blocks/CU at the bench shape) measured best;
          * DD_V2_GMAX / DD_V2_WPB override for experiments */
         int wpb = 8;
@@ -269,10 +279,18 @@ extern "C" dd_status dd_partitioner_create(const dd_batch_desc *batch, const int
               halloc((void **)&p->counts, (size_t)nchunks * P * 4) &&
               halloc((void **)&p->partials, (size_t)DD_SCAN_RANGES * P * 4) &&
               halloc((void **)&p->part_offsets, (size_t)(P + 1) * 8);
-    if (ok && nvar > 0)
-        ok = halloc((void **)&p->bcounts, (size_t)nvar * nchunks * P * 4) &&
-             halloc((void **)&p->bpartials, (size_t)nvar * DD_SCAN_RANGES * P * 4) &&
-             halloc((void **)&p->part_boffsets, (size_t)nvar * (P + 1) * 8);
+    if (ok && nvar > 0) {
+        ok = halloc((void **)&p->part_boffsets, (size_t)nvar * (P + 1) * 8);
+        if (p->staged) {
+            ok = ok && halloc((void **)&p->src_row, (size_t)n * 4) &&
+                 halloc((void **)&p->k4_partials, (size_t)8192 * 8);
+            for (int v = 0; v < nvar && ok; v++)
+                ok = halloc((void **)&p->out_off[v], (size_t)(n + 1) * 8);
+        } else {
+            ok = ok && halloc((void **)&p->bcounts, (size_t)nvar * nchunks * P * 4) &&
+                 halloc((void **)&p->bpartials, (size_t)nvar * DD_SCAN_RANGES * P * 4);
+        }
+    }
     if (!ok) return fail(DD_ERR_HIP, "allocation failed (workspace)");
 
     for (int c = 0; c < batch->n_cols && ok; c++) {
@@ -303,6 +321,28 @@ extern "C" dd_status dd_partitioner_create(const dd_batch_desc *batch, const int
     }
     if (!ok) return fail(DD_ERR_HIP, "allocation failed (outputs)");
 
+    if (p->staged && nvar > 0) {
+        /* synthetic staged columns: per var col its LENGTHS (u32, written partition-major
+         * into out_lengths) and one ROWID permutation column (u32 -> src_row); the var
+         * BYTES are materialized by K4 after the scatter (dd_kernels.hip K4 header) */
+        for (int v = 0; v < nvar; v++) {
+            const int ci = ka.var_idx[v];
+            dd_kcol &kc = ka.cols[ka.n_cols];
+            memset(&kc, 0, sizeof(kc));
+            kc.dtype = DD_KDT_VARLEN;
+            kc.elem = 4;
+            kc.data = batch->cols[ci].offsets;
+            kc.out_data = p->out_lengths[ci];
+            ka.n_cols++;
+        }
+        dd_kcol &kr = ka.cols[ka.n_cols];
+        memset(&kr, 0, sizeof(kr));
+        kr.dtype = DD_KDT_ROWID;
+        kr.elem = 4;
+        kr.out_data = p->src_row;
+        ka.n_cols++;
+    }
+
     for (auto &e : p->ev)
         if (hipEventCreate(&e) != hipSuccess) return fail(DD_ERR_HIP, "event create failed");
 
@@ -324,17 +364,28 @@ extern "C" dd_status dd_partitioner_run(dd_partitioner *p, void *stream) {
     HIP_TRY(hipEventRecord(p->ev[1], s));
     HIP_TRY(dd_launch_scan(p->counts, p->nchunks, p->nparts, DD_SCAN_RANGES, p->partials,
                            p->part_offsets, s));
-    for (int v = 0; v < p->ka.n_var; v++) {
-        HIP_TRY(dd_launch_scan(p->bcounts + (size_t)v * p->nchunks * p->nparts, p->nchunks,
-                               p->nparts, DD_SCAN_RANGES,
-                               p->bpartials + (size_t)v * DD_SCAN_RANGES * p->nparts,
-                               p->part_boffsets + (size_t)v * (p->nparts + 1), s));
+    if (!p->staged) {
+        for (int v = 0; v < p->ka.n_var; v++) {
+            HIP_TRY(dd_launch_scan(p->bcounts + (size_t)v * p->nchunks * p->nparts,
+                                   p->nchunks, p->nparts, DD_SCAN_RANGES,
+                                   p->bpartials + (size_t)v * DD_SCAN_RANGES * p->nparts,
+                                   p->part_boffsets + (size_t)v * (p->nparts + 1), s));
+        }
     }
     HIP_TRY(hipEventRecord(p->ev[2], s));
     if (p->staged) {
         HIP_TRY(dd_launch_scatter_staged(&p->ka, p->nchunks, p->chunk_rows, p->nparts,
                                          p->nbits, p->pid, p->counts, p->part_offsets,
                                          p->gmax, p->wpb, p->lds_k3, s));
+        for (int v = 0; v < p->ka.n_var; v++) {
+            const int ci = p->ka.var_idx[v];
+            const dd_col_desc &cd = p->batch.cols[ci];
+            HIP_TRY(dd_launch_var_bytes(
+                p->out_lengths[ci], p->src_row, cd.offsets, (const uint8_t *)cd.data,
+                p->ka.n_rows, cd.data_len, p->k4_partials, p->out_off[v],
+                (uint8_t *)p->out_data[ci], p->part_offsets, p->nparts,
+                p->part_boffsets + (size_t)v * (p->nparts + 1), s));
+        }
     } else {
         HIP_TRY(dd_launch_scatter(&p->ka, p->nchunks, p->chunk_rows, p->nparts, p->nbits,
                                   p->pid, p->counts, p->part_offsets, p->bcounts,

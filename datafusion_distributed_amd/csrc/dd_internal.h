@@ -25,6 +25,8 @@ enum {
     DD_KDT_BOOL = 7,
     DD_KDT_UTF8 = 8,
     DD_KDT_DICT32 = 9,
+    DD_KDT_VARLEN = 100, /* synthetic (staged-var): data = var col's offsets; value = len */
+    DD_KDT_ROWID = 101,  /* synthetic (staged-var): value = input row index */
 };
 
 struct dd_kcol {
@@ -70,6 +72,12 @@ hipError_t dd_launch_scatter(const dd_kargs *a, int64_t nchunks, int64_t chunk_r
                              const uint32_t *chunk_off, const uint64_t *part_offsets,
                              const uint32_t *chunk_boff, const uint64_t *part_boffsets,
                              size_t lds_bytes, hipStream_t s);
+hipError_t dd_launch_var_bytes(const uint32_t *lens, const uint32_t *src_row,
+                               const int32_t *in_offsets, const uint8_t *in_bytes,
+                               int64_t n, int64_t total_bytes, uint64_t *partials,
+                               uint64_t *out_off, uint8_t *out_bytes,
+                               const uint64_t *part_offsets, uint32_t nparts,
+                               uint64_t *part_boffsets, hipStream_t s);
 }
 
 #endif

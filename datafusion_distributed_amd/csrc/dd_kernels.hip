@@ -536,11 +536,19 @@ __global__ __launch_bounds__(WPB * WAVE) void k_scatter_staged(
             for (int c = 0; c < MAXC; c++) {
                 if (c >= a.n_cols) break;
                 const dd_kcol &col = a.cols[c];
-                switch (col.elem) {
-                case 1: colv[g][c] = ((const uint8_t *)col.data)[row]; break;
-                case 2: colv[g][c] = ((const uint16_t *)col.data)[row]; break;
-                case 4: colv[g][c] = ((const uint32_t *)col.data)[row]; break;
-                case 8: colv[g][c] = ((const uint64_t *)col.data)[row]; break;
+                if (col.dtype == DD_KDT_VARLEN) {
+                    /* synthetic: the var column's per-row byte length (K4 header) */
+                    const int32_t *off = (const int32_t *)col.data;
+                    colv[g][c] = (uint32_t)(off[row + 1] - off[row]);
+                } else if (col.dtype == DD_KDT_ROWID) {
+                    colv[g][c] = (uint32_t)row; /* synthetic: permutation for K4d */
+                } else {
+                    switch (col.elem) {
+                    case 1: colv[g][c] = ((const uint8_t *)col.data)[row]; break;
+                    case 2: colv[g][c] = ((const uint16_t *)col.data)[row]; break;
+                    case 4: colv[g][c] = ((const uint32_t *)col.data)[row]; break;
+                    case 8: colv[g][c] = ((const uint64_t *)col.data)[row]; break;
+                    }
                 }
                 if (col.valid) valv[g][c] = col.valid[row];
             }
@@ -662,9 +670,147 @@ __global__ __launch_bounds__(WPB * WAVE) void k_scatter_staged(
     }
 }
 
+/* ================= K4: var-width bytes for the staged path =================
+ * The staged scatter (v2) handles var columns' LENGTHS and a ROWID permutation as
+ * synthetic fixed u32 columns (DD_KDT_VARLEN / DD_KDT_ROWID, set up by dd_host.cpp).
+ * K4 then materializes the byte buffers in partition-major output order:
+ *   K4a: per-wave partial sums of lengths[s] over contiguous slot ranges
+ *   K4b: one block exclusive-scans the partials (u64)
+ *   K4c: per-wave rewrite: out_off[s] = global exclusive byte offset (u64, n+1)
+ *        == the Arrow offsets of the partition-major output (rebuilt, not just lengths)
+ *   K4d: grid-stride gather-copy: slot s copies input row src_row[s]'s bytes to
+ *        out_off[s] (consecutive slots -> consecutive output bytes: coalesced writes)
+ *   K4e: part_byte_offsets[p] = out_off[part_row_offsets[p]]
+ * No barriers anywhere except K4b. */
+
+#define K4_RANGES 8192
+
+__global__ void k4_len_partials(const uint32_t *lens, int64_t n, uint64_t *partials) {
+    const int64_t w = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) / WAVE;
+    const int lane = threadIdx.x % WAVE;
+    if (w >= K4_RANGES) return;
+    const int64_t lo = n * w / K4_RANGES, hi = n * (w + 1) / K4_RANGES;
+    uint64_t s = 0;
+    for (int64_t i = lo + lane; i < hi; i += WAVE) s += lens[i];
+#pragma unroll
+    for (int d = 32; d >= 1; d >>= 1) s += (uint64_t)__shfl_down((long long)s, d);
+    if (lane == 0) partials[w] = s;
+}
+
+__global__ void k4_scan_partials(uint64_t *partials, int nranges, int64_t total_bytes,
+                                 uint64_t *out_off, int64_t n) {
+    /* 256 threads: span sums -> one-wave shfl scan -> span rewrite */
+    __shared__ uint64_t tmp[256];
+    const int tid = threadIdx.x;
+    const int span = (nranges + 255) / 256;
+    const int lo = tid * span, hi = (lo + span < nranges) ? lo + span : nranges;
+    uint64_t s = 0;
+    for (int i = lo; i < hi; i++) s += partials[i];
+    tmp[tid] = s;
+    __syncthreads();
+    if (tid < WAVE) {
+        uint64_t carry = 0;
+        for (int k = 0; k < 256 / WAVE; k++) {
+            uint64_t v = tmp[k * WAVE + tid];
+#pragma unroll
+            for (int d = 1; d < WAVE; d <<= 1) {
+                uint64_t u = (uint64_t)__shfl_up((long long)v, d);
+                if (tid >= d) v += u;
+            }
+            v += carry;
+            tmp[k * WAVE + tid] = v;
+            carry = (uint64_t)__shfl((long long)v, WAVE - 1);
+        }
+    }
+    __syncthreads();
+    uint64_t run = (tid > 0) ? tmp[tid - 1] : 0;
+    for (int i = lo; i < hi; i++) {
+        uint64_t v = partials[i];
+        partials[i] = run;
+        run += v;
+    }
+    if (tid == 0) out_off[n] = (uint64_t)total_bytes;
+}
+
+__global__ void k4_off_rewrite(const uint32_t *lens, int64_t n, const uint64_t *partials,
+                               uint64_t *out_off) {
+    const int64_t w = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) / WAVE;
+    const int lane = threadIdx.x % WAVE;
+    if (w >= K4_RANGES) return;
+    const int64_t lo = n * w / K4_RANGES, hi = n * (w + 1) / K4_RANGES;
+    uint64_t carry = partials[w];
+    for (int64_t base = lo; base < hi; base += WAVE) {
+        const int64_t s = base + lane;
+        uint32_t len = (s < hi) ? lens[s] : 0;
+        uint64_t v = len;
+#pragma unroll
+        for (int d = 1; d < WAVE; d <<= 1) {
+            uint64_t u = (uint64_t)__shfl_up((long long)v, d);
+            if (lane >= d) v += u;
+        }
+        if (s < hi) out_off[s] = carry + v - len; /* exclusive */
+        carry += (uint64_t)__shfl((long long)v, WAVE - 1);
+    }
+}
+
+__global__ void k4_copy(const uint32_t *src_row, const uint64_t *out_off,
+                        const int32_t *in_offsets, const uint8_t *in_bytes, int64_t n,
+                        uint8_t *out_bytes) {
+    for (int64_t s = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; s < n;
+         s += (int64_t)gridDim.x * blockDim.x) {
+        const uint32_t r = src_row[s];
+        const int32_t o0 = in_offsets[r];
+        const uint32_t len = (uint32_t)(in_offsets[r + 1] - o0);
+        const uint8_t *src = in_bytes + o0;
+        uint8_t *d = out_bytes + out_off[s];
+        uint32_t b = 0;
+        for (; b + 8 <= len; b += 8) {
+            uint64_t t;
+            __builtin_memcpy(&t, src + b, 8);
+            __builtin_memcpy(d + b, &t, 8);
+        }
+        for (; b < len; b++) d[b] = src[b];
+    }
+}
+
+__global__ void k4_part_boffsets(const uint64_t *out_off, const uint64_t *part_offsets,
+                                 uint32_t nparts, int64_t n, int64_t total_bytes,
+                                 uint64_t *part_boffsets) {
+    const uint32_t p = blockIdx.x * blockDim.x + threadIdx.x;
+    if (p > nparts) return;
+    if (p == nparts) {
+        part_boffsets[p] = (uint64_t)total_bytes;
+        return;
+    }
+    const int64_t s = (int64_t)part_offsets[p];
+    part_boffsets[p] = (s < n) ? out_off[s] : (uint64_t)total_bytes;
+}
+
 /* ---------------- launchers (called from dd_host.cpp) ---------------- */
 
 extern "C" {
+
+hipError_t dd_launch_var_bytes(const uint32_t *lens, const uint32_t *src_row,
+                               const int32_t *in_offsets, const uint8_t *in_bytes,
+                               int64_t n, int64_t total_bytes, uint64_t *partials,
+                               uint64_t *out_off, uint8_t *out_bytes,
+                               const uint64_t *part_offsets, uint32_t nparts,
+                               uint64_t *part_boffsets, hipStream_t s) {
+    const int wavegrid = K4_RANGES * WAVE / 256;
+    hipLaunchKernelGGL(k4_len_partials, dim3(wavegrid), dim3(256), 0, s, lens, n, partials);
+    hipLaunchKernelGGL(k4_scan_partials, dim3(1), dim3(256), 0, s, partials, K4_RANGES,
+                       total_bytes, out_off, n);
+    hipLaunchKernelGGL(k4_off_rewrite, dim3(wavegrid), dim3(256), 0, s, lens, n, partials,
+                       out_off);
+    int copy_blocks = (int)((n + 255) / 256);
+    if (copy_blocks > 2048) copy_blocks = 2048;
+    if (copy_blocks < 1) copy_blocks = 1;
+    hipLaunchKernelGGL(k4_copy, dim3(copy_blocks), dim3(256), 0, s, src_row, out_off,
+                       in_offsets, in_bytes, n, out_bytes);
+    hipLaunchKernelGGL(k4_part_boffsets, dim3((nparts + 256) / 256 + 1), dim3(256), 0, s,
+                       out_off, part_offsets, nparts, n, total_bytes, part_boffsets);
+    return hipGetLastError();
+}
 
 hipError_t dd_launch_dict_hashes(const uint8_t *bytes, const int32_t *offsets, int64_t n,
                                  uint64_t *out, hipStream_t s) {

@@ -182,3 +182,26 @@ def test_gpu_task_cache_lifecycle():
     st = api.lib().dd_execute_task(ctypes.byref(keyc), 0, 1, None, ctypes.byref(h))
     assert st == 5  # dropped
     batch.free()
+
+
+def test_gpu_staged_var_two_utf8_with_nulls():
+    # two var cols + nulls + f64: exercises the staged-var path (synthetic VARLEN/ROWID
+    # columns + K4 byte materialization) with multiple byte streams
+    rng = np.random.default_rng(17)
+    n = 300000
+
+    def utf8(maxlen, null_p):
+        lens = rng.integers(0, maxlen, n)
+        off = np.zeros(n + 1, dtype=np.int32)
+        off[1:] = np.cumsum(lens)
+        data = rng.integers(32, 127, int(off[-1]), dtype=np.int64).astype(np.uint8)
+        valid = (rng.random(n) > null_p).astype(np.uint8) if null_p else None
+        return {"dtype": "utf8", "data": data, "offsets": off, "valid": valid}
+
+    cols = [
+        {"dtype": "i64", "data": rng.integers(0, 10**12, n, dtype=np.int64), "valid": None},
+        utf8(48, 0.15),
+        utf8(8, 0.0),
+        {"dtype": "f64", "data": rng.normal(size=n), "valid": None},
+    ]
+    check_against_oracle(cols, [0], 64)
