@@ -474,7 +474,7 @@ __device__ __forceinline__ void dd_block_excl_scan(const uint32_t *vals, uint32_
     __syncthreads();
 }
 
-template <int GMAX, int WPB, int MAXC>
+template <int GMAX, int WPB, int MAXC, bool HASVAR>
 __global__ __launch_bounds__(WPB * WAVE) void k_scatter_staged(
     dd_kargs a, int64_t tile_rows, uint32_t nparts, int nbits, const uint32_t *pid_in,
     const uint32_t *tile_off /* [nblocks][P] excl within partition */,
@@ -536,13 +536,19 @@ __global__ __launch_bounds__(WPB * WAVE) void k_scatter_staged(
             for (int c = 0; c < MAXC; c++) {
                 if (c >= a.n_cols) break;
                 const dd_kcol &col = a.cols[c];
-                if (col.dtype == DD_KDT_VARLEN) {
-                    /* synthetic: the var column's per-row byte length (K4 header) */
-                    const int32_t *off = (const int32_t *)col.data;
-                    colv[g][c] = (uint32_t)(off[row + 1] - off[row]);
-                } else if (col.dtype == DD_KDT_ROWID) {
-                    colv[g][c] = (uint32_t)row; /* synthetic: permutation for K4d */
-                } else {
+                bool synthetic = false;
+                if constexpr (HASVAR) {
+                    if (col.dtype == DD_KDT_VARLEN) {
+                        /* synthetic: the var column's per-row byte length (K4 header) */
+                        const int32_t *off = (const int32_t *)col.data;
+                        colv[g][c] = (uint32_t)(off[row + 1] - off[row]);
+                        synthetic = true;
+                    } else if (col.dtype == DD_KDT_ROWID) {
+                        colv[g][c] = (uint32_t)row; /* synthetic: permutation for K4d */
+                        synthetic = true;
+                    }
+                }
+                if (!synthetic) {
                     switch (col.elem) {
                     case 1: colv[g][c] = ((const uint8_t *)col.data)[row]; break;
                     case 2: colv[g][c] = ((const uint16_t *)col.data)[row]; break;
@@ -871,30 +877,44 @@ hipError_t dd_launch_scatter_staged(const dd_kargs *a, int64_t nblocks, int64_t 
                                     int gmax, int wpb, size_t lds_bytes, hipStream_t s) {
     dim3 grid((unsigned)nblocks);
     const int maxc = (a->n_cols <= 4) ? 4 : 8;
-#define DD_CASE(G, W, C)                                                                     \
-    if (gmax == G && wpb == W && maxc == C) {                                                \
+    const bool hasvar = a->n_var > 0;
+#define DD_CASE(G, W, C, V)                                                                  \
+    if (gmax == G && wpb == W && maxc == C && hasvar == V) {                                 \
         if (lds_bytes > 65536) {                                                             \
-            hipError_t e = hipFuncSetAttribute((const void *)k_scatter_staged<G, W, C>,      \
+            hipError_t e = hipFuncSetAttribute((const void *)k_scatter_staged<G, W, C, V>,   \
                                                hipFuncAttributeMaxDynamicSharedMemorySize,   \
                                                (int)lds_bytes);                              \
             if (e != hipSuccess) return e;                                                   \
         }                                                                                    \
-        hipLaunchKernelGGL((k_scatter_staged<G, W, C>), grid, dim3(W * WAVE), lds_bytes, s,  \
-                           *a, tile_rows, nparts, nbits, pid_in, tile_off, part_offsets);    \
+        hipLaunchKernelGGL((k_scatter_staged<G, W, C, V>), grid, dim3(W * WAVE), lds_bytes,  \
+                           s, *a, tile_rows, nparts, nbits, pid_in, tile_off,                \
+                           part_offsets);                                                    \
         return hipGetLastError();                                                            \
     }
-    DD_CASE(2, 4, 4)
-    DD_CASE(4, 4, 4)
-    DD_CASE(8, 4, 4)
-    DD_CASE(2, 8, 4)
-    DD_CASE(4, 8, 4)
-    DD_CASE(8, 8, 4)
-    DD_CASE(2, 4, 8)
-    DD_CASE(4, 4, 8)
-    DD_CASE(8, 4, 8)
-    DD_CASE(2, 8, 8)
-    DD_CASE(4, 8, 8)
-    DD_CASE(8, 8, 8)
+    DD_CASE(2, 4, 4, false)
+    DD_CASE(4, 4, 4, false)
+    DD_CASE(8, 4, 4, false)
+    DD_CASE(2, 8, 4, false)
+    DD_CASE(4, 8, 4, false)
+    DD_CASE(8, 8, 4, false)
+    DD_CASE(2, 4, 8, false)
+    DD_CASE(4, 4, 8, false)
+    DD_CASE(8, 4, 8, false)
+    DD_CASE(2, 8, 8, false)
+    DD_CASE(4, 8, 8, false)
+    DD_CASE(8, 8, 8, false)
+    DD_CASE(2, 4, 4, true)
+    DD_CASE(4, 4, 4, true)
+    DD_CASE(8, 4, 4, true)
+    DD_CASE(2, 8, 4, true)
+    DD_CASE(4, 8, 4, true)
+    DD_CASE(8, 8, 4, true)
+    DD_CASE(2, 4, 8, true)
+    DD_CASE(4, 4, 8, true)
+    DD_CASE(8, 4, 8, true)
+    DD_CASE(2, 8, 8, true)
+    DD_CASE(4, 8, 8, true)
+    DD_CASE(8, 8, 8, true)
 #undef DD_CASE
     return hipErrorInvalidValue;
 }
