@@ -184,9 +184,12 @@ def test_gpu_task_cache_lifecycle():
     batch.free()
 
 
-def test_gpu_staged_var_two_utf8_with_nulls():
-    # two var cols + nulls + f64: exercises the staged-var path (synthetic VARLEN/ROWID
-    # columns + K4 byte materialization) with multiple byte streams
+@pytest.mark.parametrize("staged_var", ["0", "1"])
+def test_gpu_staged_var_two_utf8_with_nulls(staged_var, monkeypatch):
+    # two var cols + nulls + f64 through BOTH var paths: v1 direct (default) and the
+    # staged-var path (DD_V2_VAR=1: synthetic VARLEN/ROWID columns + K4 byte
+    # materialization with device-rebuilt Arrow offsets)
+    monkeypatch.setenv("DD_V2_VAR", staged_var)
     rng = np.random.default_rng(17)
     n = 300000
 
