@@ -759,23 +759,78 @@ __global__ void k4_off_rewrite(const uint32_t *lens, int64_t n, const uint64_t *
     }
 }
 
-__global__ void k4_copy(const uint32_t *src_row, const uint64_t *out_off,
-                        const int32_t *in_offsets, const uint8_t *in_bytes, int64_t n,
-                        uint8_t *out_bytes) {
-    for (int64_t s = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; s < n;
-         s += (int64_t)gridDim.x * blockDim.x) {
-        const uint32_t r = src_row[s];
-        const int32_t o0 = in_offsets[r];
-        const uint32_t len = (uint32_t)(in_offsets[r + 1] - o0);
-        const uint8_t *src = in_bytes + o0;
-        uint8_t *d = out_bytes + out_off[s];
-        uint32_t b = 0;
-        for (; b + 8 <= len; b += 8) {
-            uint64_t t;
-            __builtin_memcpy(&t, src + b, 8);
-            __builtin_memcpy(d + b, &t, 8);
+/* Wave-cooperative gather-copy: a wave owns 64 consecutive slots and its 64 lanes copy
+ * the group's CONTIGUOUS output byte range in 8-byte chunks (lane l writes bytes
+ * [l*8, l*8+8) of the range, striding 512) — stores are fully coalesced; loads are
+ * within-string contiguous (neighboring lanes usually read the same string). A chunk that
+ * crosses string boundaries falls back to sub-chunk copies. One-string-per-thread copying
+ * (the first version) scattered every store instruction across 64 strings: 9.8 GB written
+ * for 1.28 GB of payload (profiles/). */
+__global__ __launch_bounds__(BLOCK_THREADS) void k4_copy(
+    const uint32_t *src_row, const uint64_t *out_off, const int32_t *in_offsets,
+    const uint8_t *in_bytes, int64_t n, uint8_t *out_bytes) {
+    __shared__ uint32_t loc_s[WAVES_PER_BLOCK][WAVE + 1];
+    __shared__ uint32_t srcb_s[WAVES_PER_BLOCK][WAVE];
+    const int wid = threadIdx.x / WAVE;
+    const int lane = threadIdx.x % WAVE;
+    uint32_t *loc = loc_s[wid];
+    uint32_t *srcb = srcb_s[wid];
+    const int64_t ngroups = (n + WAVE - 1) / WAVE;
+    const int64_t wave_id = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) / WAVE;
+    const int64_t nwaves = ((int64_t)gridDim.x * blockDim.x) / WAVE;
+
+    for (int64_t g = wave_id; g < ngroups; g += nwaves) {
+        const int64_t s0 = g * WAVE;
+        const int64_t s = s0 + lane;
+        uint32_t len = 0, sb = 0;
+        if (s < n) {
+            const uint32_t r = src_row[s];
+            const int32_t o0 = in_offsets[r];
+            len = (uint32_t)(in_offsets[r + 1] - o0);
+            sb = (uint32_t)o0;
         }
-        for (; b < len; b++) d[b] = src[b];
+        /* exclusive scan of lens across the wave -> group-local byte offsets */
+        uint32_t v = len;
+#pragma unroll
+        for (int d = 1; d < WAVE; d <<= 1) {
+            uint32_t u = (uint32_t)__shfl_up((int)v, d);
+            if (lane >= d) v += u;
+        }
+        const uint32_t T = (uint32_t)__shfl((int)v, WAVE - 1);
+        loc[lane] = v - len; /* exclusive */
+        srcb[lane] = sb;
+        if (lane == 0) loc[WAVE] = T;
+        /* single wave: LDS program order; no barrier */
+        const uint64_t obase = out_off[s0];
+        for (uint32_t p0 = lane * 8; p0 < T; p0 += WAVE * 8) {
+            uint32_t rem = (T - p0 < 8) ? (T - p0) : 8;
+            uint32_t p = p0;
+            /* binary search: largest j with loc[j] <= p */
+            int lo = 0, hi = WAVE;
+            while (lo < hi) {
+                int mid = (lo + hi + 1) >> 1;
+                if (loc[mid] <= p) lo = mid;
+                else hi = mid - 1;
+            }
+            int j = lo;
+            while (rem > 0) {
+                while (loc[j + 1] <= p) j++; /* skip empty strings */
+                const uint32_t within = p - loc[j];
+                const uint32_t avail = loc[j + 1] - p;
+                const uint32_t m = (rem < avail) ? rem : avail;
+                const uint8_t *sp = in_bytes + srcb[j] + within;
+                uint8_t *dp = out_bytes + obase + p;
+                if (m == 8) {
+                    uint64_t t;
+                    __builtin_memcpy(&t, sp, 8);
+                    __builtin_memcpy(dp, &t, 8);
+                } else {
+                    for (uint32_t b = 0; b < m; b++) dp[b] = sp[b];
+                }
+                p += m;
+                rem -= m;
+            }
+        }
     }
 }
 
