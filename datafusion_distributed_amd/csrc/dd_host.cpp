@@ -538,6 +538,159 @@ extern "C" dd_status dd_comm_init(const void *bytes128, int rank, int nranks, dd
 
 extern "C" void dd_comm_destroy(dd_comm *c) { delete c; }
 
+/* ---------------- broadcast (dd_bcast) ---------------- */
+
+struct dd_bcast {
+    int64_t n_rows = 0;
+    int32_t n_cols = 0;
+    int32_t dtypes[DD_MAX_COLS] = {};
+    int64_t data_lens[DD_MAX_COLS] = {};
+    void *data[DD_MAX_COLS] = {};
+    uint8_t *valid[DD_MAX_COLS] = {};
+    int32_t *offsets[DD_MAX_COLS] = {};
+    void *dict_bytes[DD_MAX_COLS] = {};
+    int32_t *dict_offsets[DD_MAX_COLS] = {};
+    bool owns = false; /* non-root ranks own their buffers */
+    ~dd_bcast() {
+        if (!owns) return;
+        for (int i = 0; i < DD_MAX_COLS; i++) {
+            hipFree(data[i]);
+            hipFree(valid[i]);
+            hipFree(offsets[i]);
+            hipFree(dict_bytes[i]);
+            hipFree(dict_offsets[i]);
+        }
+    }
+};
+
+/* header: [n_rows, n_cols] + per col [dtype, data_bytes, has_valid, n_offsets, dict_n,
+ * dict_bytes_len] — broadcast first so non-root ranks can allocate */
+#define DD_BHDR (2 + DD_MAX_COLS * 6)
+
+extern "C" dd_status dd_broadcast_run(dd_comm *c, const dd_batch_desc *batch, int root,
+                                      void *stream, dd_bcast **out) {
+    if (!c || !out) return set_err(DD_ERR_INVALID, "null argument");
+    hipStream_t s = (hipStream_t)stream;
+    const bool is_root = c->rank == root;
+    if (is_root && !batch) return set_err(DD_ERR_INVALID, "root needs a batch");
+
+    int64_t hdr[DD_BHDR] = {};
+    if (is_root) {
+        hdr[0] = batch->n_rows;
+        hdr[1] = batch->n_cols;
+        for (int i = 0; i < batch->n_cols; i++) {
+            const dd_col_desc &cd = batch->cols[i];
+            int64_t *h = hdr + 2 + (size_t)i * 6;
+            h[0] = cd.dtype;
+            h[1] = (cd.dtype == DD_DT_UTF8) ? cd.data_len
+                                            : batch->n_rows * fixed_elem_size(cd.dtype);
+            h[2] = cd.validity ? 1 : 0;
+            h[3] = cd.offsets ? batch->n_rows + 1 : 0;
+            h[4] = cd.dict_n;
+            h[5] = (cd.dtype == DD_DT_DICT32 && cd.dict_offsets)
+                       ? 0 /* filled below via D2H of last offset */
+                       : 0;
+        }
+        /* dict byte lengths require the last dict offset */
+        for (int i = 0; i < batch->n_cols; i++) {
+            const dd_col_desc &cd = batch->cols[i];
+            if (cd.dtype == DD_DT_DICT32 && cd.dict_n > 0) {
+                int32_t last = 0;
+                HIP_TRY(hipMemcpy(&last, cd.dict_offsets + cd.dict_n, 4,
+                                  hipMemcpyDeviceToHost));
+                hdr[2 + (size_t)i * 6 + 5] = last;
+            }
+        }
+    }
+    int64_t *d_hdr = nullptr;
+    HIP_TRY(hipMalloc(&d_hdr, sizeof(hdr)));
+    if (is_root)
+        HIP_TRY(hipMemcpyAsync(d_hdr, hdr, sizeof(hdr), hipMemcpyHostToDevice, s));
+    NCCL_TRY(ncclBroadcast(d_hdr, d_hdr, DD_BHDR, ncclInt64, root, c->comm, s));
+    HIP_TRY(hipStreamSynchronize(s));
+    HIP_TRY(hipMemcpy(hdr, d_hdr, sizeof(hdr), hipMemcpyDeviceToHost));
+    hipFree(d_hdr);
+
+    auto b = new dd_bcast();
+    b->n_rows = hdr[0];
+    b->n_cols = (int32_t)hdr[1];
+    b->owns = !is_root;
+    auto fail = [&](const char *m) {
+        delete b;
+        return set_err(DD_ERR_HIP, m);
+    };
+    for (int i = 0; i < b->n_cols; i++) {
+        const int64_t *h = hdr + 2 + (size_t)i * 6;
+        b->dtypes[i] = (int32_t)h[0];
+        b->data_lens[i] = h[1];
+        if (is_root) {
+            const dd_col_desc &cd = batch->cols[i];
+            b->data[i] = const_cast<void *>(cd.data);
+            b->valid[i] = const_cast<uint8_t *>(cd.validity);
+            b->offsets[i] = const_cast<int32_t *>(cd.offsets);
+            b->dict_bytes[i] = const_cast<void *>(cd.dict_bytes);
+            b->dict_offsets[i] = const_cast<int32_t *>(cd.dict_offsets);
+        } else {
+            if (hipMalloc(&b->data[i], (size_t)h[1] + 1) != hipSuccess)
+                return fail("bcast alloc data");
+            if (h[2] && hipMalloc((void **)&b->valid[i], (size_t)b->n_rows + 1) != hipSuccess)
+                return fail("bcast alloc valid");
+            if (h[3] && hipMalloc((void **)&b->offsets[i], (size_t)h[3] * 4) != hipSuccess)
+                return fail("bcast alloc offsets");
+            if (h[4]) {
+                if (hipMalloc(&b->dict_bytes[i], (size_t)h[5] + 1) != hipSuccess ||
+                    hipMalloc((void **)&b->dict_offsets[i], ((size_t)h[4] + 1) * 4) !=
+                        hipSuccess)
+                    return fail("bcast alloc dict");
+            }
+        }
+    }
+    NCCL_TRY(ncclGroupStart());
+    for (int i = 0; i < b->n_cols; i++) {
+        const int64_t *h = hdr + 2 + (size_t)i * 6;
+        if (h[1] > 0)
+            NCCL_TRY(ncclBroadcast(b->data[i], b->data[i], h[1], ncclUint8, root, c->comm, s));
+        if (h[2] && b->n_rows > 0)
+            NCCL_TRY(ncclBroadcast(b->valid[i], b->valid[i], b->n_rows, ncclUint8, root,
+                                   c->comm, s));
+        if (h[3])
+            NCCL_TRY(ncclBroadcast(b->offsets[i], b->offsets[i], h[3] * 4, ncclUint8, root,
+                                   c->comm, s));
+        if (h[4]) {
+            if (h[5] > 0)
+                NCCL_TRY(ncclBroadcast(b->dict_bytes[i], b->dict_bytes[i], h[5], ncclUint8,
+                                       root, c->comm, s));
+            NCCL_TRY(ncclBroadcast(b->dict_offsets[i], b->dict_offsets[i], (h[4] + 1) * 4,
+                                   ncclUint8, root, c->comm, s));
+        }
+    }
+    NCCL_TRY(ncclGroupEnd());
+    HIP_TRY(hipStreamSynchronize(s));
+    *out = b;
+    return DD_OK;
+}
+
+extern "C" void dd_bcast_destroy(dd_bcast *b) { delete b; }
+extern "C" int64_t dd_bcast_n_rows(const dd_bcast *b) { return b->n_rows; }
+extern "C" int32_t dd_bcast_n_cols(const dd_bcast *b) { return b->n_cols; }
+extern "C" const void *dd_bcast_col_data(const dd_bcast *b, int32_t col) {
+    return (col >= 0 && col < b->n_cols) ? b->data[col] : nullptr;
+}
+extern "C" const uint8_t *dd_bcast_col_validity(const dd_bcast *b, int32_t col) {
+    return (col >= 0 && col < b->n_cols) ? b->valid[col] : nullptr;
+}
+extern "C" const int32_t *dd_bcast_col_offsets(const dd_bcast *b, int32_t col) {
+    return (col >= 0 && col < b->n_cols) ? b->offsets[col] : nullptr;
+}
+extern "C" dd_status dd_bcast_col_meta(const dd_bcast *b, int32_t col, int32_t *dtype,
+                                       int64_t *data_len) {
+    if (col < 0 || col >= b->n_cols) return set_err(DD_ERR_INVALID, "col out of range");
+    *dtype = b->dtypes[col];
+    *data_len = b->data_lens[col];
+    return DD_OK;
+}
+
+
 struct dd_exchanged {
     int32_t n_cols = 0;
     int nranks = 1;

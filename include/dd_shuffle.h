@@ -147,6 +147,26 @@ dd_status dd_exchanged_byte_counts(const dd_exchanged *e, int32_t col, int64_t *
  * this rank to OTHER ranks (xGMI egress) */
 dd_status dd_exchanged_stats(const dd_exchanged *e, float *ms, int64_t *egress_bytes);
 
+/* ---------------- broadcast (build side of CollectLeft joins) ----------------
+ * Replaces BroadcastExec's cache + NetworkBroadcastExec's per-consumer fetch
+ * (src/execution_plans/broadcast.rs:24-28,163; network_broadcast.rs:245-266) with one
+ * RCCL ncclBroadcast over xGMI: the root rank's device batch is replicated on every rank.
+ * A size header is broadcast first so non-root ranks can allocate. */
+
+typedef struct dd_bcast dd_bcast; /* opaque: replicated batch */
+
+/* batch: the root's device batch (ignored on other ranks, may be NULL there) */
+dd_status dd_broadcast_run(dd_comm *c, const dd_batch_desc *batch, int root, void *stream,
+                           dd_bcast **out);
+void dd_bcast_destroy(dd_bcast *b);
+int64_t dd_bcast_n_rows(const dd_bcast *b);
+int32_t dd_bcast_n_cols(const dd_bcast *b);
+const void *dd_bcast_col_data(const dd_bcast *b, int32_t col);
+const uint8_t *dd_bcast_col_validity(const dd_bcast *b, int32_t col);
+const int32_t *dd_bcast_col_offsets(const dd_bcast *b, int32_t col);
+dd_status dd_bcast_col_meta(const dd_bcast *b, int32_t col, int32_t *dtype,
+                            int64_t *data_len);
+
 /* ---------------- task cache (worker execute path) ----------------
  * Mirrors SetPlanRequest / ExecuteTaskRequest (src/protocol/worker_channel.rs:74-93,163-176)
  * and the worker's TaskData cache (src/worker/task_data.rs:16-29,104-116;
