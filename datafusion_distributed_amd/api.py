@@ -89,6 +89,11 @@ def lib():
         L.dd_exchanged_col_validity.restype = ctypes.c_void_p
         L.dd_exchanged_col_lengths.restype = ctypes.c_void_p
         L.dd_exchanged_total_rows.restype = ctypes.c_int64
+        L.dd_bcast_n_rows.restype = ctypes.c_int64
+        L.dd_bcast_n_cols.restype = ctypes.c_int32
+        L.dd_bcast_col_data.restype = ctypes.c_void_p
+        L.dd_bcast_col_validity.restype = ctypes.c_void_p
+        L.dd_bcast_col_offsets.restype = ctypes.c_void_p
         _lib = L
     return _lib
 
@@ -252,9 +257,55 @@ class Comm:
         _check(lib().dd_exchange_run(self.h, part.h, stream, ctypes.byref(h)))
         return Exchanged(h, part, self)
 
+    def broadcast(self, batch, root=0, stream=None):
+        """Replicate the root's device batch on every rank (dd_broadcast_run: the
+        BroadcastExec/NetworkBroadcastExec data plane)."""
+        h = ctypes.c_void_p()
+        desc = ctypes.byref(batch.desc) if batch is not None else None
+        _check(lib().dd_broadcast_run(self.h, desc, root, stream, ctypes.byref(h)))
+        return Broadcasted(h)
+
     def destroy(self):
         if self.h:
             lib().dd_comm_destroy(self.h)
+            self.h = None
+
+
+class Broadcasted:
+    def __init__(self, h):
+        self.h = h
+
+    @property
+    def n_rows(self):
+        return lib().dd_bcast_n_rows(self.h)
+
+    @property
+    def n_cols(self):
+        return lib().dd_bcast_n_cols(self.h)
+
+    def col(self, i):
+        dtype = ctypes.c_int32()
+        dlen = ctypes.c_int64()
+        _check(lib().dd_bcast_col_meta(self.h, i, ctypes.byref(dtype),
+                                       ctypes.byref(dlen)))
+        name = {v: k for k, v in DTYPE_CODE.items()}[dtype.value]
+        n = self.n_rows
+        out = {"dtype": name}
+        if name == "utf8":
+            out["data"] = _d2h(lib().dd_bcast_col_data(self.h, i), int(dlen.value), np.uint8)
+            out["offsets"] = _d2h(lib().dd_bcast_col_offsets(self.h, i), (n + 1) * 4,
+                                  np.int32)
+        else:
+            out["data"] = _d2h(lib().dd_bcast_col_data(self.h, i), int(dlen.value),
+                               FIXED_NP[name])
+        vp = lib().dd_bcast_col_validity(self.h, i)
+        if vp:
+            out["valid"] = _d2h(vp, n, np.uint8)
+        return out
+
+    def destroy(self):
+        if self.h:
+            lib().dd_bcast_destroy(self.h)
             self.h = None
 
 
