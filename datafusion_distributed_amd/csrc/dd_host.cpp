@@ -129,24 +129,24 @@ struct dd_partitioner {
     bool has_run = false;
 
     ~dd_partitioner() {
-        hipFree(pid);
-        hipFree(counts);
-        hipFree(partials);
-        hipFree(part_offsets);
-        hipFree(bcounts);
-        hipFree(bpartials);
-        hipFree(part_boffsets);
-        hipFree(src_row);
+        (void)hipFree(pid);
+        (void)hipFree(counts);
+        (void)hipFree(partials);
+        (void)hipFree(part_offsets);
+        (void)hipFree(bcounts);
+        (void)hipFree(bpartials);
+        (void)hipFree(part_boffsets);
+        (void)hipFree(src_row);
         for (auto &o : out_off) hipFree(o);
-        hipFree(k4_partials);
+        (void)hipFree(k4_partials);
         for (int i = 0; i < DD_KMAX_COLS; i++) {
-            hipFree(dict_hashes[i]);
-            hipFree(out_data[i]);
-            hipFree(out_valid[i]);
-            hipFree(out_lengths[i]);
+            (void)hipFree(dict_hashes[i]);
+            (void)hipFree(out_data[i]);
+            (void)hipFree(out_valid[i]);
+            (void)hipFree(out_lengths[i]);
         }
         for (auto &e : ev)
-            if (e) hipEventDestroy(e);
+            if (e) (void)hipEventDestroy(e);
     }
 };
 
@@ -258,6 +258,10 @@ extern "C" dd_status dd_partitioner_create(const dd_batch_desc *batch, const int
         int64_t nblocks = (n + 16383) / 16384;
         if (nblocks < 8) nblocks = 8;
         if (nblocks > 2048) nblocks = 2048;
+        if (const char *e = getenv("DD_V2_BLOCKS")) { /* tile-count experiment knob */
+            int64_t v = atoll(e);
+            if (v >= 8 && v <= 8192) nblocks = v;
+        }
         p->nchunks = nblocks;
         p->chunk_rows = (n + nblocks - 1) / nblocks;
         if (p->chunk_rows < 1) p->chunk_rows = 1;
@@ -554,11 +558,11 @@ struct dd_bcast {
     ~dd_bcast() {
         if (!owns) return;
         for (int i = 0; i < DD_MAX_COLS; i++) {
-            hipFree(data[i]);
-            hipFree(valid[i]);
-            hipFree(offsets[i]);
-            hipFree(dict_bytes[i]);
-            hipFree(dict_offsets[i]);
+            (void)hipFree(data[i]);
+            (void)hipFree(valid[i]);
+            (void)hipFree(offsets[i]);
+            (void)hipFree(dict_bytes[i]);
+            (void)hipFree(dict_offsets[i]);
         }
     }
 };
@@ -609,7 +613,7 @@ extern "C" dd_status dd_broadcast_run(dd_comm *c, const dd_batch_desc *batch, in
     NCCL_TRY(ncclBroadcast(d_hdr, d_hdr, DD_BHDR, ncclInt64, root, c->comm, s));
     HIP_TRY(hipStreamSynchronize(s));
     HIP_TRY(hipMemcpy(hdr, d_hdr, sizeof(hdr), hipMemcpyDeviceToHost));
-    hipFree(d_hdr);
+    (void)hipFree(d_hdr);
 
     auto b = new dd_bcast();
     b->n_rows = hdr[0];
@@ -706,12 +710,12 @@ struct dd_exchanged {
     hipEvent_t e0 = nullptr, e1 = nullptr;
     ~dd_exchanged() {
         for (int i = 0; i < DD_MAX_COLS; i++) {
-            hipFree(data[i]);
-            hipFree(valid[i]);
-            hipFree(lengths[i]);
+            (void)hipFree(data[i]);
+            (void)hipFree(valid[i]);
+            (void)hipFree(lengths[i]);
         }
-        if (e0) hipEventDestroy(e0);
-        if (e1) hipEventDestroy(e1);
+        if (e0) (void)hipEventDestroy(e0);
+        if (e1) (void)hipEventDestroy(e1);
     }
 };
 
@@ -751,8 +755,8 @@ extern "C" dd_status dd_exchange_run(dd_comm *c, const dd_partitioner *p, void *
     HIP_TRY(hipStreamSynchronize(s));
     HIP_TRY(hipMemcpy(all_meta.data(), d_meta_all, (size_t)R * meta_n * 8,
                       hipMemcpyDeviceToHost));
-    hipFree(d_meta_in);
-    hipFree(d_meta_all);
+    (void)hipFree(d_meta_in);
+    (void)hipFree(d_meta_all);
 
     auto e = new dd_exchanged();
     e->n_cols = p->batch.n_cols;
