@@ -184,8 +184,12 @@ def main():
         import torch.distributed as dist
 
         os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
-        dist.init_process_group("nccl", rank=rank, world_size=world)
-        torch.cuda.set_device(local_rank)
+        ndev = torch.cuda.device_count()
+        # gloo for the control plane when GPUs are oversubscribed (testing); the data
+        # plane is always our own RCCL comm (exchange.create_comm)
+        backend = "nccl" if ndev >= world else "gloo"
+        dist.init_process_group(backend, rank=rank, world_size=world)
+        torch.cuda.set_device(local_rank % max(ndev, 1))
     elif torch.cuda.is_available():
         torch.cuda.set_device(0)
 
@@ -242,8 +246,9 @@ def main():
     if world > 1:
         import torch.distributed as dist
 
-        t = torch.tensor([elapsed], dtype=torch.float64,
-                         device="cuda" if torch.cuda.is_available() else "cpu")
+        dev = ("cuda" if (torch.cuda.is_available() and
+                          dist.get_backend() == "nccl") else "cpu")
+        t = torch.tensor([elapsed], dtype=torch.float64, device=dev)
         dist.all_reduce(t, op=dist.ReduceOp.MAX)
         elapsed = float(t.item())
 
