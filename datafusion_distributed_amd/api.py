@@ -94,6 +94,7 @@ def lib():
         L.dd_bcast_col_data.restype = ctypes.c_void_p
         L.dd_bcast_col_validity.restype = ctypes.c_void_p
         L.dd_bcast_col_offsets.restype = ctypes.c_void_p
+        L.dd_reducer_n_rows.restype = ctypes.c_int64
         _lib = L
     return _lib
 
@@ -360,3 +361,37 @@ class Exchanged:
         if self.h:
             lib().dd_exchanged_destroy(self.h)
             self.h = None
+
+
+AGG_OPS = {"sum_f64": 0, "count": 1, "sum_i64": 2}
+
+
+def partial_reduce(batch: DeviceBatch, key_idx, aggs):
+    """GPU partial aggregation below the shuffle (dd_partial_reduce_run).
+
+    aggs: list of (col_index_or_None, op) with op in {"sum_f64", "count", "sum_i64"}.
+    Returns {"keys": u64[n, nk] canonical bits, "keynull": u32[n] per-key null bitmask,
+    "aggs": f64[n, na]} — integer aggregates (count / sum_i64) are bit-cast in the f64
+    slots (view with .view(np.int64)). May contain duplicate groups (partial semantics).
+    """
+    nk = len(key_idx)
+    na = len(aggs)
+    keysc = (ctypes.c_int32 * nk)(*key_idx)
+    aggc = (ctypes.c_int32 * na)(*[c if c is not None else 0 for c, _ in aggs])
+    opsc = (ctypes.c_int32 * na)(*[AGG_OPS[o] for _, o in aggs])
+    h = ctypes.c_void_p()
+    _check(lib().dd_partial_reduce_run(ctypes.byref(batch.desc), keysc, nk, aggc, opsc,
+                                       na, None, ctypes.byref(h)))
+    try:
+        n = lib().dd_reducer_n_rows(h)
+        keys = np.empty((n, nk), dtype=np.uint64)
+        keynull = np.empty(n, dtype=np.uint32)
+        vals = np.empty((n, na), dtype=np.float64)
+        if n:
+            _check(lib().dd_reducer_fetch(
+                h, keys.ctypes.data_as(ctypes.c_void_p),
+                keynull.ctypes.data_as(ctypes.c_void_p),
+                vals.ctypes.data_as(ctypes.c_void_p)))
+        return {"keys": keys, "keynull": keynull, "aggs": vals}
+    finally:
+        lib().dd_reducer_destroy(h)

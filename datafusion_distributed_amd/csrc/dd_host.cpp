@@ -907,3 +907,123 @@ extern "C" dd_status dd_exchanged_stats(const dd_exchanged *e, float *ms,
     *egress_bytes = e->egress;
     return DD_OK;
 }
+
+/* ---------------- partial aggregation (dd_reducer) ---------------- */
+
+struct dd_reducer {
+    int32_t n_keys = 0, n_aggs = 0;
+    int64_t n_rows = 0; /* output rows */
+    uint64_t *keys = nullptr;
+    uint32_t *keynull = nullptr;
+    double *aggs = nullptr;
+    uint64_t *n_dev = nullptr;
+    int32_t *meta = nullptr; /* device: agg_cols + agg_ops */
+    ~dd_reducer() {
+        (void)hipFree(keys);
+        (void)hipFree(keynull);
+        (void)hipFree(aggs);
+        (void)hipFree(n_dev);
+        (void)hipFree(meta);
+    }
+};
+
+extern "C" dd_status dd_partial_reduce_run(const dd_batch_desc *batch,
+                                           const int32_t *key_cols, int32_t n_keys,
+                                           const int32_t *agg_cols, const int32_t *agg_ops,
+                                           int32_t n_aggs, void *stream, dd_reducer **out) {
+    if (!batch || !key_cols || !agg_ops || !out)
+        return set_err(DD_ERR_INVALID, "null argument");
+    if (dd_device_count() == 0) return set_err(DD_ERR_NO_DEVICE, "no HIP device");
+    if (n_keys < 1 || n_keys > 4) return set_err(DD_ERR_UNSUPPORTED, "1..4 key columns");
+    if (n_aggs < 1 || n_aggs > 4) return set_err(DD_ERR_UNSUPPORTED, "1..4 aggregates");
+    for (int k = 0; k < n_keys; k++) {
+        if (key_cols[k] < 0 || key_cols[k] >= batch->n_cols)
+            return set_err(DD_ERR_INVALID, "key column out of range");
+        if (batch->cols[key_cols[k]].dtype == DD_DT_UTF8 ||
+            batch->cols[key_cols[k]].dtype == DD_DT_DICT32)
+            return set_err(DD_ERR_UNSUPPORTED, "partial reduce: fixed-width keys only");
+    }
+    for (int g = 0; g < n_aggs; g++) {
+        if (agg_ops[g] == DD_AGG_COUNT) continue;
+        if (!agg_cols || agg_cols[g] < 0 || agg_cols[g] >= batch->n_cols)
+            return set_err(DD_ERR_INVALID, "agg column out of range");
+        int dt = batch->cols[agg_cols[g]].dtype;
+        if (agg_ops[g] == DD_AGG_SUM_F64 && dt != DD_DT_F64)
+            return set_err(DD_ERR_UNSUPPORTED, "SUM_F64 needs an f64 column");
+        if (agg_ops[g] == DD_AGG_SUM_I64 && dt != DD_DT_I64)
+            return set_err(DD_ERR_UNSUPPORTED, "SUM_I64 needs an i64 column");
+    }
+
+    dd_kargs ka;
+    memset(&ka, 0, sizeof(ka));
+    const int64_t n = batch->n_rows;
+    ka.n_rows = n;
+    ka.n_cols = batch->n_cols;
+    ka.n_keys = n_keys;
+    for (int k = 0; k < n_keys; k++) ka.key_idx[k] = key_cols[k];
+    for (int c = 0; c < batch->n_cols; c++) {
+        const dd_col_desc &cd = batch->cols[c];
+        dd_kcol &kc = ka.cols[c];
+        kc.dtype = cd.dtype;
+        kc.elem = fixed_elem_size(cd.dtype);
+        kc.data = cd.data;
+        kc.valid = cd.validity;
+        kc.offsets = cd.offsets;
+    }
+
+    int64_t nblocks = (n + 16383) / 16384;
+    if (nblocks < 8) nblocks = 8;
+    if (nblocks > 1024) nblocks = 1024;
+    const int64_t chunk = (n + nblocks - 1) / nblocks;
+    const int64_t max_rows = n + nblocks * 1024 + 1; /* spill worst case + table flush */
+
+    auto r = new dd_reducer();
+    r->n_keys = n_keys;
+    r->n_aggs = n_aggs;
+    auto fail = [&](const char *m) {
+        delete r;
+        return set_err(DD_ERR_HIP, m);
+    };
+    if (hipMalloc((void **)&r->keys, (size_t)max_rows * n_keys * 8) != hipSuccess ||
+        hipMalloc((void **)&r->keynull, (size_t)max_rows * 4) != hipSuccess ||
+        hipMalloc((void **)&r->aggs, (size_t)max_rows * n_aggs * 8) != hipSuccess ||
+        hipMalloc((void **)&r->n_dev, 8) != hipSuccess ||
+        hipMalloc((void **)&r->meta, 2 * 4 * 4) != hipSuccess)
+        return fail("partial reduce alloc");
+    hipStream_t s = (hipStream_t)stream;
+    if (hipMemsetAsync(r->n_dev, 0, 8, s) != hipSuccess) return fail("memset");
+    int32_t meta_h[8] = {};
+    for (int g = 0; g < n_aggs; g++) {
+        meta_h[g] = agg_cols ? agg_cols[g] : 0;
+        meta_h[4 + g] = agg_ops[g];
+    }
+    if (hipMemcpyAsync(r->meta, meta_h, sizeof(meta_h), hipMemcpyHostToDevice, s) !=
+        hipSuccess)
+        return fail("meta upload");
+    hipError_t e = dd_launch_partial_reduce(&ka, nblocks, chunk, n_aggs, r->meta,
+                                            r->meta + 4, r->keys, r->keynull, r->aggs,
+                                            r->n_dev, s);
+    if (e != hipSuccess) return fail(hipGetErrorString(e));
+    if (hipStreamSynchronize(s) != hipSuccess) return fail("sync");
+    uint64_t n_out = 0;
+    if (hipMemcpy(&n_out, r->n_dev, 8, hipMemcpyDeviceToHost) != hipSuccess)
+        return fail("n_out copy");
+    r->n_rows = (int64_t)n_out;
+    *out = r;
+    return DD_OK;
+}
+
+extern "C" int64_t dd_reducer_n_rows(const dd_reducer *r) { return r->n_rows; }
+
+extern "C" dd_status dd_reducer_fetch(const dd_reducer *r, uint64_t *host_keys,
+                                      uint32_t *host_keynull, double *host_aggs) {
+    HIP_TRY(hipMemcpy(host_keys, r->keys, (size_t)r->n_rows * r->n_keys * 8,
+                      hipMemcpyDeviceToHost));
+    HIP_TRY(hipMemcpy(host_keynull, r->keynull, (size_t)r->n_rows * 4,
+                      hipMemcpyDeviceToHost));
+    HIP_TRY(hipMemcpy(host_aggs, r->aggs, (size_t)r->n_rows * r->n_aggs * 8,
+                      hipMemcpyDeviceToHost));
+    return DD_OK;
+}
+
+extern "C" void dd_reducer_destroy(dd_reducer *r) { delete r; }

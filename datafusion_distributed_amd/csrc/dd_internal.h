@@ -78,6 +78,11 @@ hipError_t dd_launch_var_bytes(const uint32_t *lens, const uint32_t *src_row,
                                uint64_t *out_off, uint8_t *out_bytes,
                                const uint64_t *part_offsets, uint32_t nparts,
                                uint64_t *part_boffsets, hipStream_t s);
+hipError_t dd_launch_partial_reduce(const dd_kargs *a, int64_t nblocks, int64_t chunk_rows,
+                                    int n_aggs, const int32_t *agg_cols,
+                                    const int32_t *agg_ops, uint64_t *out_keys,
+                                    uint32_t *out_keynull, double *out_aggs,
+                                    uint64_t *out_n, hipStream_t s);
 }
 
 #endif

@@ -1,0 +1,207 @@
+/* dd_reduce.hip — GPU partial aggregation below the shuffle.
+ *
+ * Mirrors the reference's partial-reduce pass (/root/reference/src/distributed_planner/
+ * partial_reduce_below_network_shuffles.rs, enabled by `distributed.partial_reduce`,
+ * distributed_config.rs:50-54): an AggregateExec mode=Partial is placed BELOW the
+ * repartition so the exchange moves per-group partials instead of raw rows. Semantics are
+ * DataFusion's partial aggregate: each invocation may emit DUPLICATE groups (one per
+ * block here, one per input partition there); the downstream final aggregate merges them.
+ * Output row order is unspecified (the reference's is stream-order; both are merged by
+ * the order-insensitive final aggregate — test_utils/property_based.rs:15-41).
+ *
+ * Design: per-block LDS open-addressing table keyed by the normative row hash
+ * (dd_row_hash — the same hash the shuffle uses). Claim a slot with an LDS CAS on the
+ * hash word, publish key values behind a ready flag, aggregate with LDS atomics
+ * (f64/u64 add). Rows that do not fit after a bounded probe spill to the output as
+ * singleton groups (sum = value, count = 1) — correct for any cardinality, fast for the
+ * low-cardinality keys partial-reduce targets (q1: 6 groups).
+ *
+ * Round-1 coverage: fixed-width key columns (nullable), aggregate ops
+ * sum(f64) / sum(i64) / count(*). */
+
+#include <hip/hip_runtime.h>
+#include <stdint.h>
+
+#include "dd_hash_device.h"
+#include "dd_internal.h"
+
+#define RWAVE 64
+#define R_THREADS 256
+#define R_CAP 1024      /* table slots per block */
+#define R_PROBE 64      /* bounded probe; then spill */
+
+/* agg op codes (mirrored in include/dd_shuffle.h) */
+enum { DD_AGG_SUM_F64 = 0, DD_AGG_COUNT = 1, DD_AGG_SUM_I64 = 2 };
+
+/* canonical 64-bit key bits (zero-extended; floats canonicalized like the hash) */
+__device__ __forceinline__ uint64_t dd_key_bits(const dd_kcol &c, int64_t i) {
+    switch (c.dtype) {
+    case DD_KDT_U8:
+    case DD_KDT_BOOL:
+        return (uint64_t)((const uint8_t *)c.data)[i];
+    case DD_KDT_I16:
+        return (uint64_t)((const uint16_t *)c.data)[i];
+    case DD_KDT_I32:
+        return (uint64_t)((const uint32_t *)c.data)[i];
+    case DD_KDT_I64:
+        return ((const uint64_t *)c.data)[i];
+    case DD_KDT_F32: {
+        float v = ((const float *)c.data)[i];
+        if (v == 0.0f) v = 0.0f;
+        uint32_t b = __float_as_uint(v);
+        if (v != v) b = 0x7fc00000u;
+        return (uint64_t)b;
+    }
+    case DD_KDT_F64: {
+        double v = ((const double *)c.data)[i];
+        if (v == 0.0) v = 0.0;
+        uint64_t b = __double_as_longlong(v);
+        if (v != v) b = 0x7ff8000000000000ULL;
+        return b;
+    }
+    default:
+        return 0;
+    }
+}
+
+extern "C" __global__ __launch_bounds__(R_THREADS) void k_partial_reduce(
+    dd_kargs a, int64_t chunk_rows, int n_aggs, const int32_t *agg_cols,
+    const int32_t *agg_ops, uint64_t *out_keys /* [max_rows][n_keys] interleaved */,
+    uint32_t *out_keynull /* [max_rows] bitmask */,
+    double *out_aggs /* [max_rows][n_aggs] interleaved (i64 sums bit-cast) */,
+    uint64_t *out_n /* global row counter */) {
+    __shared__ uint64_t t_hash[R_CAP];
+    __shared__ uint32_t t_ready[R_CAP];
+    __shared__ uint64_t t_keys[DD_KMAX_KEYS > 4 ? 4 : DD_KMAX_KEYS][R_CAP];
+    __shared__ uint32_t t_null[R_CAP];
+    __shared__ unsigned long long t_agg[4][R_CAP]; /* f64 or i64 state, bit pattern */
+
+    const int nk = a.n_keys > 4 ? 4 : a.n_keys;
+    for (int s = threadIdx.x; s < R_CAP; s += R_THREADS) {
+        t_hash[s] = 0;
+        t_ready[s] = 0;
+        t_null[s] = 0;
+        for (int k = 0; k < nk; k++) t_keys[k][s] = 0;
+        for (int g = 0; g < n_aggs; g++)
+            t_agg[g][s] = (agg_ops[g] == DD_AGG_SUM_F64) ? __double_as_longlong(0.0) : 0ull;
+    }
+    __syncthreads();
+
+    const int64_t start = (int64_t)blockIdx.x * chunk_rows;
+    const int64_t end = (start + chunk_rows < a.n_rows) ? (start + chunk_rows) : a.n_rows;
+
+    for (int64_t row = start + threadIdx.x; row < end; row += R_THREADS) {
+        uint64_t h = dd_row_hash(a, row);
+        if (h == 0) h = 1; /* 0 is the empty sentinel */
+        uint64_t kb[4];
+        uint32_t knull = 0;
+        for (int k = 0; k < nk; k++) {
+            const dd_kcol &c = a.cols[a.key_idx[k]];
+            const bool isnull = c.valid && !c.valid[row];
+            if (isnull) knull |= 1u << k;
+            kb[k] = isnull ? 0 : dd_key_bits(c, row);
+        }
+        int slot = -1;
+        uint32_t s = (uint32_t)(h % R_CAP);
+        for (int probe = 0; probe < R_PROBE; probe++, s = (s + 1) % R_CAP) {
+            uint64_t cur = atomicCAS((unsigned long long *)&t_hash[s], 0ull,
+                                     (unsigned long long)h);
+            if (cur == 0) { /* claimed: publish keys, then ready */
+                for (int k = 0; k < nk; k++) t_keys[k][s] = kb[k];
+                t_null[s] = knull;
+                __threadfence_block();
+                atomicExch(&t_ready[s], 1u);
+                slot = (int)s;
+                break;
+            }
+            if (cur == h) {
+                /* NEVER spin on the publisher: the claimer may be a divergent lane of
+                 * THIS wave (no independent thread scheduling on CDNA — a spin would
+                 * deadlock the wave). Unpublished -> keep probing; worst case the row
+                 * spills as a duplicate group, which partial aggregation permits. */
+                if (atomicAdd(&t_ready[s], 0u) != 0) {
+                    __threadfence_block();
+                    bool eq = t_null[s] == knull;
+                    for (int k = 0; k < nk && eq; k++) eq = t_keys[k][s] == kb[k];
+                    if (eq) {
+                        slot = (int)s;
+                        break;
+                    }
+                }
+            }
+            /* different group (or unpublished slot): keep probing */
+        }
+        if (slot >= 0) {
+            for (int g = 0; g < n_aggs; g++) {
+                switch (agg_ops[g]) {
+                case DD_AGG_SUM_F64: {
+                    const dd_kcol &c = a.cols[agg_cols[g]];
+                    if (c.valid && !c.valid[row]) break;
+                    atomicAdd((double *)&t_agg[g][slot], ((const double *)c.data)[row]);
+                    break;
+                }
+                case DD_AGG_SUM_I64: {
+                    const dd_kcol &c = a.cols[agg_cols[g]];
+                    if (c.valid && !c.valid[row]) break;
+                    atomicAdd(&t_agg[g][slot],
+                              (unsigned long long)((const uint64_t *)c.data)[row]);
+                    break;
+                }
+                case DD_AGG_COUNT:
+                    atomicAdd(&t_agg[g][slot], 1ull);
+                    break;
+                }
+            }
+        } else {
+            /* spill: emit a singleton group */
+            uint64_t o = atomicAdd((unsigned long long *)out_n, 1ull);
+            for (int k = 0; k < nk; k++) out_keys[o * nk + k] = kb[k];
+            out_keynull[o] = knull;
+            for (int g = 0; g < n_aggs; g++) {
+                double v = 0;
+                switch (agg_ops[g]) {
+                case DD_AGG_SUM_F64: {
+                    const dd_kcol &c = a.cols[agg_cols[g]];
+                    v = (c.valid && !c.valid[row]) ? 0.0 : ((const double *)c.data)[row];
+                    out_aggs[o * n_aggs + g] = v;
+                    break;
+                }
+                case DD_AGG_SUM_I64: {
+                    const dd_kcol &c = a.cols[agg_cols[g]];
+                    uint64_t iv = (c.valid && !c.valid[row])
+                                      ? 0
+                                      : ((const uint64_t *)c.data)[row];
+                    out_aggs[o * n_aggs + g] = __longlong_as_double((long long)iv);
+                    break;
+                }
+                case DD_AGG_COUNT:
+                    out_aggs[o * n_aggs + g] = __longlong_as_double(1ll);
+                    break;
+                }
+            }
+        }
+    }
+    __syncthreads();
+
+    /* flush occupied slots */
+    for (int s = threadIdx.x; s < R_CAP; s += R_THREADS) {
+        if (t_hash[s] == 0) continue;
+        uint64_t o = atomicAdd((unsigned long long *)out_n, 1ull);
+        for (int k = 0; k < nk; k++) out_keys[o * nk + k] = t_keys[k][s];
+        out_keynull[o] = t_null[s];
+        for (int g = 0; g < n_aggs; g++)
+            out_aggs[o * n_aggs + g] = __longlong_as_double((long long)t_agg[g][s]);
+    }
+}
+
+extern "C" hipError_t dd_launch_partial_reduce(const dd_kargs *a, int64_t nblocks,
+                                               int64_t chunk_rows, int n_aggs,
+                                               const int32_t *agg_cols,
+                                               const int32_t *agg_ops, uint64_t *out_keys,
+                                               uint32_t *out_keynull, double *out_aggs,
+                                               uint64_t *out_n, hipStream_t s) {
+    hipLaunchKernelGGL(k_partial_reduce, dim3((unsigned)nblocks), dim3(R_THREADS), 0, s, *a,
+                       chunk_rows, n_aggs, agg_cols, agg_ops, out_keys, out_keynull,
+                       out_aggs, out_n);
+    return hipGetLastError();
+}

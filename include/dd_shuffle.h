@@ -190,6 +190,30 @@ dd_status dd_execute_task(const dd_task_key *key, uint32_t part_lo, uint32_t par
                           void *stream, dd_partitioner **out);
 dd_status dd_drop_task(const dd_task_key *key); /* task cleanup (stateful_data_cleanup) */
 
+/* ---------------- partial aggregation (below the shuffle) ----------------
+ * Mirrors the partial-reduce pass (src/distributed_planner/
+ * partial_reduce_below_network_shuffles.rs; `distributed.partial_reduce`,
+ * distributed_config.rs:50-54): a mode=Partial aggregate run before dd_partition so the
+ * exchange moves per-group partials. May emit duplicate groups (one per block); the
+ * downstream final aggregate merges them. Fixed-width keys (<=4), aggregate ops below. */
+
+typedef struct dd_reducer dd_reducer;
+
+#define DD_AGG_SUM_F64 0 /* sum of an f64 column (nulls skipped) */
+#define DD_AGG_COUNT 1   /* count(*) (agg_col ignored) */
+#define DD_AGG_SUM_I64 2 /* sum of an i64 column (nulls skipped) */
+
+dd_status dd_partial_reduce_run(const dd_batch_desc *batch, const int32_t *key_cols,
+                                int32_t n_keys, const int32_t *agg_cols,
+                                const int32_t *agg_ops, int32_t n_aggs, void *stream,
+                                dd_reducer **out);
+int64_t dd_reducer_n_rows(const dd_reducer *r);
+/* host_keys[n][n_keys] canonical 64-bit key bits; host_keynull[n] per-key null bitmask;
+ * host_aggs[n][n_aggs] (i64 sums / counts bit-cast into the double slot) */
+dd_status dd_reducer_fetch(const dd_reducer *r, uint64_t *host_keys, uint32_t *host_keynull,
+                           double *host_aggs);
+void dd_reducer_destroy(dd_reducer *r);
+
 /* ---------------- device helpers (harness convenience; not part of the seam) ------- */
 dd_status dd_dev_alloc(int64_t bytes, void **out);
 dd_status dd_dev_free(void *p);
