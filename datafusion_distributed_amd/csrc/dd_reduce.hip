@@ -115,11 +115,16 @@ extern "C" __global__ __launch_bounds__(R_THREADS) void k_partial_reduce(
                 break;
             }
             if (cur == h) {
-                /* NEVER spin on the publisher: the claimer may be a divergent lane of
-                 * THIS wave (no independent thread scheduling on CDNA — a spin would
-                 * deadlock the wave). Unpublished -> keep probing; worst case the row
-                 * spills as a duplicate group, which partial aggregation permits. */
-                if (atomicAdd(&t_ready[s], 0u) != 0) {
+                /* NEVER spin unboundedly on the publisher: the claimer may be a divergent
+                 * lane of THIS wave (no independent thread scheduling on CDNA — an
+                 * unbounded spin would deadlock the wave). A BOUNDED recheck drains the
+                 * common race (claimer in another wave); if still unpublished, keep
+                 * probing — worst case the row lands as a duplicate group, which partial
+                 * aggregation permits. */
+                uint32_t rdy = 0;
+                for (int retry = 0; retry < 16 && !rdy; retry++)
+                    rdy = atomicAdd(&t_ready[s], 0u);
+                if (rdy) {
                     __threadfence_block();
                     bool eq = t_null[s] == knull;
                     for (int k = 0; k < nk && eq; k++) eq = t_keys[k][s] == kb[k];

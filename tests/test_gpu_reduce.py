@@ -56,8 +56,9 @@ def test_partial_reduce_q1_low_cardinality():
     res = api.partial_reduce(batch, [0, 1], [(2, "sum_f64"), (3, "sum_f64"),
                                              (None, "count")])
     batch.free()
-    # partial output is tiny: <= blocks * 6 groups (plus none spilled)
-    assert len(res["keynull"]) <= 1024 * 8
+    # partial output is small: 6 true groups per block plus bounded claim-race
+    # duplicates (dd_reduce.hip probe notes); far below the input row count
+    assert len(res["keynull"]) <= 64 * 64
     merged = merge_partials_mixed(res, int_ops={2})
 
     tbl = pa.table({"rf": rf, "ls": ls, "qty": qty, "price": price})
