@@ -102,8 +102,16 @@ extern "C" __global__ __launch_bounds__(R_THREADS) void k_partial_reduce(
             kb[k] = isnull ? 0 : dd_key_bits(c, row);
         }
         int slot = -1;
+        /* Two probe attempts. Within one divergent loop iteration a follower lane may not
+         * observe its own wave's claimer publish (no independent thread scheduling, and
+         * masked branch bodies need not interleave); after the first attempt's loop
+         * reconverges, every publish from this wave IS visible, so attempt 2 matches
+         * deterministically instead of claiming a duplicate slot per lane. */
+        for (int attempt = 0; attempt < 2 && slot < 0; attempt++) {
         uint32_t s = (uint32_t)(h % R_CAP);
         for (int probe = 0; probe < R_PROBE; probe++, s = (s + 1) % R_CAP) {
+            if (attempt == 0 && probe > 0) break; /* attempt 1: claim-or-match at home slot
+                                                     only; defer the walk to attempt 2 */
             uint64_t cur = atomicCAS((unsigned long long *)&t_hash[s], 0ull,
                                      (unsigned long long)h);
             if (cur == 0) { /* claimed: publish keys, then ready */
@@ -121,9 +129,7 @@ extern "C" __global__ __launch_bounds__(R_THREADS) void k_partial_reduce(
                  * common race (claimer in another wave); if still unpublished, keep
                  * probing — worst case the row lands as a duplicate group, which partial
                  * aggregation permits. */
-                uint32_t rdy = 0;
-                for (int retry = 0; retry < 16 && !rdy; retry++)
-                    rdy = atomicAdd(&t_ready[s], 0u);
+                uint32_t rdy = atomicAdd(&t_ready[s], 0u);
                 if (rdy) {
                     __threadfence_block();
                     bool eq = t_null[s] == knull;
@@ -135,6 +141,7 @@ extern "C" __global__ __launch_bounds__(R_THREADS) void k_partial_reduce(
                 }
             }
             /* different group (or unpublished slot): keep probing */
+        }
         }
         if (slot >= 0) {
             for (int g = 0; g < n_aggs; g++) {
