@@ -95,6 +95,7 @@ def lib():
         L.dd_bcast_col_validity.restype = ctypes.c_void_p
         L.dd_bcast_col_offsets.restype = ctypes.c_void_p
         L.dd_reducer_n_rows.restype = ctypes.c_int64
+        L.dd_reducer_kernel_ms.restype = ctypes.c_float
         _lib = L
     return _lib
 
@@ -383,6 +384,7 @@ def partial_reduce(batch: DeviceBatch, key_idx, aggs):
     _check(lib().dd_partial_reduce_run(ctypes.byref(batch.desc), keysc, nk, aggc, opsc,
                                        na, None, ctypes.byref(h)))
     try:
+        kernel_ms = float(lib().dd_reducer_kernel_ms(h))
         n = lib().dd_reducer_n_rows(h)
         keys = np.empty((n, nk), dtype=np.uint64)
         keynull = np.empty(n, dtype=np.uint32)
@@ -392,6 +394,6 @@ def partial_reduce(batch: DeviceBatch, key_idx, aggs):
                 h, keys.ctypes.data_as(ctypes.c_void_p),
                 keynull.ctypes.data_as(ctypes.c_void_p),
                 vals.ctypes.data_as(ctypes.c_void_p)))
-        return {"keys": keys, "keynull": keynull, "aggs": vals}
+        return {"keys": keys, "keynull": keynull, "aggs": vals, "kernel_ms": kernel_ms}
     finally:
         lib().dd_reducer_destroy(h)

@@ -918,6 +918,7 @@ struct dd_reducer {
     double *aggs = nullptr;
     uint64_t *n_dev = nullptr;
     int32_t *meta = nullptr; /* device: agg_cols + agg_ops */
+    float kernel_ms = 0;     /* hipEvent time of the reduce kernel alone */
     ~dd_reducer() {
         (void)hipFree(keys);
         (void)hipFree(keynull);
@@ -1000,11 +1001,19 @@ extern "C" dd_status dd_partial_reduce_run(const dd_batch_desc *batch,
     if (hipMemcpyAsync(r->meta, meta_h, sizeof(meta_h), hipMemcpyHostToDevice, s) !=
         hipSuccess)
         return fail("meta upload");
+    hipEvent_t e0 = nullptr, e1 = nullptr;
+    (void)hipEventCreate(&e0);
+    (void)hipEventCreate(&e1);
+    (void)hipEventRecord(e0, s);
     hipError_t e = dd_launch_partial_reduce(&ka, nblocks, chunk, n_aggs, r->meta,
                                             r->meta + 4, r->keys, r->keynull, r->aggs,
                                             r->n_dev, s);
+    (void)hipEventRecord(e1, s);
     if (e != hipSuccess) return fail(hipGetErrorString(e));
     if (hipStreamSynchronize(s) != hipSuccess) return fail("sync");
+    (void)hipEventElapsedTime(&r->kernel_ms, e0, e1);
+    (void)hipEventDestroy(e0);
+    (void)hipEventDestroy(e1);
     uint64_t n_out = 0;
     if (hipMemcpy(&n_out, r->n_dev, 8, hipMemcpyDeviceToHost) != hipSuccess)
         return fail("n_out copy");
@@ -1014,6 +1023,7 @@ extern "C" dd_status dd_partial_reduce_run(const dd_batch_desc *batch,
 }
 
 extern "C" int64_t dd_reducer_n_rows(const dd_reducer *r) { return r->n_rows; }
+extern "C" float dd_reducer_kernel_ms(const dd_reducer *r) { return r->kernel_ms; }
 
 extern "C" dd_status dd_reducer_fetch(const dd_reducer *r, uint64_t *host_keys,
                                       uint32_t *host_keynull, double *host_aggs) {

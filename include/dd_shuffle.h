@@ -208,6 +208,7 @@ dd_status dd_partial_reduce_run(const dd_batch_desc *batch, const int32_t *key_c
                                 const int32_t *agg_ops, int32_t n_aggs, void *stream,
                                 dd_reducer **out);
 int64_t dd_reducer_n_rows(const dd_reducer *r);
+float dd_reducer_kernel_ms(const dd_reducer *r); /* reduce-kernel time of the last run */
 /* host_keys[n][n_keys] canonical 64-bit key bits; host_keynull[n] per-key null bitmask;
  * host_aggs[n][n_aggs] (i64 sums / counts bit-cast into the double slot) */
 dd_status dd_reducer_fetch(const dd_reducer *r, uint64_t *host_keys, uint32_t *host_keynull,
