@@ -231,7 +231,7 @@ extern "C" dd_status dd_partitioner_create(const dd_batch_desc *batch, const int
         int wpb = 8;
         if (const char *e = getenv("DD_V2_WPB")) {
             int v = atoi(e);
-            if (v == 4 || v == 8) wpb = v;
+            if (v == 4 || v == 8 || v == 16) wpb = v;
         }
         /* aim for flush runs of >= 16 rows per partition (R >= 16P), floor g=2 (measured
          * best at the bench shape: wpb=8, g=4, P=128) */

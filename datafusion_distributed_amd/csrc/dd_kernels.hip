@@ -504,19 +504,48 @@ __global__ __launch_bounds__(WPB * WAVE) void k_scatter_staged(
         }
         __syncthreads();
 
-        /* cross-wave exclusive scan per partition + round totals */
-        for (uint32_t p = tid; p < nparts; p += BT) {
-            uint32_t run = 0;
+        /* cross-wave exclusive scan per partition + round totals, fused with the
+         * partition-offset scan (contiguous spans; one wave shfl-scans the span sums) */
+        {
+            const uint32_t span = (nparts + BT - 1) / BT;
+            const uint32_t plo = tid * span;
+            const uint32_t phi = (plo + span < nparts) ? plo + span : nparts;
+            uint32_t ssum = 0;
+            for (uint32_t p = plo; p < phi; p++) {
+                uint32_t run = 0;
 #pragma unroll
-            for (int w = 0; w < WPB; w++) {
-                uint32_t v = seghist[(size_t)w * nparts + p];
-                seghist[(size_t)w * nparts + p] = run;
-                run += v;
+                for (int w = 0; w < WPB; w++) {
+                    uint32_t v = seghist[(size_t)w * nparts + p];
+                    seghist[(size_t)w * nparts + p] = run;
+                    run += v;
+                }
+                roundcnt[p] = run;
+                ssum += run;
             }
-            roundcnt[p] = run;
+            scan_tmp[tid] = ssum;
+            __syncthreads();
+            if (tid < WAVE) {
+                uint32_t carry = 0;
+                for (int k = 0; k < BT / WAVE; k++) {
+                    uint32_t v = scan_tmp[k * WAVE + tid];
+#pragma unroll
+                    for (int d = 1; d < WAVE; d <<= 1) {
+                        uint32_t u = (uint32_t)__shfl_up((int)v, d);
+                        if (tid >= d) v += u;
+                    }
+                    v += carry;
+                    scan_tmp[k * WAVE + tid] = v;
+                    carry = (uint32_t)__shfl((int)v, WAVE - 1);
+                }
+            }
+            __syncthreads();
+            uint32_t run = (tid > 0) ? scan_tmp[tid - 1] : 0;
+            for (uint32_t p = plo; p < phi; p++) {
+                round_off[p] = run;
+                run += roundcnt[p];
+            }
+            __syncthreads();
         }
-        __syncthreads();
-        dd_block_excl_scan<BT>(roundcnt, round_off, nparts, scan_tmp);
 
         /* place rows into the partition-major LDS image (registers -> LDS) */
 #pragma unroll
@@ -880,6 +909,14 @@ hipError_t dd_launch_scatter_staged(const dd_kargs *a, int64_t nblocks, int64_t 
     DD_CASE(2, 8, 8, true)
     DD_CASE(4, 8, 8, true)
     DD_CASE(8, 8, 8, true)
+    DD_CASE(2, 16, 4, false)
+    DD_CASE(4, 16, 4, false)
+    DD_CASE(2, 16, 8, false)
+    DD_CASE(4, 16, 8, false)
+    DD_CASE(2, 16, 4, true)
+    DD_CASE(4, 16, 4, true)
+    DD_CASE(2, 16, 8, true)
+    DD_CASE(4, 16, 8, true)
 #undef DD_CASE
     return hipErrorInvalidValue;
 }
