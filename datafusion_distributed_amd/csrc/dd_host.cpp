@@ -228,28 +228,41 @@ extern "C" dd_status dd_partitioner_create(const dd_batch_desc *batch, const int
         row_stage += 4 * nvar + (nvar > 0 ? 4 : 0); /* VARLEN cols + ROWID col */
         /* auto: 8 waves x 4 groups (R=2048, 2 blocks/CU at the bench shape) measured best;
          * DD_V2_GMAX / DD_V2_WPB override for experiments */
-        int wpb = 8;
+        /* wpb cascade: 16-wave blocks measured best at the bench shape (wpb=16 g=4:
+         * K3 1.381 ms vs 1.408 at 8/4) but their partition arrays outgrow LDS at large P
+         * — fall back to 8 then 4 waves. Aim for flush runs >= 16 rows (R >= 16P),
+         * floor g=2. DD_V2_WPB / DD_V2_GMAX override for experiments. */
+        int wpb_cands[3] = {16, 8, 4};
         if (const char *e = getenv("DD_V2_WPB")) {
             int v = atoi(e);
-            if (v == 4 || v == 8 || v == 16) wpb = v;
+            if (v == 4 || v == 8 || v == 16) {
+                wpb_cands[0] = v;
+                wpb_cands[1] = wpb_cands[2] = 0;
+            }
         }
-        /* aim for flush runs of >= 16 rows per partition (R >= 16P), floor g=2 (measured
-         * best at the bench shape: wpb=8, g=4, P=128) */
-        int gtop = 2;
-        while (gtop < 8 && (size_t)gtop * wpb * 64 < 16 * (size_t)P) gtop *= 2;
-        if (const char *e = getenv("DD_V2_GMAX")) {
-            int v = atoi(e);
-            if (v == 2 || v == 4 || v == 8) gtop = v;
-        }
-        for (int g = gtop; g >= 2; g /= 2) {
-            const size_t part_lds = (size_t)P * (8 + 4 * wpb + 4 + 4) + (size_t)wpb * 64 * 4;
-            size_t lds = part_lds + (size_t)g * wpb * 64 * row_stage;
-            if (lds <= 163840) {
-                p->staged = true;
-                p->gmax = g;
-                p->wpb = wpb;
-                p->lds_k3 = lds;
-                break;
+        for (int wi = 0; wi < 3 && !p->staged; wi++) {
+            const int wpb = wpb_cands[wi];
+            if (!wpb) continue;
+            /* k_scatter_staged is instantiated for (gmax, wpb) in {2,4,8}x{4,8} and
+             * {2,4}x{16} */
+            int gcap = (wpb == 16) ? 4 : 8;
+            int gtop = 2;
+            while (gtop < gcap && (size_t)gtop * wpb * 64 < 16 * (size_t)P) gtop *= 2;
+            if (const char *e = getenv("DD_V2_GMAX")) {
+                int v = atoi(e);
+                if (v == 2 || v == 4 || v == 8) gtop = (v <= gcap) ? v : gcap;
+            }
+            for (int g = gtop; g >= 2; g /= 2) {
+                const size_t part_lds =
+                    (size_t)P * (8 + 4 * wpb + 4 + 4) + (size_t)wpb * 64 * 4;
+                size_t lds = part_lds + (size_t)g * wpb * 64 * row_stage;
+                if (lds <= 163840) {
+                    p->staged = true;
+                    p->gmax = g;
+                    p->wpb = wpb;
+                    p->lds_k3 = lds;
+                    break;
+                }
             }
         }
     }
