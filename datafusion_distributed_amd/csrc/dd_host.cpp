@@ -230,7 +230,7 @@ extern "C" dd_status dd_partitioner_create(const dd_batch_desc *batch, const int
          * DD_V2_GMAX / DD_V2_WPB override for experiments */
         /* wpb cascade: 16-wave blocks measured best at the bench shape (wpb=16 g=4:
          * K3 1.381 ms vs 1.408 at 8/4) but their partition arrays outgrow LDS at large P
-         * — fall back to 8 then 4 waves. Aim for flush runs >= 16 rows (R >= 16P),
+         * — fall back to 8 then 4 waves. Aim for R >= 32P (16/4 measured best at P=128),
          * floor g=2. DD_V2_WPB / DD_V2_GMAX override for experiments. */
         int wpb_cands[3] = {16, 8, 4};
         if (const char *e = getenv("DD_V2_WPB")) {
@@ -247,7 +247,7 @@ extern "C" dd_status dd_partitioner_create(const dd_batch_desc *batch, const int
              * {2,4}x{16} */
             int gcap = (wpb == 16) ? 4 : 8;
             int gtop = 2;
-            while (gtop < gcap && (size_t)gtop * wpb * 64 < 16 * (size_t)P) gtop *= 2;
+            while (gtop < gcap && (size_t)gtop * wpb * 64 < 32 * (size_t)P) gtop *= 2;
             if (const char *e = getenv("DD_V2_GMAX")) {
                 int v = atoi(e);
                 if (v == 2 || v == 4 || v == 8) gtop = (v <= gcap) ? v : gcap;
