@@ -1,0 +1,60 @@
+"""Randomized parity fuzz (marked gpu): 20 random batch shapes — dtype mixes, null
+densities, partition counts (pow2 and not), sizes including tiny/ragged — every one must
+match the CPU oracle bit-exactly through the full partition path. Catches configuration
+corners the fixed tests miss (both staged and v1 kernels get hit depending on shape)."""
+
+import numpy as np
+import pytest
+
+from tests.test_gpu_parity import check_against_oracle
+
+pytestmark = pytest.mark.gpu
+
+FIXED = ["u8", "i16", "i32", "i64", "f32", "f64", "bool"]
+
+
+def random_col(rng, n, dtype, null_p):
+    valid = (rng.random(n) >= null_p).astype(np.uint8) if null_p > 0 else None
+    if dtype == "utf8":
+        lens = rng.integers(0, rng.integers(1, 40), n)
+        off = np.zeros(n + 1, dtype=np.int32)
+        off[1:] = np.cumsum(lens)
+        data = rng.integers(0, 256, int(off[-1]), dtype=np.int64).astype(np.uint8)
+        return {"dtype": "utf8", "data": data, "offsets": off, "valid": valid}
+    if dtype == "dict32":
+        nvals = int(rng.integers(1, 50))
+        vals = [bytes(rng.integers(0, 256, rng.integers(0, 12), dtype=np.int64)
+                      .astype(np.uint8)) for _ in range(nvals)]
+        doff = np.zeros(nvals + 1, dtype=np.int32)
+        doff[1:] = np.cumsum([len(v) for v in vals])
+        return {"dtype": "dict32",
+                "data": rng.integers(0, nvals, n).astype(np.int32),
+                "dict_bytes": np.frombuffer(b"".join(vals), dtype=np.uint8).copy()
+                if doff[-1] else np.zeros(0, dtype=np.uint8),
+                "dict_offsets": doff, "valid": valid}
+    npdt = {"u8": np.uint8, "bool": np.uint8, "i16": np.int16, "i32": np.int32,
+            "i64": np.int64, "f32": np.float32, "f64": np.float64}[dtype]
+    if dtype in ("f32", "f64"):
+        data = rng.normal(size=n).astype(npdt)
+        data[rng.random(n) < 0.01] = 0.0
+        data[rng.random(n) < 0.01] = -0.0
+        data[rng.random(n) < 0.005] = np.nan
+    elif dtype == "bool":
+        data = (rng.random(n) > 0.5).astype(np.uint8)
+    else:
+        info = np.iinfo(npdt)
+        data = rng.integers(info.min, int(info.max) + 1, n, dtype=np.int64).astype(npdt)
+    return {"dtype": dtype, "data": data, "valid": valid}
+
+
+@pytest.mark.parametrize("case", range(20))
+def test_fuzz_parity(case):
+    rng = np.random.default_rng(1000 + case)
+    n = int(rng.choice([1, 2, 63, 64, 65, 1000, 4096, 30000, 250000]))
+    ncols = int(rng.integers(1, 9))
+    dtypes = list(rng.choice(FIXED + ["utf8", "dict32"], ncols))
+    cols = [random_col(rng, n, dt, float(rng.choice([0, 0, 0.1, 0.5]))) for dt in dtypes]
+    nkeys = int(rng.integers(1, min(ncols, 4) + 1))
+    key_idx = list(rng.choice(ncols, nkeys, replace=False))
+    nparts = int(rng.choice([1, 2, 3, 7, 8, 16, 100, 128, 777, 2048]))
+    check_against_oracle(cols, [int(k) for k in key_idx], nparts)
