@@ -56,6 +56,7 @@ __device__ __forceinline__ uint64_t dd_canon_f32_dev(float v) {
 __device__ __forceinline__ uint64_t dd_value_hash_dev(const dd_kcol &c, int64_t i) {
     switch (c.dtype) {
     case DD_KDT_U8:
+    case DD_KDT_BOOL: /* unpacked u8 0/1 hashes like u8 (oracle dd_value_hash) */
         return dd_mix64((uint64_t)((const uint8_t *)c.data)[i]);
     case DD_KDT_I16:
         return dd_mix64((uint64_t)((const uint16_t *)c.data)[i]);
