@@ -132,6 +132,9 @@ def main():
     # P=2048 (round-1 max partition count)
     cols2 = make_bench_batch(16384, seed=123)
     save_case(os.path.join(d, "bench_i64key_p2048.npz"), cols2, [0], 2048)
+    # bool + narrow-int keys (a missing BOOL case in the device hash escaped the
+    # original fixtures — pin it)
+    save_case(os.path.join(d, "bench_boolkey_p7.npz"), cols, [2, 5, 7], 7)
     # scalar hash known-answer vectors (pin mix64 / hash_bytes directly)
     from oracle.pyref import hash_bytes_scalar, mix64_scalar
     vec_in = [0, 1, 0x9E3779B97F4A7C15, 2**64 - 1, 42]
