@@ -430,60 +430,43 @@ __global__ __launch_bounds__(WPB * WAVE) void k_scatter_staged(
     uint64_t colv[GMAX][MAXC];
     uint8_t valv[GMAX][MAXC];
 
-    /* Column-OUTER preload: the (wave-uniform) dtype switch is hoisted out of the
-     * per-row path, so each column's GMAX loads are a straight unrolled run the compiler
-     * can cover with counted s_waitcnt instead of vmcnt(0) full drains (162 of which
-     * serialized the cross-round store/load overlap in the row-major version). */
     auto preload = [&](int64_t rstart, int64_t rend) {
         const int64_t segstart = rstart + (int64_t)wid * SEG;
 #pragma unroll
         for (int g = 0; g < GMAX; g++) {
             const int64_t row = segstart + g * WAVE + lane;
-            actr[g] = row < rend;
-            pidr[g] = actr[g] ? pid_in[row] : 0;
-        }
-        for (int c = 0; c < a.n_cols; c++) {
-            const dd_kcol &col = a.cols[c];
-            bool synthetic = false;
-            if constexpr (HASVAR) {
-                if (col.dtype == DD_KDT_VARLEN) {
-                    const int32_t *off = (const int32_t *)col.data;
+            const bool active = row < rend;
+            actr[g] = active;
+            if (!active) {
+                pidr[g] = 0;
+                continue;
+            }
+            pidr[g] = pid_in[row];
 #pragma unroll
-                    for (int g = 0; g < GMAX; g++) {
-                        const int64_t row = segstart + g * WAVE + lane;
-                        if (actr[g]) colv[g][c] = (uint32_t)(off[row + 1] - off[row]);
+            for (int c = 0; c < MAXC; c++) {
+                if (c >= a.n_cols) break;
+                const dd_kcol &col = a.cols[c];
+                bool synthetic = false;
+                if constexpr (HASVAR) {
+                    if (col.dtype == DD_KDT_VARLEN) {
+                        /* synthetic: the var column's per-row byte length (K4 header) */
+                        const int32_t *off = (const int32_t *)col.data;
+                        colv[g][c] = (uint32_t)(off[row + 1] - off[row]);
+                        synthetic = true;
+                    } else if (col.dtype == DD_KDT_ROWID) {
+                        colv[g][c] = (uint32_t)row; /* synthetic: permutation for K4d */
+                        synthetic = true;
                     }
-                    synthetic = true;
-                } else if (col.dtype == DD_KDT_ROWID) {
-#pragma unroll
-                    for (int g = 0; g < GMAX; g++)
-                        colv[g][c] = (uint32_t)(segstart + g * WAVE + lane);
-                    synthetic = true;
                 }
-            }
-            if (!synthetic) {
-                switch (col.elem) {
-#define DD_LOADC(T)                                                                          \
-    {                                                                                        \
-        const T *d = (const T *)col.data;                                                    \
-        _Pragma("unroll") for (int g = 0; g < GMAX; g++) {                                   \
-            const int64_t row = segstart + g * WAVE + lane;                                  \
-            if (actr[g]) colv[g][c] = d[row];                                                \
-        }                                                                                    \
-    }
-                case 1: DD_LOADC(uint8_t) break;
-                case 2: DD_LOADC(uint16_t) break;
-                case 4: DD_LOADC(uint32_t) break;
-                case 8: DD_LOADC(uint64_t) break;
-#undef DD_LOADC
+                if (!synthetic) {
+                    switch (col.elem) {
+                    case 1: colv[g][c] = ((const uint8_t *)col.data)[row]; break;
+                    case 2: colv[g][c] = ((const uint16_t *)col.data)[row]; break;
+                    case 4: colv[g][c] = ((const uint32_t *)col.data)[row]; break;
+                    case 8: colv[g][c] = ((const uint64_t *)col.data)[row]; break;
+                    }
                 }
-            }
-            if (col.valid) {
-#pragma unroll
-                for (int g = 0; g < GMAX; g++) {
-                    const int64_t row = segstart + g * WAVE + lane;
-                    if (actr[g]) valv[g][c] = col.valid[row];
-                }
+                if (col.valid) valv[g][c] = col.valid[row];
             }
         }
     };
@@ -564,37 +547,28 @@ __global__ __launch_bounds__(WPB * WAVE) void k_scatter_staged(
             __syncthreads();
         }
 
-        /* place rows into the partition-major LDS image (registers -> LDS);
-         * slots computed once, then column-outer writes */
-        uint32_t slotr[GMAX];
+        /* place rows into the partition-major LDS image (registers -> LDS) */
 #pragma unroll
         for (int g = 0; g < GMAX; g++) {
             if (!actr[g]) continue;
             const uint32_t pid = pidr[g];
             const uint32_t rank_r = myseg[pid] + rankr[g];
-            slotr[g] = round_off[pid] + rank_r;
-            dstg[slotr[g]] = (uint32_t)(dstbase[pid] + rank_r);
-        }
-        {
+            const uint32_t slot = round_off[pid] + rank_r;
+            dstg[slot] = (uint32_t)(dstbase[pid] + rank_r);
             char *stage = stage0;
-            for (int c = 0; c < a.n_cols; c++) {
+#pragma unroll
+            for (int c = 0; c < MAXC; c++) {
+                if (c >= a.n_cols) break;
                 const dd_kcol &col = a.cols[c];
                 switch (col.elem) {
-#define DD_STAGEC(T)                                                                         \
-    _Pragma("unroll") for (int g = 0; g < GMAX; g++) {                                       \
-        if (actr[g]) ((T *)stage)[slotr[g]] = (T)colv[g][c];                                 \
-    }
-                case 1: DD_STAGEC(uint8_t) break;
-                case 2: DD_STAGEC(uint16_t) break;
-                case 4: DD_STAGEC(uint32_t) break;
-                case 8: DD_STAGEC(uint64_t) break;
-#undef DD_STAGEC
+                case 1: ((uint8_t *)stage)[slot] = (uint8_t)colv[g][c]; break;
+                case 2: ((uint16_t *)stage)[slot] = (uint16_t)colv[g][c]; break;
+                case 4: ((uint32_t *)stage)[slot] = (uint32_t)colv[g][c]; break;
+                case 8: ((uint64_t *)stage)[slot] = colv[g][c]; break;
                 }
                 stage += (size_t)R * col.elem;
                 if (col.valid) {
-#pragma unroll
-                    for (int g = 0; g < GMAX; g++)
-                        if (actr[g]) ((uint8_t *)stage)[slotr[g]] = valv[g][c];
+                    ((uint8_t *)stage)[slot] = valv[g][c];
                     stage += R;
                 }
             }
@@ -609,25 +583,28 @@ __global__ __launch_bounds__(WPB * WAVE) void k_scatter_staged(
 
         /* flush: consecutive LDS slots -> consecutive global rows within each partition
          * run; stores coalesce into run-sized segments */
-        {
+        for (int i = tid; i < round_rows; i += BT) {
+            const uint64_t dst = dstg[i];
             char *stage = stage0;
-            for (int c = 0; c < a.n_cols; c++) {
+#pragma unroll
+            for (int c = 0; c < MAXC; c++) {
+                if (c >= a.n_cols) break;
                 const dd_kcol &col = a.cols[c];
                 switch (col.elem) {
-#define DD_FLUSHC(T)                                                                         \
-    for (int i = tid; i < round_rows; i += BT) {                                             \
-        ((T *)col.out_data)[dstg[i]] = ((const T *)stage)[i];                                \
-    }
-                case 1: DD_FLUSHC(uint8_t) break;
-                case 2: DD_FLUSHC(uint16_t) break;
-                case 4: DD_FLUSHC(uint32_t) break;
-                case 8: DD_FLUSHC(uint64_t) break;
-#undef DD_FLUSHC
+                case 1: ((uint8_t *)col.out_data)[dst] = ((const uint8_t *)stage)[i]; break;
+                case 2:
+                    ((uint16_t *)col.out_data)[dst] = ((const uint16_t *)stage)[i];
+                    break;
+                case 4:
+                    ((uint32_t *)col.out_data)[dst] = ((const uint32_t *)stage)[i];
+                    break;
+                case 8:
+                    ((uint64_t *)col.out_data)[dst] = ((const uint64_t *)stage)[i];
+                    break;
                 }
                 stage += (size_t)R * col.elem;
                 if (col.valid) {
-                    for (int i = tid; i < round_rows; i += BT)
-                        col.out_valid[dstg[i]] = ((const uint8_t *)stage)[i];
+                    col.out_valid[dst] = ((const uint8_t *)stage)[i];
                     stage += R;
                 }
             }
