@@ -62,8 +62,7 @@ def _ptr(a):
 def _make_ocol(col):
     oc = _OCol()
     oc.dtype = _DTYPE_CODE[col["dtype"]]
-    oc.data = ctypes.cast(_ptr(np.ascontiguousarray(col["data"])), ctypes.c_void_p)
-    # keep refs alive via attributes
+    # keep refs alive via attributes until the ctypes call returns
     oc._keep = [np.ascontiguousarray(col["data"])]
     oc.data = ctypes.cast(oc._keep[0].ctypes.data, ctypes.c_void_p)
     if col.get("valid") is not None:
