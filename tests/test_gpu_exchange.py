@@ -53,3 +53,43 @@ def test_self_exchange_roundtrip():
     comm.destroy()
     part.destroy()
     batch.free()
+
+
+def test_self_exchange_empty_batch():
+    """Zero-row batch through partition + exchange (the reference's empty-batch wire edge
+    case, tests/empty_columns_between_workers.rs analog at our boundary)."""
+    cols = [{"dtype": "i64", "data": np.zeros(0, dtype=np.int64), "valid": None}]
+    batch = api.DeviceBatch(cols)
+    part = api.Partitioner(batch, [0], 4)
+    part.run()
+    part.sync()
+    comm = api.Comm(api.Comm.unique_id(), 0, 1)
+    ex = comm.exchange(part)
+    assert ex.total_rows == 0
+    assert (ex.row_counts() == 0).all()
+    ex.destroy()
+    comm.destroy()
+    part.destroy()
+    batch.free()
+
+
+def test_self_exchange_sparse_partitions():
+    """Highly skewed input: most partitions empty (zero-size slices must pair correctly)."""
+    n = 10000
+    cols = [{"dtype": "i64", "data": np.full(n, 123456789, dtype=np.int64), "valid": None},
+            {"dtype": "f64", "data": np.arange(n, dtype=np.float64), "valid": None}]
+    batch = api.DeviceBatch(cols)
+    part = api.Partitioner(batch, [0], 64)
+    part.run()
+    part.sync()
+    comm = api.Comm(api.Comm.unique_id(), 0, 1)
+    ex = comm.exchange(part)
+    assert ex.total_rows == n
+    rc = ex.row_counts()
+    assert (rc > 0).sum() == 1  # single key -> single partition
+    got = ex.col_data(1)["data"]
+    assert (np.sort(got) == np.arange(n, dtype=np.float64)).all()
+    ex.destroy()
+    comm.destroy()
+    part.destroy()
+    batch.free()
