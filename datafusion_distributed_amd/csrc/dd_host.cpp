@@ -1050,3 +1050,184 @@ extern "C" dd_status dd_reducer_fetch(const dd_reducer *r, uint64_t *host_keys,
 }
 
 extern "C" void dd_reducer_destroy(dd_reducer *r) { delete r; }
+
+/* ---------------- coalesce (dd_coalesce_run) ---------------- */
+
+/* task_group (network_coalesce.rs:376-400): contiguous ceil-split of producers */
+static void dd_task_group(int input_tasks, int task_index, int task_count, int *start,
+                          int *len) {
+    const int base = input_tasks / task_count;
+    const int extra = input_tasks % task_count;
+    *len = base + (task_index < extra ? 1 : 0);
+    *start = task_index * base + (task_index < extra ? task_index : extra);
+}
+
+extern "C" dd_status dd_coalesce_run(dd_comm *c, const dd_partitioner *p,
+                                     int32_t consumer_tasks, void *stream,
+                                     dd_exchanged **out) {
+    if (!c || !p || !out) return set_err(DD_ERR_INVALID, "null argument");
+    if (!p->has_run) return set_err(DD_ERR_INVALID, "partitioner has not run");
+    if (consumer_tasks < 1 || consumer_tasks > c->nranks)
+        return set_err(DD_ERR_INVALID, "consumer_tasks must be in [1, nranks]");
+    hipStream_t s = (hipStream_t)stream;
+    const int R = c->nranks;
+    const uint32_t P = p->nparts;
+    const int nvar = p->ka.n_var;
+
+    /* 1) size matrix allgather (same meta layout as dd_exchange_run) */
+    const size_t meta_n = (size_t)P * (1 + nvar);
+    std::vector<int64_t> my_meta(meta_n);
+    std::vector<int64_t> off_h(P + 1);
+    dd_status st = dd_partitioner_row_offsets(p, off_h.data());
+    if (st != DD_OK) return st;
+    for (uint32_t q = 0; q < P; q++) my_meta[q] = off_h[q + 1] - off_h[q];
+    std::vector<std::vector<int64_t>> boff_h(nvar, std::vector<int64_t>(P + 1));
+    for (int v = 0; v < nvar; v++) {
+        st = dd_partitioner_byte_offsets(p, p->ka.var_idx[v], boff_h[v].data());
+        if (st != DD_OK) return st;
+        for (uint32_t q = 0; q < P; q++)
+            my_meta[(size_t)(1 + v) * P + q] = boff_h[v][q + 1] - boff_h[v][q];
+    }
+    int64_t *d_in = nullptr, *d_all = nullptr;
+    HIP_TRY(hipMalloc(&d_in, meta_n * 8));
+    HIP_TRY(hipMalloc(&d_all, (size_t)R * meta_n * 8));
+    HIP_TRY(hipMemcpyAsync(d_in, my_meta.data(), meta_n * 8, hipMemcpyHostToDevice, s));
+    NCCL_TRY(ncclAllGather(d_in, d_all, meta_n, ncclInt64, c->comm, s));
+    std::vector<int64_t> all_meta((size_t)R * meta_n);
+    HIP_TRY(hipStreamSynchronize(s));
+    HIP_TRY(hipMemcpy(all_meta.data(), d_all, (size_t)R * meta_n * 8,
+                      hipMemcpyDeviceToHost));
+    (void)hipFree(d_in);
+    (void)hipFree(d_all);
+
+    /* 2) my group as a consumer (ranks >= consumer_tasks consume nothing) */
+    int gstart = 0, glen = 0;
+    if (c->rank < consumer_tasks) dd_task_group(R, c->rank, consumer_tasks, &gstart, &glen);
+    /* my consumer as a producer: the consumer whose group contains me */
+    int my_consumer = -1;
+    for (int t = 0; t < consumer_tasks; t++) {
+        int s0, l0;
+        dd_task_group(R, t, consumer_tasks, &s0, &l0);
+        if (c->rank >= s0 && c->rank < s0 + l0) my_consumer = t;
+    }
+
+    auto e = new dd_exchanged();
+    e->n_cols = p->batch.n_cols;
+    e->nranks = glen; /* producers I consume */
+    e->P = P;
+    e->row_counts.assign((size_t)(glen > 0 ? glen : 1) * P, 0);
+    e->byte_counts.assign(p->batch.n_cols, {});
+    std::vector<int64_t> recv_rows(glen, 0);
+    for (int gi = 0; gi < glen; gi++) {
+        const int r = gstart + gi;
+        for (uint32_t q = 0; q < P; q++) {
+            int64_t n = all_meta[(size_t)r * meta_n + q];
+            e->row_counts[(size_t)gi * P + q] = n;
+            recv_rows[gi] += n;
+            e->total_rows += n;
+        }
+    }
+    auto fail = [&](dd_status sc, const char *m) {
+        delete e;
+        return set_err(sc, m);
+    };
+
+    std::vector<std::vector<int64_t>> recv_bytes(p->batch.n_cols,
+                                                 std::vector<int64_t>(glen, 0));
+    for (int ci = 0; ci < p->batch.n_cols; ci++) {
+        const dd_kcol &kc = p->ka.cols[ci];
+        if (kc.elem > 0) {
+            if (hipMalloc(&e->data[ci], (size_t)e->total_rows * kc.elem + 1) != hipSuccess)
+                return fail(DD_ERR_HIP, "coalesce recv alloc");
+        } else {
+            int v = -1;
+            for (int vv = 0; vv < nvar; vv++)
+                if (p->ka.var_idx[vv] == ci) v = vv;
+            e->byte_counts[ci].assign((size_t)(glen > 0 ? glen : 1) * P, 0);
+            int64_t total_b = 0;
+            for (int gi = 0; gi < glen; gi++) {
+                const int r = gstart + gi;
+                for (uint32_t q = 0; q < P; q++) {
+                    int64_t b = all_meta[(size_t)r * meta_n + (size_t)(1 + v) * P + q];
+                    e->byte_counts[ci][(size_t)gi * P + q] = b;
+                    recv_bytes[ci][gi] += b;
+                    total_b += b;
+                }
+            }
+            if (hipMalloc(&e->data[ci], (size_t)total_b + 1) != hipSuccess)
+                return fail(DD_ERR_HIP, "coalesce recv alloc (var)");
+            if (hipMalloc((void **)&e->lengths[ci], (size_t)e->total_rows * 4 + 1) !=
+                hipSuccess)
+                return fail(DD_ERR_HIP, "coalesce recv alloc (lengths)");
+        }
+        if (kc.valid) {
+            if (hipMalloc((void **)&e->valid[ci], (size_t)e->total_rows + 1) != hipSuccess)
+                return fail(DD_ERR_HIP, "coalesce recv alloc (validity)");
+        }
+    }
+    if (hipEventCreate(&e->e0) != hipSuccess || hipEventCreate(&e->e1) != hipSuccess)
+        return fail(DD_ERR_HIP, "event create");
+
+    /* 3) grouped p2p: I SEND my whole output to my_consumer; as a consumer I RECV from
+     * each producer in my group, producer-major. Post order per peer: per col (data,
+     * [lengths], [valid]) — symmetric on both sides. */
+    const int64_t my_rows = off_h[P];
+    HIP_TRY(hipEventRecord(e->e0, s));
+    NCCL_TRY(ncclGroupStart());
+    for (int ci = 0; ci < p->batch.n_cols; ci++) {
+        const dd_kcol &kc = p->ka.cols[ci];
+        /* sends (every rank has exactly one consumer) */
+        if (my_consumer >= 0) {
+            if (kc.elem > 0) {
+                if (my_rows > 0)
+                    NCCL_TRY(ncclSend(p->out_data[ci], my_rows * kc.elem, ncclUint8,
+                                      my_consumer, c->comm, s));
+            } else {
+                int v = -1;
+                for (int vv = 0; vv < nvar; vv++)
+                    if (p->ka.var_idx[vv] == ci) v = vv;
+                const int64_t my_b = boff_h[v][P];
+                if (my_b > 0)
+                    NCCL_TRY(ncclSend(p->out_data[ci], my_b, ncclUint8, my_consumer,
+                                      c->comm, s));
+                if (my_rows > 0)
+                    NCCL_TRY(ncclSend(p->out_lengths[ci], my_rows * 4, ncclUint8,
+                                      my_consumer, c->comm, s));
+            }
+            if (kc.valid && my_rows > 0)
+                NCCL_TRY(ncclSend(p->out_valid[ci], my_rows, ncclUint8, my_consumer,
+                                  c->comm, s));
+            if (my_consumer != c->rank)
+                e->egress += (kc.elem > 0 ? my_rows * kc.elem : 0);
+        }
+        /* recvs */
+        int64_t rrow0 = 0, rb0 = 0;
+        for (int gi = 0; gi < glen; gi++) {
+            const int r = gstart + gi;
+            if (kc.elem > 0) {
+                if (recv_rows[gi] > 0)
+                    NCCL_TRY(ncclRecv((uint8_t *)e->data[ci] + rrow0 * kc.elem,
+                                      recv_rows[gi] * kc.elem, ncclUint8, r, c->comm, s));
+            } else {
+                if (recv_bytes[ci][gi] > 0)
+                    NCCL_TRY(ncclRecv((uint8_t *)e->data[ci] + rb0, recv_bytes[ci][gi],
+                                      ncclUint8, r, c->comm, s));
+                if (recv_rows[gi] > 0)
+                    NCCL_TRY(ncclRecv((uint8_t *)e->lengths[ci] + rrow0 * 4,
+                                      recv_rows[gi] * 4, ncclUint8, r, c->comm, s));
+                rb0 += recv_bytes[ci][gi];
+            }
+            if (kc.valid && recv_rows[gi] > 0)
+                NCCL_TRY(ncclRecv(e->valid[ci] + rrow0, recv_rows[gi], ncclUint8, r,
+                                  c->comm, s));
+            rrow0 += recv_rows[gi];
+        }
+    }
+    NCCL_TRY(ncclGroupEnd());
+    HIP_TRY(hipEventRecord(e->e1, s));
+    HIP_TRY(hipStreamSynchronize(s));
+    HIP_TRY(hipEventElapsedTime(&e->ms, e->e0, e->e1));
+    *out = e;
+    return DD_OK;
+}
+

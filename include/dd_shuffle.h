@@ -147,6 +147,16 @@ dd_status dd_exchanged_byte_counts(const dd_exchanged *e, int32_t col, int64_t *
  * this rank to OTHER ranks (xGMI egress) */
 dd_status dd_exchanged_stats(const dd_exchanged *e, float *ms, int64_t *egress_bytes);
 
+/* ---------------- coalesce (N->M task coalesce, no repartition) ----------------
+ * Data plane for NetworkCoalesceExec (src/execution_plans/network_coalesce.rs:24-70):
+ * consumer rank t receives the WHOLE partitioned output of every producer rank in its
+ * contiguous group (task_group, :376-400; groups = ceil-split of nranks over
+ * consumer_tasks). No repartition head (ProducerHead::None, :209-212). Reuses the
+ * dd_exchanged result shape: producer-major concatenation, per-(producer, partition)
+ * row counts. Ranks outside any group's producer set still participate (empty sends). */
+dd_status dd_coalesce_run(dd_comm *c, const dd_partitioner *p, int32_t consumer_tasks,
+                          void *stream, dd_exchanged **out);
+
 /* ---------------- broadcast (build side of CollectLeft joins) ----------------
  * Replaces BroadcastExec's cache + NetworkBroadcastExec's per-consumer fetch
  * (src/execution_plans/broadcast.rs:24-28,163; network_broadcast.rs:245-266) with one
