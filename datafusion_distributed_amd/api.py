@@ -259,6 +259,15 @@ class Comm:
         _check(lib().dd_exchange_run(self.h, part.h, stream, ctypes.byref(h)))
         return Exchanged(h, part, self)
 
+    def coalesce(self, part: Partitioner, consumer_tasks: int, stream=None):
+        """NetworkCoalesceExec data plane (dd_coalesce_run): this consumer rank receives
+        the whole partitioned output of every producer rank in its contiguous group."""
+        h = ctypes.c_void_p()
+        _check(lib().dd_coalesce_run(self.h, part.h, consumer_tasks, stream,
+                                     ctypes.byref(h)))
+        ex = Exchanged(h, part, self)
+        return ex
+
     def broadcast(self, batch, root=0, stream=None):
         """Replicate the root's device batch on every rank (dd_broadcast_run: the
         BroadcastExec/NetworkBroadcastExec data plane)."""
@@ -321,12 +330,16 @@ class Exchanged:
     def total_rows(self):
         return lib().dd_exchanged_total_rows(self.h)
 
-    def row_counts(self):
-        out = np.empty(self.comm.nranks * (self.part.nparts // self.comm.nranks),
-                       dtype=np.int64)
+    def row_counts(self, producers=None, parts=None):
+        """Per-(producer, partition) row counts. Defaults fit dd_exchange_run's window
+        shape; pass producers/parts explicitly for dd_coalesce_run results (group size,
+        full P)."""
+        producers = self.comm.nranks if producers is None else producers
+        parts = (self.part.nparts // self.comm.nranks) if parts is None else parts
+        out = np.empty(max(producers * parts, 1), dtype=np.int64)
         _check(lib().dd_exchanged_row_counts(
             self.h, out.ctypes.data_as(ctypes.POINTER(ctypes.c_int64))))
-        return out.reshape(self.comm.nranks, -1)
+        return out.reshape(max(producers, 1), -1)
 
     def byte_counts(self, col):
         out = np.empty(self.comm.nranks * (self.part.nparts // self.comm.nranks),

@@ -93,3 +93,34 @@ def test_self_exchange_sparse_partitions():
     comm.destroy()
     part.destroy()
     batch.free()
+
+
+def test_self_coalesce_roundtrip():
+    """dd_coalesce_run at nranks=1, consumer_tasks=1: the consumer's group is {rank 0},
+    so the received buffers equal the partitioner's own output (whole-partition fetch,
+    no repartition — NetworkCoalesceExec's data plane, network_coalesce.rs:24-70)."""
+    rng = np.random.default_rng(71)
+    n, P = 50000, 8
+    cols = [
+        {"dtype": "i64", "data": rng.integers(0, 10**10, n, dtype=np.int64), "valid": None},
+        {"dtype": "i32", "data": rng.integers(0, 100, n, dtype=np.int64).astype(np.int32),
+         "valid": (rng.random(n) > 0.1).astype(np.uint8)},
+    ]
+    batch = api.DeviceBatch(cols)
+    part = api.Partitioner(batch, [0], P)
+    part.run()
+    part.sync()
+    ref = oracle.repartition(cols, [0], P)
+    comm = api.Comm(api.Comm.unique_id(), 0, 1)
+    co = comm.coalesce(part, consumer_tasks=1)
+    assert co.total_rows == n
+    rc = co.row_counts(producers=1, parts=P)
+    assert (rc[0] == ref["part_offsets"][1:] - ref["part_offsets"][:-1]).all()
+    got = co.col_data(0)["data"]
+    assert (got == ref["cols"][0]["data"]).all()
+    v = co.col_validity(1)
+    assert (v == ref["cols"][1]["valid"]).all()
+    co.destroy()
+    comm.destroy()
+    part.destroy()
+    batch.free()
