@@ -172,6 +172,10 @@ def main():
                     help="measured HBM bytes per K3 launch from a separate rocprofv3 "
                          "--pmc run (profiles/); null if not provided")
     ap.add_argument("--skip-cpu-baseline", action="store_true")
+    ap.add_argument("--no-pipeline", action="store_true",
+                    help="disable the two-stream batch pipeline at N=1 (phase1 of batch "
+                         "s+1 overlapped with phase2 of batch s, mirroring the "
+                         "reference's continuous batch streaming)")
     args = ap.parse_args()
 
     rank = int(os.environ.get("RANK", "0"))
@@ -214,6 +218,11 @@ def main():
 
     part = api.Partitioner(batch, key_idx, p_total)
     comm = create_comm(rank, world) if world > 1 else None
+    pipeline = world == 1 and not args.no_pipeline and torch.cuda.is_available()
+    if pipeline:
+        part2 = api.Partitioner(batch, key_idx, p_total)
+        parts2 = [part, part2]
+        s1, s2 = torch.cuda.Stream(), torch.cuda.Stream()
 
     def step():
         part.run()
@@ -222,6 +231,22 @@ def main():
             ex.destroy()
         else:
             part.sync()
+
+    def pipelined_steps(k):
+        """k batch passes with phase1(s+1) on stream 1 overlapped with phase2(s) on
+        stream 2 (events order phases of the same partitioner across streams)."""
+        for i in range(k):
+            cur = parts2[i % 2]
+            cur.wait_phase2(s1)   # don't overwrite pid/counts its last scatter still reads
+            cur.run_phase1(s1)
+            if i > 0:
+                prv = parts2[(i + 1) % 2]
+                prv.wait_phase1(s2)
+                prv.run_phase2(s2)
+        last = parts2[(k - 1) % 2]
+        last.wait_phase1(s2)
+        last.run_phase2(s2)
+        torch.cuda.synchronize()
 
     def barrier_sync():
         if world > 1:
@@ -232,14 +257,20 @@ def main():
         torch.cuda.synchronize() if torch.cuda.is_available() else api.lib().dd_device_sync()
 
     for _ in range(args.warmup):
+        if pipeline:
+            part2.run()
         step()
     barrier_sync()
 
     k3_ms_total = 0.0
     t0 = time.perf_counter()
-    for _ in range(args.steps):
-        step()
-        k3_ms_total += part.kernel_ms()[2]
+    if pipeline:
+        pipelined_steps(args.steps)
+        k3_ms_total = part.kernel_ms()[2] * args.steps  # last-run K3 (events, per pass)
+    else:
+        for _ in range(args.steps):
+            step()
+            k3_ms_total += part.kernel_ms()[2]
     barrier_sync()
     elapsed = time.perf_counter() - t0
 
@@ -274,6 +305,7 @@ def main():
 
     config = {
         "workload": args.workload,
+        "batch_pipeline": bool(pipeline),
         "rows_per_rank": rows,
         "row_bytes": round(total_bytes_in / max(rows, 1), 1),
         "p_total": p_total,

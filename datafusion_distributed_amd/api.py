@@ -109,6 +109,14 @@ def device_count():
     return lib().dd_device_count()
 
 
+def _stream_arg(stream):
+    """Accept None, a raw int hipStream_t, or a torch.cuda.Stream."""
+    if stream is None:
+        return None
+    h = getattr(stream, "cuda_stream", stream)
+    return ctypes.c_void_p(h)
+
+
 def _dev_alloc(nbytes):
     p = ctypes.c_void_p()
     _check(lib().dd_dev_alloc(ctypes.c_int64(max(nbytes, 1)), ctypes.byref(p)))
@@ -188,6 +196,20 @@ class Partitioner:
 
     def run(self, stream=None):
         _check(lib().dd_partitioner_run(self.h, stream))
+
+    def run_phase1(self, stream=None):
+        """K1 hash+count + K2 scans (batch-pipelining split; see dd_shuffle.h)."""
+        _check(lib().dd_partitioner_run_phase1(self.h, _stream_arg(stream)))
+
+    def run_phase2(self, stream=None):
+        """K3 scatter (+K4 var bytes)."""
+        _check(lib().dd_partitioner_run_phase2(self.h, _stream_arg(stream)))
+
+    def wait_phase1(self, stream):
+        _check(lib().dd_partitioner_wait_phase1(self.h, _stream_arg(stream)))
+
+    def wait_phase2(self, stream):
+        _check(lib().dd_partitioner_wait_phase2(self.h, _stream_arg(stream)))
 
     def sync(self):
         _check(lib().dd_device_sync())

@@ -125,7 +125,7 @@ struct dd_partitioner {
     uint8_t *out_valid[DD_KMAX_COLS] = {};
     uint32_t *out_lengths[DD_KMAX_COLS] = {};
 
-    hipEvent_t ev[4] = {};
+    hipEvent_t ev[5] = {}; /* 0 K1start 1 K1end 2 scans-end 3 K3end 4 K3start */
     bool has_run = false;
 
     ~dd_partitioner() {
@@ -372,7 +372,7 @@ extern "C" dd_status dd_partitioner_create(const dd_batch_desc *batch, const int
     return DD_OK;
 }
 
-extern "C" dd_status dd_partitioner_run(dd_partitioner *p, void *stream) {
+extern "C" dd_status dd_partitioner_run_phase1(dd_partitioner *p, void *stream) {
     if (!p) return set_err(DD_ERR_INVALID, "null partitioner");
     hipStream_t s = (hipStream_t)stream;
     HIP_TRY(hipEventRecord(p->ev[0], s));
@@ -395,6 +395,13 @@ extern "C" dd_status dd_partitioner_run(dd_partitioner *p, void *stream) {
         }
     }
     HIP_TRY(hipEventRecord(p->ev[2], s));
+    return DD_OK;
+}
+
+extern "C" dd_status dd_partitioner_run_phase2(dd_partitioner *p, void *stream) {
+    if (!p) return set_err(DD_ERR_INVALID, "null partitioner");
+    hipStream_t s = (hipStream_t)stream;
+    HIP_TRY(hipEventRecord(p->ev[4], s)); /* K3 start on the phase2 stream */
     if (p->staged) {
         HIP_TRY(dd_launch_scatter_staged(&p->ka, p->nchunks, p->chunk_rows, p->nparts,
                                          p->nbits, p->pid, p->counts, p->part_offsets,
@@ -416,6 +423,22 @@ extern "C" dd_status dd_partitioner_run(dd_partitioner *p, void *stream) {
     HIP_TRY(hipEventRecord(p->ev[3], s));
     p->has_run = true;
     return DD_OK;
+}
+
+extern "C" dd_status dd_partitioner_wait_phase1(dd_partitioner *p, void *stream) {
+    HIP_TRY(hipStreamWaitEvent((hipStream_t)stream, p->ev[2], 0));
+    return DD_OK;
+}
+
+extern "C" dd_status dd_partitioner_wait_phase2(dd_partitioner *p, void *stream) {
+    HIP_TRY(hipStreamWaitEvent((hipStream_t)stream, p->ev[3], 0));
+    return DD_OK;
+}
+
+extern "C" dd_status dd_partitioner_run(dd_partitioner *p, void *stream) {
+    dd_status st = dd_partitioner_run_phase1(p, stream);
+    if (st != DD_OK) return st;
+    return dd_partitioner_run_phase2(p, stream);
 }
 
 extern "C" void dd_partitioner_destroy(dd_partitioner *p) { delete p; }
@@ -456,7 +479,7 @@ extern "C" dd_status dd_partitioner_kernel_ms(const dd_partitioner *p, float out
     HIP_TRY(hipEventSynchronize(p->ev[3]));
     HIP_TRY(hipEventElapsedTime(&out_ms[0], p->ev[0], p->ev[1]));
     HIP_TRY(hipEventElapsedTime(&out_ms[1], p->ev[1], p->ev[2]));
-    HIP_TRY(hipEventElapsedTime(&out_ms[2], p->ev[2], p->ev[3]));
+    HIP_TRY(hipEventElapsedTime(&out_ms[2], p->ev[4], p->ev[3]));
     return DD_OK;
 }
 

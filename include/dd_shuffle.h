@@ -94,6 +94,16 @@ dd_status dd_partitioner_create(const dd_batch_desc *batch, const int32_t *key_c
 /* Runs K1 (hash+count), K2 (scan), K3 (stable scatter) on `stream` (a hipStream_t, or NULL
  * for the default stream). Asynchronous; results valid after stream sync. */
 dd_status dd_partitioner_run(dd_partitioner *p, void *stream);
+/* Phase split for batch pipelining (the reference streams batches through
+ * RepartitionExec continuously; overlapping batch s+1's hash/count with batch s's
+ * scatter mirrors that): phase1 = K1+K2 (+byte scans), phase2 = K3 (+K4). phase2 must be
+ * ordered after phase1 of the SAME partitioner (same stream, or an event). */
+dd_status dd_partitioner_run_phase1(dd_partitioner *p, void *stream);
+dd_status dd_partitioner_run_phase2(dd_partitioner *p, void *stream);
+/* make `stream` wait for this partitioner's last phase1 / phase2 completion (for
+ * cross-stream batch pipelining) */
+dd_status dd_partitioner_wait_phase1(dd_partitioner *p, void *stream);
+dd_status dd_partitioner_wait_phase2(dd_partitioner *p, void *stream);
 void dd_partitioner_destroy(dd_partitioner *p);
 
 /* result accessors (pointers are device memory owned by the partitioner) */
