@@ -736,6 +736,9 @@ struct dd_exchanged {
     int nranks = 1;
     uint32_t P = 0;
     int64_t total_rows = 0;
+    hipStream_t alloc_stream = nullptr; /* buffers are stream-ordered (hipMallocAsync):
+                                           repeated exchanges reuse the pool instead of
+                                           paying a device malloc per step */
     std::vector<int64_t> row_counts;                 /* [nranks][P] */
     std::vector<std::vector<int64_t>> byte_counts;   /* per col: [nranks][P] (var only) */
     void *data[DD_MAX_COLS] = {};
@@ -746,9 +749,9 @@ struct dd_exchanged {
     hipEvent_t e0 = nullptr, e1 = nullptr;
     ~dd_exchanged() {
         for (int i = 0; i < DD_MAX_COLS; i++) {
-            (void)hipFree(data[i]);
-            (void)hipFree(valid[i]);
-            (void)hipFree(lengths[i]);
+            if (data[i]) (void)hipFreeAsync(data[i], alloc_stream);
+            if (valid[i]) (void)hipFreeAsync(valid[i], alloc_stream);
+            if (lengths[i]) (void)hipFreeAsync(lengths[i], alloc_stream);
         }
         if (e0) (void)hipEventDestroy(e0);
         if (e1) (void)hipEventDestroy(e1);
@@ -795,6 +798,7 @@ extern "C" dd_status dd_exchange_run(dd_comm *c, const dd_partitioner *p, void *
     (void)hipFree(d_meta_all);
 
     auto e = new dd_exchanged();
+    e->alloc_stream = s;
     e->n_cols = p->batch.n_cols;
     e->nranks = R;
     e->P = P;
@@ -823,7 +827,7 @@ extern "C" dd_status dd_exchange_run(dd_comm *c, const dd_partitioner *p, void *
     for (int ci = 0; ci < p->batch.n_cols; ci++) {
         const dd_kcol &kc = p->ka.cols[ci];
         if (kc.elem > 0) {
-            if (hipMalloc(&e->data[ci], (size_t)e->total_rows * kc.elem + 1) != hipSuccess)
+            if (hipMallocAsync(&e->data[ci], (size_t)e->total_rows * kc.elem + 1, s) != hipSuccess)
                 return fail(DD_ERR_HIP, "recv alloc");
         } else {
             /* var col: find its v index, total bytes of my window */
@@ -840,13 +844,13 @@ extern "C" dd_status dd_exchange_run(dd_comm *c, const dd_partitioner *p, void *
                     recv_bytes_per_producer[ci][r] += b;
                     total_b += b;
                 }
-            if (hipMalloc(&e->data[ci], (size_t)total_b + 1) != hipSuccess)
+            if (hipMallocAsync(&e->data[ci], (size_t)total_b + 1, s) != hipSuccess)
                 return fail(DD_ERR_HIP, "recv alloc (var)");
-            if (hipMalloc((void **)&e->lengths[ci], (size_t)e->total_rows * 4 + 1) != hipSuccess)
+            if (hipMallocAsync((void **)&e->lengths[ci], (size_t)e->total_rows * 4 + 1, s) != hipSuccess)
                 return fail(DD_ERR_HIP, "recv alloc (lengths)");
         }
         if (kc.valid) {
-            if (hipMalloc((void **)&e->valid[ci], (size_t)e->total_rows + 1) != hipSuccess)
+            if (hipMallocAsync((void **)&e->valid[ci], (size_t)e->total_rows + 1, s) != hipSuccess)
                 return fail(DD_ERR_HIP, "recv alloc (validity)");
         }
     }
@@ -1135,6 +1139,7 @@ extern "C" dd_status dd_coalesce_run(dd_comm *c, const dd_partitioner *p,
     }
 
     auto e = new dd_exchanged();
+    e->alloc_stream = s;
     e->n_cols = p->batch.n_cols;
     e->nranks = glen; /* producers I consume */
     e->P = P;
@@ -1160,7 +1165,7 @@ extern "C" dd_status dd_coalesce_run(dd_comm *c, const dd_partitioner *p,
     for (int ci = 0; ci < p->batch.n_cols; ci++) {
         const dd_kcol &kc = p->ka.cols[ci];
         if (kc.elem > 0) {
-            if (hipMalloc(&e->data[ci], (size_t)e->total_rows * kc.elem + 1) != hipSuccess)
+            if (hipMallocAsync(&e->data[ci], (size_t)e->total_rows * kc.elem + 1, s) != hipSuccess)
                 return fail(DD_ERR_HIP, "coalesce recv alloc");
         } else {
             int v = -1;
@@ -1177,14 +1182,14 @@ extern "C" dd_status dd_coalesce_run(dd_comm *c, const dd_partitioner *p,
                     total_b += b;
                 }
             }
-            if (hipMalloc(&e->data[ci], (size_t)total_b + 1) != hipSuccess)
+            if (hipMallocAsync(&e->data[ci], (size_t)total_b + 1, s) != hipSuccess)
                 return fail(DD_ERR_HIP, "coalesce recv alloc (var)");
-            if (hipMalloc((void **)&e->lengths[ci], (size_t)e->total_rows * 4 + 1) !=
-                hipSuccess)
+            if (hipMallocAsync((void **)&e->lengths[ci], (size_t)e->total_rows * 4 + 1,
+                               s) != hipSuccess)
                 return fail(DD_ERR_HIP, "coalesce recv alloc (lengths)");
         }
         if (kc.valid) {
-            if (hipMalloc((void **)&e->valid[ci], (size_t)e->total_rows + 1) != hipSuccess)
+            if (hipMallocAsync((void **)&e->valid[ci], (size_t)e->total_rows + 1, s) != hipSuccess)
                 return fail(DD_ERR_HIP, "coalesce recv alloc (validity)");
         }
     }
