@@ -299,6 +299,8 @@ int main() {
     const size_t lds = P * 8 + (size_t)WPB * P * 4 + P * 4 + P * 4 + BT * 4 + (size_t)R * 4 +
                        (size_t)R * 28;
     printf("lds=%zu\n", lds);
+    fflush(stdout);
+    const char *only = getenv("ABLATE_ONLY");
 
     hipEvent_t e0, e1;
     HC(hipEventCreate(&e0));
@@ -323,13 +325,14 @@ int main() {
         HC(hipDeviceSynchronize());                                                          \
         float ms;                                                                            \
         HC(hipEventElapsedTime(&ms, e0, e1));                                                \
-        printf("%-28s %.3f ms\n", NAME, ms / 5);                                             \
+        printf("%-28s %.3f ms\n", NAME, ms / 5);                                         \
+        fflush(stdout);                                             \
     }
 
-    RUN(0, "full")
-    RUN(1, "no-flush-stores")
-    RUN(2, "no-column-loads")
-    RUN(3, "no-rank-machinery (timing)")
-    RUN(4, "no-LDS-staging (direct)")
+    if (!only || *only == '0') RUN(0, "full")
+    if (!only || *only == '1') RUN(1, "no-flush-stores")
+    if (!only || *only == '2') RUN(2, "no-column-loads")
+    if (!only || *only == '3') RUN(3, "no-rank-machinery (timing)")
+    if (!only || *only == '4') RUN(4, "no-LDS-staging (direct)")
     return 0;
 }
