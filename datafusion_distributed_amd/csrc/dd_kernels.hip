@@ -384,7 +384,10 @@ __device__ __forceinline__ void dd_block_excl_scan(const uint32_t *vals, uint32_
     __syncthreads();
 }
 
-template <int GMAX, int WPB, int MAXC, bool HASVAR>
+/* NC > 0: compile-time column count, no validity, no var (specialized fast path — the
+ * generic runtime column loop with break-guards and validity checks measured ~8-15%
+ * slower at the bench shape, tools/ablate_k3.cpp). NC == 0: generic. */
+template <int GMAX, int WPB, int MAXC, bool HASVAR, int NC = 0>
 __global__ __launch_bounds__(WPB * WAVE) void k_scatter_staged(
     dd_kargs a, int64_t tile_rows, uint32_t nparts, int nbits, const uint32_t *pid_in,
     const uint32_t *tile_off /* [nblocks][P] excl within partition */,
@@ -443,8 +446,8 @@ __global__ __launch_bounds__(WPB * WAVE) void k_scatter_staged(
             }
             pidr[g] = pid_in[row];
 #pragma unroll
-            for (int c = 0; c < MAXC; c++) {
-                if (c >= a.n_cols) break;
+            for (int c = 0; c < (NC > 0 ? NC : MAXC); c++) {
+                if (NC == 0 && c >= a.n_cols) break;
                 const dd_kcol &col = a.cols[c];
                 bool synthetic = false;
                 if constexpr (HASVAR) {
@@ -466,7 +469,7 @@ __global__ __launch_bounds__(WPB * WAVE) void k_scatter_staged(
                     case 8: colv[g][c] = ((const uint64_t *)col.data)[row]; break;
                     }
                 }
-                if (col.valid) valv[g][c] = col.valid[row];
+                if (NC == 0 && col.valid) valv[g][c] = col.valid[row];
             }
         }
     };
@@ -557,8 +560,8 @@ __global__ __launch_bounds__(WPB * WAVE) void k_scatter_staged(
             dstg[slot] = (uint32_t)(dstbase[pid] + rank_r);
             char *stage = stage0;
 #pragma unroll
-            for (int c = 0; c < MAXC; c++) {
-                if (c >= a.n_cols) break;
+            for (int c = 0; c < (NC > 0 ? NC : MAXC); c++) {
+                if (NC == 0 && c >= a.n_cols) break;
                 const dd_kcol &col = a.cols[c];
                 switch (col.elem) {
                 case 1: ((uint8_t *)stage)[slot] = (uint8_t)colv[g][c]; break;
@@ -567,7 +570,7 @@ __global__ __launch_bounds__(WPB * WAVE) void k_scatter_staged(
                 case 8: ((uint64_t *)stage)[slot] = colv[g][c]; break;
                 }
                 stage += (size_t)R * col.elem;
-                if (col.valid) {
+                if (NC == 0 && col.valid) {
                     ((uint8_t *)stage)[slot] = valv[g][c];
                     stage += R;
                 }
@@ -587,8 +590,8 @@ __global__ __launch_bounds__(WPB * WAVE) void k_scatter_staged(
             const uint64_t dst = dstg[i];
             char *stage = stage0;
 #pragma unroll
-            for (int c = 0; c < MAXC; c++) {
-                if (c >= a.n_cols) break;
+            for (int c = 0; c < (NC > 0 ? NC : MAXC); c++) {
+                if (NC == 0 && c >= a.n_cols) break;
                 const dd_kcol &col = a.cols[c];
                 switch (col.elem) {
                 case 1: ((uint8_t *)col.out_data)[dst] = ((const uint8_t *)stage)[i]; break;
@@ -603,7 +606,7 @@ __global__ __launch_bounds__(WPB * WAVE) void k_scatter_staged(
                     break;
                 }
                 stage += (size_t)R * col.elem;
-                if (col.valid) {
+                if (NC == 0 && col.valid) {
                     col.out_valid[dst] = ((const uint8_t *)stage)[i];
                     stage += R;
                 }
@@ -872,6 +875,33 @@ hipError_t dd_launch_scatter_staged(const dd_kargs *a, int64_t nblocks, int64_t 
     dim3 grid((unsigned)nblocks);
     const int maxc = (a->n_cols <= 4) ? 4 : 8;
     const bool hasvar = a->n_var > 0;
+    /* specialized fast path: wpb 16, <=4 fixed columns, no validity, no var */
+    bool can_spec = !hasvar && wpb == 16 && a->n_cols <= 4;
+    for (int c = 0; c < a->n_cols && can_spec; c++)
+        if (a->cols[c].valid || a->cols[c].elem == 0) can_spec = false;
+#define DD_SPEC(G, N)                                                                        \
+    if (can_spec && gmax == G && a->n_cols == N) {                                           \
+        if (lds_bytes > 65536) {                                                             \
+            hipError_t e =                                                                   \
+                hipFuncSetAttribute((const void *)k_scatter_staged<G, 16, 4, false, N>,      \
+                                    hipFuncAttributeMaxDynamicSharedMemorySize,              \
+                                    (int)lds_bytes);                                         \
+            if (e != hipSuccess) return e;                                                   \
+        }                                                                                    \
+        hipLaunchKernelGGL((k_scatter_staged<G, 16, 4, false, N>), grid, dim3(16 * WAVE),    \
+                           lds_bytes, s, *a, tile_rows, nparts, nbits, pid_in, tile_off,     \
+                           part_offsets);                                                    \
+        return hipGetLastError();                                                            \
+    }
+    DD_SPEC(2, 1)
+    DD_SPEC(2, 2)
+    DD_SPEC(2, 3)
+    DD_SPEC(2, 4)
+    DD_SPEC(4, 1)
+    DD_SPEC(4, 2)
+    DD_SPEC(4, 3)
+    DD_SPEC(4, 4)
+#undef DD_SPEC
 #define DD_CASE(G, W, C, V)                                                                  \
     if (gmax == G && wpb == W && maxc == C && hasvar == V) {                                 \
         if (lds_bytes > 65536) {                                                             \
