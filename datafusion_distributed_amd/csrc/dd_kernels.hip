@@ -474,17 +474,12 @@ __global__ __launch_bounds__(WPB * WAVE) void k_scatter_staged(
         }
     };
 
-    {
-        const int64_t rend0 = (tstart + R < tend) ? (tstart + R) : tend;
-        if (tstart < tend) preload(tstart, rend0);
-    }
-
-    for (int64_t rstart = tstart; rstart < tend; rstart += R) {
-        const int64_t rend = (rstart + R < tend) ? (rstart + R) : tend;
-        const int round_rows = (int)(rend - rstart);
-
-        /* rank (LDS-only; rows already in registers). Each wave zeroes and owns its own
-         * seghist row — no barrier needed before ranking. */
+    /* rank: LDS-only, consumes the preloaded pidr. Each wave zeroes and owns its own
+     * seghist row. Runs for round r+1 BEFORE the flush of round r: the compiler's
+     * vmcnt(0) drain at the first pidr use then hits only stores issued a full round ago
+     * (the ablation showed flush-store time fully serialized with the next round
+     * otherwise — profiles/r01_k3_ablation.json). */
+    auto rank = [&]() {
         for (uint32_t p = lane; p < nparts; p += WAVE) myseg[p] = 0;
 #pragma unroll
         for (int g = 0; g < GMAX; g++) {
@@ -505,7 +500,20 @@ __global__ __launch_bounds__(WPB * WAVE) void k_scatter_staged(
             }
             rankr[g] = rk;
         }
-        __syncthreads();
+    };
+
+    {
+        const int64_t rend0 = (tstart + R < tend) ? (tstart + R) : tend;
+        if (tstart < tend) {
+            preload(tstart, rend0);
+            rank();
+        }
+    }
+    __syncthreads();
+
+    for (int64_t rstart = tstart; rstart < tend; rstart += R) {
+        const int64_t rend = (rstart + R < tend) ? (rstart + R) : tend;
+        const int round_rows = (int)(rend - rstart);
 
         /* cross-wave exclusive scan per partition + round totals, fused with the
          * partition-offset scan (contiguous spans; one wave shfl-scans the span sums) */
@@ -578,10 +586,12 @@ __global__ __launch_bounds__(WPB * WAVE) void k_scatter_staged(
         }
         __syncthreads();
 
-        /* issue next round's loads BEFORE the flush so they overlap the stores */
+        /* round r+1: issue its loads AND rank it before the flush — the loads overlap
+         * the stores, and the pid-consumption drain lands on old stores */
         if (rstart + R < tend) {
             const int64_t nrend = (rstart + 2 * R < tend) ? (rstart + 2 * R) : tend;
             preload(rstart + R, nrend);
+            rank();
         }
 
         /* flush: consecutive LDS slots -> consecutive global rows within each partition
