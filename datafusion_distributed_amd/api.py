@@ -399,7 +399,8 @@ class Exchanged:
             self.h = None
 
 
-AGG_OPS = {"sum_f64": 0, "count": 1, "sum_i64": 2}
+AGG_OPS = {"sum_f64": 0, "count": 1, "sum_i64": 2, "min_f64": 3, "max_f64": 4,
+           "min_i64": 5, "max_i64": 6}
 
 
 def partial_reduce(batch: DeviceBatch, key_idx, aggs):

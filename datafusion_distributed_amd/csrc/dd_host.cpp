@@ -989,10 +989,16 @@ extern "C" dd_status dd_partial_reduce_run(const dd_batch_desc *batch,
         if (!agg_cols || agg_cols[g] < 0 || agg_cols[g] >= batch->n_cols)
             return set_err(DD_ERR_INVALID, "agg column out of range");
         int dt = batch->cols[agg_cols[g]].dtype;
-        if (agg_ops[g] == DD_AGG_SUM_F64 && dt != DD_DT_F64)
-            return set_err(DD_ERR_UNSUPPORTED, "SUM_F64 needs an f64 column");
-        if (agg_ops[g] == DD_AGG_SUM_I64 && dt != DD_DT_I64)
-            return set_err(DD_ERR_UNSUPPORTED, "SUM_I64 needs an i64 column");
+        if ((agg_ops[g] == DD_AGG_SUM_F64 || agg_ops[g] == DD_AGG_MIN_F64 ||
+             agg_ops[g] == DD_AGG_MAX_F64) &&
+            dt != DD_DT_F64)
+            return set_err(DD_ERR_UNSUPPORTED, "f64 aggregate needs an f64 column");
+        if ((agg_ops[g] == DD_AGG_SUM_I64 || agg_ops[g] == DD_AGG_MIN_I64 ||
+             agg_ops[g] == DD_AGG_MAX_I64) &&
+            dt != DD_DT_I64)
+            return set_err(DD_ERR_UNSUPPORTED, "i64 aggregate needs an i64 column");
+        if (agg_ops[g] < 0 || agg_ops[g] > DD_AGG_MAX_I64)
+            return set_err(DD_ERR_INVALID, "unknown aggregate op");
     }
 
     dd_kargs ka;

@@ -31,7 +31,43 @@
 #define R_PROBE 64      /* bounded probe; then spill */
 
 /* agg op codes (mirrored in include/dd_shuffle.h) */
-enum { DD_AGG_SUM_F64 = 0, DD_AGG_COUNT = 1, DD_AGG_SUM_I64 = 2 };
+enum {
+    DD_AGG_SUM_F64 = 0,
+    DD_AGG_COUNT = 1,
+    DD_AGG_SUM_I64 = 2,
+    DD_AGG_MIN_F64 = 3,
+    DD_AGG_MAX_F64 = 4,
+    DD_AGG_MIN_I64 = 5,
+    DD_AGG_MAX_I64 = 6,
+};
+
+/* order-preserving f64 <-> u64 map (IEEE total order; NaN sorts above +inf, matching
+ * Arrow's max semantics): x >= 0 -> bits | sign, x < 0 -> ~bits */
+__device__ __forceinline__ uint64_t dd_f64_key(double v) {
+    uint64_t b = __double_as_longlong(v);
+    return (b & 0x8000000000000000ULL) ? ~b : (b | 0x8000000000000000ULL);
+}
+__device__ __forceinline__ double dd_f64_unkey(uint64_t k) {
+    uint64_t b = (k & 0x8000000000000000ULL) ? (k & 0x7fffffffffffffffULL) : ~k;
+    return __longlong_as_double((long long)b);
+}
+__device__ __forceinline__ uint64_t dd_i64_key(int64_t v) {
+    return (uint64_t)v ^ 0x8000000000000000ULL; /* order-preserving signed -> unsigned */
+}
+__device__ __forceinline__ int64_t dd_i64_unkey(uint64_t k) {
+    return (int64_t)(k ^ 0x8000000000000000ULL);
+}
+
+__device__ __forceinline__ unsigned long long dd_agg_identity(int op) {
+    switch (op) {
+    case DD_AGG_SUM_F64: return (unsigned long long)__double_as_longlong(0.0);
+    case DD_AGG_MIN_F64:
+    case DD_AGG_MIN_I64: return ~0ULL; /* min over mapped-u64 keys */
+    case DD_AGG_MAX_F64:
+    case DD_AGG_MAX_I64: return 0ULL;
+    default: return 0ULL; /* COUNT / SUM_I64 */
+    }
+}
 
 /* canonical 64-bit key bits (zero-extended; floats canonicalized like the hash) */
 __device__ __forceinline__ uint64_t dd_key_bits(const dd_kcol &c, int64_t i) {
@@ -82,8 +118,7 @@ extern "C" __global__ __launch_bounds__(R_THREADS) void k_partial_reduce(
         t_ready[s] = 0;
         t_null[s] = 0;
         for (int k = 0; k < nk; k++) t_keys[k][s] = 0;
-        for (int g = 0; g < n_aggs; g++)
-            t_agg[g][s] = (agg_ops[g] == DD_AGG_SUM_F64) ? __double_as_longlong(0.0) : 0ull;
+        for (int g = 0; g < n_aggs; g++) t_agg[g][s] = dd_agg_identity(agg_ops[g]);
     }
     __syncthreads();
 
@@ -159,6 +194,28 @@ extern "C" __global__ __launch_bounds__(R_THREADS) void k_partial_reduce(
                               (unsigned long long)((const uint64_t *)c.data)[row]);
                     break;
                 }
+                case DD_AGG_MIN_F64:
+                case DD_AGG_MAX_F64: {
+                    const dd_kcol &c = a.cols[agg_cols[g]];
+                    if (c.valid && !c.valid[row]) break;
+                    uint64_t k = dd_f64_key(((const double *)c.data)[row]);
+                    if (agg_ops[g] == DD_AGG_MIN_F64)
+                        atomicMin(&t_agg[g][slot], (unsigned long long)k);
+                    else
+                        atomicMax(&t_agg[g][slot], (unsigned long long)k);
+                    break;
+                }
+                case DD_AGG_MIN_I64:
+                case DD_AGG_MAX_I64: {
+                    const dd_kcol &c = a.cols[agg_cols[g]];
+                    if (c.valid && !c.valid[row]) break;
+                    uint64_t k = dd_i64_key(((const int64_t *)c.data)[row]);
+                    if (agg_ops[g] == DD_AGG_MIN_I64)
+                        atomicMin(&t_agg[g][slot], (unsigned long long)k);
+                    else
+                        atomicMax(&t_agg[g][slot], (unsigned long long)k);
+                    break;
+                }
                 case DD_AGG_COUNT:
                     atomicAdd(&t_agg[g][slot], 1ull);
                     break;
@@ -176,6 +233,24 @@ extern "C" __global__ __launch_bounds__(R_THREADS) void k_partial_reduce(
                     const dd_kcol &c = a.cols[agg_cols[g]];
                     v = (c.valid && !c.valid[row]) ? 0.0 : ((const double *)c.data)[row];
                     out_aggs[o * n_aggs + g] = v;
+                    break;
+                }
+                case DD_AGG_MIN_F64:
+                case DD_AGG_MAX_F64: {
+                    const dd_kcol &c = a.cols[agg_cols[g]];
+                    const bool null = c.valid && !c.valid[row];
+                    out_aggs[o * n_aggs + g] =
+                        null ? dd_f64_unkey(dd_agg_identity(agg_ops[g]))
+                             : ((const double *)c.data)[row];
+                    break;
+                }
+                case DD_AGG_MIN_I64:
+                case DD_AGG_MAX_I64: {
+                    const dd_kcol &c = a.cols[agg_cols[g]];
+                    const bool null = c.valid && !c.valid[row];
+                    out_aggs[o * n_aggs + g] = __longlong_as_double(
+                        null ? dd_i64_unkey(dd_agg_identity(agg_ops[g]))
+                             : ((const int64_t *)c.data)[row]);
                     break;
                 }
                 case DD_AGG_SUM_I64: {
@@ -201,8 +276,21 @@ extern "C" __global__ __launch_bounds__(R_THREADS) void k_partial_reduce(
         uint64_t o = atomicAdd((unsigned long long *)out_n, 1ull);
         for (int k = 0; k < nk; k++) out_keys[o * nk + k] = t_keys[k][s];
         out_keynull[o] = t_null[s];
-        for (int g = 0; g < n_aggs; g++)
-            out_aggs[o * n_aggs + g] = __longlong_as_double((long long)t_agg[g][s]);
+        for (int g = 0; g < n_aggs; g++) {
+            unsigned long long st = t_agg[g][s];
+            switch (agg_ops[g]) {
+            case DD_AGG_MIN_F64:
+            case DD_AGG_MAX_F64:
+                out_aggs[o * n_aggs + g] = dd_f64_unkey(st);
+                break;
+            case DD_AGG_MIN_I64:
+            case DD_AGG_MAX_I64:
+                out_aggs[o * n_aggs + g] = __longlong_as_double(dd_i64_unkey(st));
+                break;
+            default:
+                out_aggs[o * n_aggs + g] = __longlong_as_double((long long)st);
+            }
+        }
     }
 }
 

@@ -222,6 +222,10 @@ typedef struct dd_reducer dd_reducer;
 #define DD_AGG_SUM_F64 0 /* sum of an f64 column (nulls skipped) */
 #define DD_AGG_COUNT 1   /* count(*) (agg_col ignored) */
 #define DD_AGG_SUM_I64 2 /* sum of an i64 column (nulls skipped) */
+#define DD_AGG_MIN_F64 3 /* min/max skip nulls; NaN sorts above +inf (Arrow semantics); */
+#define DD_AGG_MAX_F64 4 /* an all-null group yields the identity (0 / +-extreme), not   */
+#define DD_AGG_MIN_I64 5 /* SQL NULL — pair with count(col) downstream to reconstruct    */
+#define DD_AGG_MAX_I64 6 /* null results */
 
 dd_status dd_partial_reduce_run(const dd_batch_desc *batch, const int32_t *key_cols,
                                 int32_t n_keys, const int32_t *agg_cols,

@@ -136,3 +136,44 @@ def test_partial_reduce_null_keys_and_null_aggs():
         assert got[1] == direct["count_all"][i].as_py()
         want = direct["v_sum"][i].as_py() or 0.0
         assert abs(got[0] - want) <= 1e-6 * max(abs(want), 1.0)
+
+
+def test_partial_reduce_min_max():
+    import pyarrow as pa
+
+    rng = np.random.default_rng(37)
+    n = 500_000
+    k = rng.integers(0, 64, n, dtype=np.int64)
+    f = rng.normal(size=n) * 1000
+    i = rng.integers(-(2**60), 2**60, n, dtype=np.int64)
+    batch = api.DeviceBatch([
+        {"dtype": "i64", "data": k, "valid": None},
+        {"dtype": "f64", "data": f, "valid": None},
+        {"dtype": "i64", "data": i, "valid": None},
+    ])
+    res = api.partial_reduce(batch, [0], [(1, "min_f64"), (1, "max_f64"),
+                                          (2, "min_i64"), (2, "max_i64")])
+    batch.free()
+    # final merge: min of mins, max of maxes
+    merged = {}
+    for r in range(len(res["keynull"])):
+        key = int(res["keys"][r, 0])
+        v = res["aggs"][r]
+        iv = res["aggs"][r].view(np.int64)
+        cur = merged.setdefault(key, [np.inf, -np.inf, 2**63 - 1, -(2**63)])
+        cur[0] = min(cur[0], v[0])
+        cur[1] = max(cur[1], v[1])
+        cur[2] = min(cur[2], int(iv[2]))
+        cur[3] = max(cur[3], int(iv[3]))
+
+    tbl = pa.table({"k": k, "f": f, "i": i})
+    direct = tbl.group_by("k").aggregate([("f", "min"), ("f", "max"), ("i", "min"),
+                                          ("i", "max")])
+    assert len(merged) == direct.num_rows
+    for r in range(direct.num_rows):
+        key = direct["k"][r].as_py()
+        got = merged[key]
+        assert got[0] == direct["f_min"][r].as_py()
+        assert got[1] == direct["f_max"][r].as_py()
+        assert got[2] == direct["i_min"][r].as_py()
+        assert got[3] == direct["i_max"][r].as_py()
