@@ -72,6 +72,16 @@ def make_lineitem_q1(rows, seed):
     return cols, [0, 1]
 
 
+def make_lineitem_q3_multikey(rows, seed):
+    """BASELINE config[2]'s 'multi-key hash shuffle': l_orderkey + l_suppkey composite
+    key over the q3/q5 projection (exercises the multi-column combine on the GPU)."""
+    cols, _ = make_lineitem_q3(rows, seed)
+    rng = np.random.default_rng(seed + 1)
+    cols.append({"dtype": "i32", "data": rng.integers(1, 100_001, rows, dtype=np.int64)
+                 .astype(np.int32), "valid": None})
+    return cols, [0, 4]
+
+
 def make_clickbench_userid(rows, seed):
     """BASELINE config[4] stand-in: GROUP BY UserID — Zipf(1.1)-skewed i64 key + wide
     var-width URL column (exercises the v1 var-width path)."""
@@ -93,6 +103,7 @@ def make_clickbench_userid(rows, seed):
 
 WORKLOADS = {
     "tpch_sf10_lineitem_shuffle": (make_lineitem_q3, SF10_ROWS),
+    "tpch_sf10_multikey_shuffle": (make_lineitem_q3_multikey, SF10_ROWS),
     "tpch_sf1_q1_repartition": (make_lineitem_q1, 6_001_215),
     "clickbench_userid_shuffle": (make_clickbench_userid, 20_000_000),
 }
