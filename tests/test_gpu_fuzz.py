@@ -58,3 +58,28 @@ def test_fuzz_parity(case):
     key_idx = list(rng.choice(ncols, nkeys, replace=False))
     nparts = int(rng.choice([1, 2, 3, 7, 8, 16, 100, 128, 777, 2048]))
     check_against_oracle(cols, [int(k) for k in key_idx], nparts)
+
+
+KNOBS = [  # (wpb, gmax, dd_v2_var)
+    ("4", "2", "0"), ("4", "8", "1"), ("8", "2", "1"),
+    ("8", "8", "0"), ("16", "2", "1"), ("16", "4", "0"),
+]
+
+
+@pytest.mark.parametrize("knob", KNOBS)
+def test_fuzz_parity_tuning_knobs(knob, monkeypatch):
+    """Every tuning configuration (waves-per-block, rows-per-round, var path) must stay
+    bit-exact — the knobs change scheduling, never results."""
+    wpb, gmax, var = knob
+    monkeypatch.setenv("DD_V2_WPB", wpb)
+    monkeypatch.setenv("DD_V2_GMAX", gmax)
+    monkeypatch.setenv("DD_V2_VAR", var)
+    rng = np.random.default_rng(7000 + int(wpb) * 10 + int(gmax))
+    n = 120000
+    cols = [
+        random_col(rng, n, "i64", 0.1),
+        random_col(rng, n, "utf8", 0.2),
+        random_col(rng, n, "f64", 0),
+        random_col(rng, n, "bool", 0.3),
+    ]
+    check_against_oracle(cols, [0, 3], 32)
