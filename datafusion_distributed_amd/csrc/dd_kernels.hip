@@ -912,6 +912,33 @@ hipError_t dd_launch_scatter_staged(const dd_kargs *a, int64_t nblocks, int64_t 
     DD_SPEC(4, 3)
     DD_SPEC(4, 4)
 #undef DD_SPEC
+    /* 5-8 fixed columns: same fast path at MAXC=8 */
+#define DD_SPEC8(G, N)                                                                       \
+    if (can_spec8 && gmax == G && a->n_cols == N) {                                          \
+        if (lds_bytes > 65536) {                                                             \
+            hipError_t e =                                                                   \
+                hipFuncSetAttribute((const void *)k_scatter_staged<G, 16, 8, false, N>,      \
+                                    hipFuncAttributeMaxDynamicSharedMemorySize,              \
+                                    (int)lds_bytes);                                         \
+            if (e != hipSuccess) return e;                                                   \
+        }                                                                                    \
+        hipLaunchKernelGGL((k_scatter_staged<G, 16, 8, false, N>), grid, dim3(16 * WAVE),    \
+                           lds_bytes, s, *a, tile_rows, nparts, nbits, pid_in, tile_off,     \
+                           part_offsets);                                                    \
+        return hipGetLastError();                                                            \
+    }
+    bool can_spec8 = !hasvar && wpb == 16 && a->n_cols >= 5 && a->n_cols <= 8;
+    for (int c = 0; c < a->n_cols && can_spec8; c++)
+        if (a->cols[c].valid || a->cols[c].elem == 0) can_spec8 = false;
+    DD_SPEC8(2, 5)
+    DD_SPEC8(2, 6)
+    DD_SPEC8(2, 7)
+    DD_SPEC8(2, 8)
+    DD_SPEC8(4, 5)
+    DD_SPEC8(4, 6)
+    DD_SPEC8(4, 7)
+    DD_SPEC8(4, 8)
+#undef DD_SPEC8
 #define DD_CASE(G, W, C, V)                                                                  \
     if (gmax == G && wpb == W && maxc == C && hasvar == V) {                                 \
         if (lds_bytes > 65536) {                                                             \
