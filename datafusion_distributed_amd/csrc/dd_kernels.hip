@@ -7,17 +7,20 @@
  * contiguous outputs. Semantics are the normative spec of DESIGN.md §3 (bit-exact vs
  * oracle/dd_oracle.c).
  *
- * Decomposition: the input is split into contiguous row "chunks", one per
- * 64-lane wave; blocks are 256 threads = 4 independent waves (no intra-block barriers on the
- * hot path). Stability: chunks are ordered, a wave walks its chunk's 64-row groups in order,
- * and lane order == row order inside a group; ballot-multisplit (log2(P) ballots) gives the
- * stable intra-group rank.
+ * Two kernel paths share the scan infrastructure (host picks in dd_host.cpp):
  *
- *   K1 hash_count : h -> pid (stored u32) + per-(chunk, partition) row/byte histograms
- *   K2 scan       : column-major exclusive scan of the histograms (3 passes, coalesced)
- *   K3 scatter    : stable scatter of every column into partition-major output buffers
+ *   v2 "staged" (default; fixed-width columns, and var-width via synthetic VARLEN/ROWID
+ *   columns + the K4 byte pass): one contiguous row TILE per block; rounds of
+ *   R = gmax*wpb*64 rows ranked with ballot-multisplit, staged partition-major in LDS,
+ *   flushed with run-granular coalesced stores. k_hash_count_tile + k_scan_* +
+ *   k_scatter_staged (+ K4). Stability: tiles ordered, wave segments contiguous within a
+ *   round, lane order == row order inside a 64-row group.
  *
- * All integer/byte work — HBM-bound; no MFMA (this is indexing, not GEMM-shaped work).
+ *   v1 "direct" (fallback beyond DD_STAGE_MAXC columns): chunk per WAVE, per-wave LDS
+ *   bases, direct scatter. k_hash_count + k_scan_* + k_scatter. Same stability guarantees.
+ *
+ * All integer/byte work — HBM-bound; no MFMA (indexing, not GEMM-shaped work).
+ * Optimization history and PMC evidence: DESIGN.md §9, profiles/.
  */
 
 #include <hip/hip_runtime.h>
