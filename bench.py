@@ -82,6 +82,23 @@ def make_lineitem_q3_multikey(rows, seed):
     return cols, [0, 4]
 
 
+def make_lineitem_utf8key(rows, seed):
+    """SURVEY §8(d) microbench variant '2-key (int64+utf8(16B avg))': composite i64+utf8
+    key over the q3 payload — exercises the chunked byte hash at full scale."""
+    rng = np.random.default_rng(seed)
+    lens = rng.integers(8, 25, rows)  # ~16 B avg
+    off = np.zeros(rows + 1, dtype=np.int32)
+    off[1:] = np.cumsum(lens)
+    data = rng.integers(48, 122, int(off[-1]), dtype=np.int64).astype(np.uint8)
+    return [
+        {"dtype": "i64", "data": rng.integers(1, 60_000_000, rows, dtype=np.int64),
+         "valid": None},
+        {"dtype": "utf8", "data": data, "offsets": off, "valid": None},
+        {"dtype": "f64", "data": rng.uniform(900.0, 105000.0, rows), "valid": None},
+        {"dtype": "f64", "data": rng.uniform(0.0, 0.1, rows), "valid": None},
+    ], [0, 1]
+
+
 def make_clickbench_userid(rows, seed):
     """BASELINE config[4] stand-in: GROUP BY UserID — Zipf(1.1)-skewed i64 key + wide
     var-width URL column (exercises the v1 var-width path)."""
@@ -104,6 +121,7 @@ def make_clickbench_userid(rows, seed):
 WORKLOADS = {
     "tpch_sf10_lineitem_shuffle": (make_lineitem_q3, SF10_ROWS),
     "tpch_sf10_multikey_shuffle": (make_lineitem_q3_multikey, SF10_ROWS),
+    "tpch_sf10_utf8key_shuffle": (make_lineitem_utf8key, 30_000_000),
     "tpch_sf1_q1_repartition": (make_lineitem_q1, 6_001_215),
     "clickbench_userid_shuffle": (make_clickbench_userid, 20_000_000),
 }
