@@ -1,0 +1,157 @@
+"""chunked.py — partition inputs beyond the single-launch caps (2^32 rows, 4 GB of var
+bytes per column — DESIGN.md §5) by row-chunking and device-side concatenation.
+
+Chunks are processed in input order and each partition's chunk segments are concatenated
+in chunk order, so the global result is exactly the stable partition of the whole input
+(bit-identical to running one oversized launch / the oracle). This is the harness-level
+orchestration of SURVEY §8(d)'s "chunked to fit" SF100/ClickBench configs; memory peak is
+final outputs + one in-flight chunk.
+"""
+
+import ctypes
+
+import numpy as np
+
+from . import api
+
+
+def _final_alloc(nbytes):
+    p = ctypes.c_void_p()
+    api._check(api.lib().dd_dev_alloc(ctypes.c_int64(max(nbytes, 1)), ctypes.byref(p)))
+    return p
+
+
+def chunked_partition(cols, key_idx, nparts, max_chunk_rows=2**31,
+                      max_chunk_var_bytes=3 << 30):
+    """Returns {"part_row_offsets", "cols": [{"dtype", "data"(np), "valid", "lengths"}...]}
+    — downloaded final buffers (harness/test use; a production caller would keep the
+    device pointers)."""
+    n = (len(cols[0]["offsets"]) - 1 if cols[0]["dtype"] == "utf8" else len(cols[0]["data"]))
+
+    # choose chunk boundaries respecting both caps
+    bounds = [0]
+    while bounds[-1] < n:
+        lo = bounds[-1]
+        hi = min(n, lo + max_chunk_rows)
+        for c in cols:
+            if c["dtype"] == "utf8":
+                off = c["offsets"]
+                # largest hi with off[hi]-off[lo] <= cap
+                cap_hi = int(np.searchsorted(off, off[lo] + max_chunk_var_bytes,
+                                             side="right")) - 1
+                hi = min(hi, max(cap_hi, lo + 1))
+        bounds.append(hi)
+    nchunks = len(bounds) - 1
+
+    # pass 1: partition each chunk, collect per-partition row/byte counts
+    chunk_results = []
+    rows_cp = np.zeros((nchunks, nparts), dtype=np.int64)
+    bytes_cp = {i: np.zeros((nchunks, nparts), dtype=np.int64)
+                for i, c in enumerate(cols) if c["dtype"] == "utf8"}
+
+    def slice_chunk(ci):
+        lo, hi = bounds[ci], bounds[ci + 1]
+        out = []
+        for c in cols:
+            sc = dict(c)
+            if c["dtype"] == "utf8":
+                off = c["offsets"]
+                sc["offsets"] = (off[lo:hi + 1] - off[lo]).astype(np.int32)
+                sc["data"] = c["data"][int(off[lo]):int(off[hi])]
+            else:
+                sc["data"] = c["data"][lo:hi]
+            if c.get("valid") is not None:
+                sc["valid"] = c["valid"][lo:hi]
+            out.append(sc)
+        return out
+
+    # final buffer sizes
+    total_rows = n
+    elem = {i: api.ELEM_SIZE[c["dtype"]] for i, c in enumerate(cols)
+            if c["dtype"] != "utf8"}
+    var_total = {i: int(np.asarray(c["data"]).nbytes) for i, c in enumerate(cols)
+                 if c["dtype"] == "utf8"}
+    final = {}
+    for i, c in enumerate(cols):
+        if c["dtype"] == "utf8":
+            final[i] = {"data": _final_alloc(var_total[i]),
+                        "lengths": _final_alloc(total_rows * 4)}
+        else:
+            final[i] = {"data": _final_alloc(total_rows * elem[i])}
+        if c.get("valid") is not None:
+            final[i]["valid"] = _final_alloc(total_rows)
+
+    L = api.lib()
+
+    def d2d(dst, dst_off, src, src_off, nbytes):
+        if nbytes:
+            api._check(L.dd_memcpy_d2d(ctypes.c_void_p(dst.value + dst_off),
+                                       ctypes.c_void_p(src + src_off),
+                                       ctypes.c_int64(nbytes)))
+
+    # process chunks sequentially: partition, then copy each partition segment out
+    part_meta = []
+    for ci in range(nchunks):
+        ccols = slice_chunk(ci)
+        batch = api.DeviceBatch(ccols)
+        part = api.Partitioner(batch, key_idx, nparts)
+        part.run()
+        part.sync()
+        roff = part.row_offsets()
+        rows_cp[ci] = roff[1:] - roff[:-1]
+        boffs = {}
+        for i in bytes_cp:
+            bo = part.byte_offsets(i)
+            bytes_cp[i][ci] = bo[1:] - bo[:-1]
+            boffs[i] = bo
+        part_meta.append((batch, part, roff, boffs))
+
+    # global offsets: partition-major, chunk order within a partition (stable)
+    part_rows = rows_cp.sum(axis=0)
+    grow = np.zeros(nparts + 1, dtype=np.int64)
+    np.cumsum(part_rows, out=grow[1:])
+    gbyte = {i: np.concatenate([[0], np.cumsum(bytes_cp[i].sum(axis=0))])
+             for i in bytes_cp}
+
+    for ci in range(nchunks):
+        batch, part, roff, boffs = part_meta[ci]
+        row_prior = rows_cp[:ci].sum(axis=0) if ci else np.zeros(nparts, dtype=np.int64)
+        for p in range(nparts):
+            gdst_row = int(grow[p] + row_prior[p])
+            src_lo, src_hi = int(roff[p]), int(roff[p + 1])
+            nrows = src_hi - src_lo
+            for i, c in enumerate(cols):
+                if c["dtype"] == "utf8":
+                    byte_prior = bytes_cp[i][:ci, p].sum() if ci else 0
+                    b_lo = int(boffs[i][p])
+                    nb = int(boffs[i][p + 1]) - b_lo
+                    d2d(final[i]["data"], int(gbyte[i][p] + byte_prior),
+                        L.dd_partitioner_col_data(part.h, i), b_lo, nb)
+                    d2d(final[i]["lengths"], gdst_row * 4,
+                        L.dd_partitioner_col_lengths(part.h, i), src_lo * 4, nrows * 4)
+                else:
+                    d2d(final[i]["data"], gdst_row * elem[i],
+                        L.dd_partitioner_col_data(part.h, i), src_lo * elem[i],
+                        nrows * elem[i])
+                if c.get("valid") is not None:
+                    d2d(final[i]["valid"], gdst_row,
+                        L.dd_partitioner_col_validity(part.h, i), src_lo, nrows)
+        part.destroy()
+        batch.free()
+
+    # download for the harness caller
+    out_cols = []
+    for i, c in enumerate(cols):
+        oc = {"dtype": c["dtype"]}
+        if c["dtype"] == "utf8":
+            oc["data"] = api._d2h(final[i]["data"].value, var_total[i], np.uint8)
+            oc["lengths"] = api._d2h(final[i]["lengths"].value, total_rows * 4, np.uint32)
+        else:
+            npdt = api.FIXED_NP[c["dtype"]]
+            oc["data"] = api._d2h(final[i]["data"].value, total_rows * elem[i], npdt)
+        if c.get("valid") is not None:
+            oc["valid"] = api._d2h(final[i]["valid"].value, total_rows, np.uint8)
+        out_cols.append(oc)
+        for buf in final[i].values():
+            L.dd_dev_free(buf)
+    return {"part_row_offsets": grow, "cols": out_cols, "n_chunks": nchunks}

@@ -83,6 +83,10 @@ extern "C" dd_status dd_memcpy_d2h(void *dst, const void *src, int64_t bytes) {
     HIP_TRY(hipMemcpy(dst, src, (size_t)bytes, hipMemcpyDeviceToHost));
     return DD_OK;
 }
+extern "C" dd_status dd_memcpy_d2d(void *dst, const void *src, int64_t bytes) {
+    HIP_TRY(hipMemcpy(dst, src, (size_t)bytes, hipMemcpyDeviceToDevice));
+    return DD_OK;
+}
 extern "C" dd_status dd_device_sync(void) {
     HIP_TRY(hipDeviceSynchronize());
     return DD_OK;

@@ -244,6 +244,7 @@ dd_status dd_dev_alloc(int64_t bytes, void **out);
 dd_status dd_dev_free(void *p);
 dd_status dd_memcpy_h2d(void *dst, const void *src, int64_t bytes);
 dd_status dd_memcpy_d2h(void *dst, const void *src, int64_t bytes);
+dd_status dd_memcpy_d2d(void *dst, const void *src, int64_t bytes);
 dd_status dd_device_sync(void);
 
 #ifdef __cplusplus
