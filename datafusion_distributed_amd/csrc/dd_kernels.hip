@@ -819,7 +819,9 @@ hipError_t dd_launch_var_bytes(const uint32_t *lens, const uint32_t *src_row,
     hipLaunchKernelGGL(k4_off_rewrite, dim3(wavegrid), dim3(256), 0, s, lens, n, partials,
                        out_off);
     int copy_blocks = (int)((n + 255) / 256);
-    if (copy_blocks > 2048) copy_blocks = 2048;
+    if (copy_blocks > 8192) copy_blocks = 8192; /* gather-copy is latency-bound: deep
+                                                   oversubscription hides the random
+                                                   string reads */
     if (copy_blocks < 1) copy_blocks = 1;
     hipLaunchKernelGGL(k4_copy, dim3(copy_blocks), dim3(256), 0, s, src_row, out_off,
                        in_offsets, in_bytes, n, out_bytes);
