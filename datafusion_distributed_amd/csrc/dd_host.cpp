@@ -192,6 +192,8 @@ extern "C" dd_status dd_partitioner_create(const dd_batch_desc *batch, const int
     ka.n_rows = n;
     ka.n_cols = batch->n_cols;
     ka.n_keys = n_keys;
+    ka.pid_total = n_partitions;
+    ka.pid_shift = 0;
 
     auto fail = [&](dd_status s, const char *m) {
         delete p;
@@ -387,6 +389,27 @@ extern "C" dd_status dd_partitioner_create(const dd_batch_desc *batch, const int
     return DD_OK;
 }
 
+/* coarse-bucket partitioner: pid = (h % pid_total) >> log2(coarse_div). The effective
+ * partition count is pid_total/coarse_div CONTIGUOUS ranges of the final partition space
+ * — pass A of the two-level var scatter. Power-of-two args only. */
+extern "C" dd_status dd_partitioner_create_ranged(const dd_batch_desc *batch,
+                                                  const int32_t *key_cols, int32_t n_keys,
+                                                  uint32_t pid_total, uint32_t coarse_div,
+                                                  dd_partitioner **out) {
+    if (coarse_div == 0 || (coarse_div & (coarse_div - 1)) != 0 ||
+        (pid_total & (pid_total - 1)) != 0 || pid_total % coarse_div != 0)
+        return set_err(DD_ERR_INVALID, "pid_total and coarse_div must be powers of two "
+                                       "with coarse_div | pid_total");
+    dd_status st = dd_partitioner_create(batch, key_cols, n_keys, pid_total / coarse_div,
+                                         out);
+    if (st != DD_OK) return st;
+    (*out)->ka.pid_total = pid_total;
+    int shift = 0;
+    while ((1u << shift) < coarse_div) shift++;
+    (*out)->ka.pid_shift = shift;
+    return DD_OK;
+}
+
 extern "C" dd_status dd_partitioner_run_phase1(dd_partitioner *p, void *stream) {
     if (!p) return set_err(DD_ERR_INVALID, "null partitioner");
     hipStream_t s = (hipStream_t)stream;
@@ -473,6 +496,21 @@ extern "C" dd_status dd_partitioner_row_offsets(const dd_partitioner *p, int64_t
     if (!p->has_run) return set_err(DD_ERR_INVALID, "partitioner has not run");
     HIP_TRY(hipMemcpy(host_out, p->part_offsets, (size_t)(p->nparts + 1) * 8,
                       hipMemcpyDeviceToHost));
+    return DD_OK;
+}
+
+extern "C" const uint64_t *dd_partitioner_var_offsets64(const dd_partitioner *p,
+                                                        int32_t col) {
+    if (!p->staged) return nullptr;
+    for (int v = 0; v < p->ka.n_var; v++)
+        if (p->ka.var_idx[v] == col) return p->out_off[v];
+    return nullptr;
+}
+
+extern "C" dd_status dd_make_offsets32(const uint64_t *off64, int64_t lo_row, int64_t n,
+                                       int32_t *out32, void *stream) {
+    if (!off64 || !out32) return set_err(DD_ERR_INVALID, "null argument");
+    HIP_TRY(dd_launch_off64_to_off32(off64, lo_row, n, out32, (hipStream_t)stream));
     return DD_OK;
 }
 

@@ -46,6 +46,8 @@ struct dd_kargs {
     int32_t n_cols;
     int32_t n_keys;
     int32_t n_var;
+    uint32_t pid_total; /* pid = (h % pid_total) >> pid_shift; nparts = pid_total >> shift */
+    int32_t pid_shift;  /* 0 for plain partitioning; >0 = contiguous coarse buckets */
     int32_t key_idx[DD_KMAX_KEYS];
     int32_t var_idx[DD_KMAX_VAR];
     dd_kcol cols[DD_KMAX_COLS];
@@ -72,6 +74,8 @@ hipError_t dd_launch_scatter(const dd_kargs *a, int64_t nchunks, int64_t chunk_r
                              const uint32_t *chunk_off, const uint64_t *part_offsets,
                              const uint32_t *chunk_boff, const uint64_t *part_boffsets,
                              size_t lds_bytes, hipStream_t s);
+hipError_t dd_launch_off64_to_off32(const uint64_t *off64, int64_t lo, int64_t n,
+                                    int32_t *out32, hipStream_t s);
 hipError_t dd_launch_var_bytes(const uint32_t *lens, const uint32_t *src_row,
                                const int32_t *in_offsets, const uint8_t *in_bytes,
                                int64_t n, int64_t total_bytes, uint64_t *partials,

@@ -70,9 +70,12 @@ __global__ __launch_bounds__(BLOCK_THREADS) void k_hash_count(
         uint32_t pid = 0;
         if (active) {
             uint64_t h = dd_row_hash(a, row);
-            /* mask == mod for power-of-two P (the common case); u64 division otherwise */
-            pid = ((nparts & (nparts - 1)) == 0) ? (uint32_t)(h & (uint64_t)(nparts - 1))
-                                                 : (uint32_t)(h % (uint64_t)nparts);
+            /* pid = (h % pid_total) >> pid_shift (coarse-bucket mode when shift > 0);
+             * mask == mod for power-of-two totals (the common case) */
+            const uint32_t tot = a.pid_total;
+            uint32_t fine = ((tot & (tot - 1)) == 0) ? (uint32_t)(h & (uint64_t)(tot - 1))
+                                                     : (uint32_t)(h % (uint64_t)tot);
+            pid = fine >> a.pid_shift;
             pid_out[row] = pid;
         }
         uint64_t act = __ballot(active);
@@ -325,10 +328,12 @@ __global__ __launch_bounds__(BLOCK_THREADS) void k_hash_count_tile(
             pidu[u] = 0;
             if (actu[u]) {
                 uint64_t h = dd_row_hash(a, row);
-                /* mask == mod for power-of-two P; u64 division otherwise */
-                pidu[u] = ((nparts & (nparts - 1)) == 0)
-                              ? (uint32_t)(h & (uint64_t)(nparts - 1))
-                              : (uint32_t)(h % (uint64_t)nparts);
+                /* pid = (h % pid_total) >> pid_shift (coarse-bucket mode) */
+                const uint32_t tot = a.pid_total;
+                uint32_t fine = ((tot & (tot - 1)) == 0)
+                                    ? (uint32_t)(h & (uint64_t)(tot - 1))
+                                    : (uint32_t)(h % (uint64_t)tot);
+                pidu[u] = fine >> a.pid_shift;
                 pid_out[row] = pidu[u];
             }
         }
@@ -789,6 +794,12 @@ __global__ __launch_bounds__(BLOCK_THREADS) void k4_copy(
     }
 }
 
+__global__ void k_off64_to_off32(const uint64_t *off64, int64_t lo, int64_t n,
+                                 int32_t *out32) {
+    const int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    if (i <= n) out32[i] = (int32_t)(off64[lo + i] - off64[lo]);
+}
+
 __global__ void k4_part_boffsets(const uint64_t *out_off, const uint64_t *part_offsets,
                                  uint32_t nparts, int64_t n, int64_t total_bytes,
                                  uint64_t *part_boffsets) {
@@ -805,6 +816,14 @@ __global__ void k4_part_boffsets(const uint64_t *out_off, const uint64_t *part_o
 /* ---------------- launchers (called from dd_host.cpp) ---------------- */
 
 extern "C" {
+
+hipError_t dd_launch_off64_to_off32(const uint64_t *off64, int64_t lo, int64_t n,
+                                    int32_t *out32, hipStream_t s) {
+    int64_t total = n + 1;
+    int blocks = (int)((total + 255) / 256);
+    hipLaunchKernelGGL(k_off64_to_off32, dim3(blocks), dim3(256), 0, s, off64, lo, n, out32);
+    return hipGetLastError();
+}
 
 hipError_t dd_launch_var_bytes(const uint32_t *lens, const uint32_t *src_row,
                                const int32_t *in_offsets, const uint8_t *in_bytes,

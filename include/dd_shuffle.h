@@ -94,6 +94,12 @@ dd_status dd_partitioner_create(const dd_batch_desc *batch, const int32_t *key_c
 /* Runs K1 (hash+count), K2 (scan), K3 (stable scatter) on `stream` (a hipStream_t, or NULL
  * for the default stream). Asynchronous; results valid after stream sync. */
 dd_status dd_partitioner_run(dd_partitioner *p, void *stream);
+/* coarse-bucket variant: pid = (h % pid_total) / coarse_div — pid_total/coarse_div
+ * CONTIGUOUS ranges of the final partition space (pass A of the two-level var scatter,
+ * DESIGN.md §11; both arguments must be powers of two) */
+dd_status dd_partitioner_create_ranged(const dd_batch_desc *batch, const int32_t *key_cols,
+                                       int32_t n_keys, uint32_t pid_total,
+                                       uint32_t coarse_div, dd_partitioner **out);
 /* Phase split for batch pipelining (the reference streams batches through
  * RepartitionExec continuously; overlapping batch s+1's hash/count with batch s's
  * scatter mirrors that): phase1 = K1+K2 (+byte scans), phase2 = K3 (+K4). phase2 must be
@@ -113,6 +119,13 @@ const uint8_t *dd_partitioner_col_validity(const dd_partitioner *p, int32_t col)
 const uint32_t *dd_partitioner_col_lengths(const dd_partitioner *p, int32_t col); /* utf8 */
 /* copies part_row_offsets[P+1] (rows) to host */
 dd_status dd_partitioner_row_offsets(const dd_partitioner *p, int64_t *host_out);
+/* staged-var path only: device pointer to the rebuilt 64-bit Arrow offsets [n_rows+1]
+ * of a var column (K4c output); NULL on the v1 path */
+const uint64_t *dd_partitioner_var_offsets64(const dd_partitioner *p, int32_t col);
+/* out32[i] = (int32)(off64[lo_row + i] - off64[lo_row]) for i in [0, n]; device buffers;
+ * lets a bucket/chunk of a var column be re-consumed as an Arrow batch view */
+dd_status dd_make_offsets32(const uint64_t *off64, int64_t lo_row, int64_t n,
+                            int32_t *out32, void *stream);
 /* copies per-var-col part_byte_offsets[P+1] to host */
 dd_status dd_partitioner_byte_offsets(const dd_partitioner *p, int32_t col, int64_t *host_out);
 /* per-kernel last-run durations in ms, measured with hipEvents on the run stream:
