@@ -169,6 +169,34 @@ class DeviceBatch:
                 cd.dict_offsets = self._up(np.ascontiguousarray(col["dict_offsets"], np.int32))
                 cd.dict_n = len(col["dict_offsets"]) - 1
 
+    @classmethod
+    def from_device(cls, view_cols, n_rows):
+        """Non-owning device-view batch: each col gives raw device pointers
+        ({"dtype", "data_ptr", "valid_ptr", "offsets_ptr", "data_len", dict...}).
+        Used by the two-level scatter to re-consume pass-A bucket slices without
+        copies."""
+        self = cls.__new__(cls)
+        self.cols = view_cols
+        self.n_rows = n_rows
+        self._bufs = []
+        self.desc = BatchDesc()
+        self.desc.n_rows = n_rows
+        self.desc.n_cols = len(view_cols)
+        for i, col in enumerate(view_cols):
+            cd = self.desc.cols[i]
+            cd.dtype = DTYPE_CODE[col["dtype"]]
+            cd.data = col["data_ptr"]
+            cd.data_len = int(col.get("data_len", 0))
+            if col.get("valid_ptr"):
+                cd.validity = col["valid_ptr"]
+            if col.get("offsets_ptr"):
+                cd.offsets = col["offsets_ptr"]
+            if col.get("dict_bytes_ptr"):
+                cd.dict_bytes = col["dict_bytes_ptr"]
+                cd.dict_offsets = col["dict_offsets_ptr"]
+                cd.dict_n = col["dict_n"]
+        return self
+
     def _up(self, arr):
         p = _dev_alloc(arr.nbytes)
         _h2d(p, arr)
@@ -184,15 +212,24 @@ class DeviceBatch:
 class Partitioner:
     """Wraps dd_partitioner_*: the producer-head replacement."""
 
-    def __init__(self, batch: DeviceBatch, key_idx, nparts):
+    def __init__(self, batch: DeviceBatch, key_idx, nparts, ranged=None):
+        """ranged=(pid_total, coarse_div): coarse-bucket mode — nparts is then
+        pid_total // coarse_div contiguous ranges (dd_partitioner_create_ranged)."""
         self.batch = batch
-        self.nparts = nparts
         self.key_idx = list(key_idx)
         keys = (ctypes.c_int32 * len(key_idx))(*key_idx)
         self.h = ctypes.c_void_p()
-        _check(lib().dd_partitioner_create(ctypes.byref(batch.desc), keys,
-                                           len(key_idx), ctypes.c_uint32(nparts),
-                                           ctypes.byref(self.h)))
+        if ranged is None:
+            self.nparts = nparts
+            _check(lib().dd_partitioner_create(ctypes.byref(batch.desc), keys,
+                                               len(key_idx), ctypes.c_uint32(nparts),
+                                               ctypes.byref(self.h)))
+        else:
+            total, div = ranged
+            self.nparts = total // div
+            _check(lib().dd_partitioner_create_ranged(
+                ctypes.byref(batch.desc), keys, len(key_idx), ctypes.c_uint32(total),
+                ctypes.c_uint32(div), ctypes.byref(self.h)))
 
     def run(self, stream=None):
         _check(lib().dd_partitioner_run(self.h, stream))
