@@ -85,6 +85,7 @@ def lib():
         L.dd_partitioner_col_data.restype = ctypes.c_void_p
         L.dd_partitioner_col_validity.restype = ctypes.c_void_p
         L.dd_partitioner_col_lengths.restype = ctypes.c_void_p
+        L.dd_partitioner_var_offsets64.restype = ctypes.c_void_p
         L.dd_exchanged_col_data.restype = ctypes.c_void_p
         L.dd_exchanged_col_validity.restype = ctypes.c_void_p
         L.dd_exchanged_col_lengths.restype = ctypes.c_void_p
