@@ -77,6 +77,8 @@ def two_level_partition(cols, key_idx, nparts, buckets=8):
         passB.append(pb)
     tB = time.perf_counter() - t1
 
+    kernel_ms = sum(passA.kernel_ms()) + sum(sum(pb.kernel_ms()) for pb in passB)
+
     # global offsets: bucket-contiguous
     grow = np.zeros(nparts + 1, dtype=np.int64)
     gbyte = {i: np.zeros(nparts + 1, dtype=np.int64) for i in var_idx}
@@ -147,4 +149,5 @@ def two_level_partition(cols, key_idx, nparts, buckets=8):
         for b in final[i].values():
             L.dd_dev_free(b)
     return {"part_row_offsets": grow, "cols": out_cols,
-            "timings_ms": {"passA": tA * 1e3, "passB": tB * 1e3, "concat": tC * 1e3}}
+            "timings_ms": {"passA": tA * 1e3, "passB": tB * 1e3, "concat": tC * 1e3,
+                           "kernels": kernel_ms}}
