@@ -31,17 +31,29 @@ def two_level_partition(cols, key_idx, nparts, buckets=8):
     var_idx = [i for i, c in enumerate(cols) if c["dtype"] == "utf8"]
 
     batch = api.DeviceBatch(cols)
+    import os
     import time
     t0 = time.perf_counter()
-    passA = api.Partitioner(batch, key_idx, None, ranged=(nparts, S))
+    # pass A on the v1 direct path: sequential reads, writes into only `buckets` streams
+    # (large runs) — avoids running the expensive K4 gather twice. Pass B (staged-var)
+    # then gathers within an L3-resident bucket.
+    prev = os.environ.get("DD_V2_VAR")
+    os.environ["DD_V2_VAR"] = "0"
+    try:
+        passA = api.Partitioner(batch, key_idx, None, ranged=(nparts, S))
+    finally:
+        if prev is None:
+            os.environ.pop("DD_V2_VAR", None)
+        else:
+            os.environ["DD_V2_VAR"] = prev
     passA.run()
     passA.sync()
     tA = time.perf_counter() - t0
     roffA = passA.row_offsets()
     boffA = {i: passA.byte_offsets(i) for i in var_idx}
-    off64 = {i: L.dd_partitioner_var_offsets64(passA.h, i) for i in var_idx}
-    for i in var_idx:
-        assert off64[i], "two-level requires the staged-var pass-A path"
+    # v1 pass A has no rebuilt offsets64: build bucket offsets from the lengths instead
+    lensA = {i: api._d2h(L.dd_partitioner_col_lengths(passA.h, i), n * 4, np.uint32)
+             for i in var_idx}
 
     # reusable i32 offsets scratch for bucket views
     max_rows = int((roffA[1:] - roffA[:-1]).max()) if n else 0
@@ -58,9 +70,12 @@ def two_level_partition(cols, key_idx, nparts, buckets=8):
         for i, c in enumerate(cols):
             v = {"dtype": c["dtype"]}
             if c["dtype"] == "utf8":
-                api._check(L.dd_make_offsets32(
-                    ctypes.c_void_p(off64[i]), ctypes.c_int64(lo), ctypes.c_int64(nrows),
-                    off32[i], None))
+                # host-built i32 offsets for the bucket (lengths prefix), uploaded
+                ob = np.zeros(nrows + 1, dtype=np.int32)
+                np.cumsum(lensA[i][lo:hi], out=ob[1:].view(np.uint32)[:nrows])
+                api._check(L.dd_memcpy_h2d(off32[i],
+                                           ob.ctypes.data_as(ctypes.c_void_p),
+                                           ctypes.c_int64(ob.nbytes)))
                 v["data_ptr"] = L.dd_partitioner_col_data(passA.h, i) + int(boffA[i][g])
                 v["offsets_ptr"] = off32[i].value
                 v["data_len"] = int(boffA[i][g + 1] - boffA[i][g])
