@@ -762,57 +762,34 @@ __global__ __launch_bounds__(BLOCK_THREADS) void k4_copy(
         if (lane == 0) loc[WAVE] = T;
         /* single wave: LDS program order; no barrier */
         const uint64_t obase = out_off[s0];
-        /* batch K4_ILP chunk loads per lane before storing: the random string reads are
-         * latency-bound, so in-flight depth matters more than instruction count. All
-         * slot arrays are indexed by the CONSTANT unrolled k (a runtime index would put
-         * them in scratch — the 400 B/lane trap). */
-        constexpr int K4_ILP = 8;
-        const uint8_t *srcp[K4_ILP];
-        uint32_t dsto[K4_ILP];
-        uint64_t tv[K4_ILP];
-        bool full[K4_ILP];
-        for (uint32_t p0 = lane * 8; p0 < T; p0 += (uint32_t)(WAVE * 8 * K4_ILP)) {
-#pragma unroll
-            for (int k = 0; k < K4_ILP; k++) {
-                full[k] = false;
-                const uint32_t p = p0 + (uint32_t)(WAVE * 8) * k;
-                if (p >= T) continue;
-                uint32_t rem = (T - p < 8) ? (T - p) : 8;
-                int lo = 0, hi = WAVE;
-                while (lo < hi) {
-                    int mid = (lo + hi + 1) >> 1;
-                    if (loc[mid] <= p) lo = mid;
-                    else hi = mid - 1;
-                }
-                int j = lo;
-                while (loc[j + 1] <= p) j++;
-                const uint32_t avail = loc[j + 1] - p;
-                if (rem == 8 && avail >= 8) {
-                    srcp[k] = in_bytes + srcb[j] + (p - loc[j]);
-                    dsto[k] = p;
-                    full[k] = true;
-                } else {
-                    /* boundary chunk: sub-copies inline (string edges / tail) */
-                    uint32_t pp = p;
-                    while (rem > 0) {
-                        while (loc[j + 1] <= pp) j++;
-                        const uint32_t within = pp - loc[j];
-                        const uint32_t av = loc[j + 1] - pp;
-                        const uint32_t m = (rem < av) ? rem : av;
-                        const uint8_t *sp = in_bytes + srcb[j] + within;
-                        uint8_t *dp = out_bytes + obase + pp;
-                        for (uint32_t b = 0; b < m; b++) dp[b] = sp[b];
-                        pp += m;
-                        rem -= m;
-                    }
-                }
+        for (uint32_t p0 = lane * 8; p0 < T; p0 += WAVE * 8) {
+            uint32_t rem = (T - p0 < 8) ? (T - p0) : 8;
+            uint32_t p = p0;
+            /* binary search: largest j with loc[j] <= p */
+            int lo = 0, hi = WAVE;
+            while (lo < hi) {
+                int mid = (lo + hi + 1) >> 1;
+                if (loc[mid] <= p) lo = mid;
+                else hi = mid - 1;
             }
-#pragma unroll
-            for (int k = 0; k < K4_ILP; k++)
-                if (full[k]) __builtin_memcpy(&tv[k], srcp[k], 8);
-#pragma unroll
-            for (int k = 0; k < K4_ILP; k++)
-                if (full[k]) __builtin_memcpy(out_bytes + obase + dsto[k], &tv[k], 8);
+            int j = lo;
+            while (rem > 0) {
+                while (loc[j + 1] <= p) j++; /* skip empty strings */
+                const uint32_t within = p - loc[j];
+                const uint32_t avail = loc[j + 1] - p;
+                const uint32_t m = (rem < avail) ? rem : avail;
+                const uint8_t *sp = in_bytes + srcb[j] + within;
+                uint8_t *dp = out_bytes + obase + p;
+                if (m == 8) {
+                    uint64_t t;
+                    __builtin_memcpy(&t, sp, 8);
+                    __builtin_memcpy(dp, &t, 8);
+                } else {
+                    for (uint32_t b = 0; b < m; b++) dp[b] = sp[b];
+                }
+                p += m;
+                rem -= m;
+            }
         }
     }
 }
