@@ -1038,9 +1038,9 @@ extern "C" dd_status dd_partial_reduce_run(const dd_batch_desc *batch,
     for (int k = 0; k < n_keys; k++) {
         if (key_cols[k] < 0 || key_cols[k] >= batch->n_cols)
             return set_err(DD_ERR_INVALID, "key column out of range");
-        if (batch->cols[key_cols[k]].dtype == DD_DT_UTF8 ||
-            batch->cols[key_cols[k]].dtype == DD_DT_DICT32)
-            return set_err(DD_ERR_UNSUPPORTED, "partial reduce: fixed-width keys only");
+        if (batch->cols[key_cols[k]].dtype == DD_DT_UTF8)
+            return set_err(DD_ERR_UNSUPPORTED,
+                           "partial reduce: fixed-width or dictionary keys only");
     }
     for (int g = 0; g < n_aggs; g++) {
         if (agg_ops[g] == DD_AGG_COUNT) continue;

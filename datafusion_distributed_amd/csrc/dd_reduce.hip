@@ -95,6 +95,11 @@ __device__ __forceinline__ uint64_t dd_key_bits(const dd_kcol &c, int64_t i) {
         if (v != v) b = 0x7ff8000000000000ULL;
         return b;
     }
+    case DD_KDT_DICT32:
+        /* group by dictionary INDEX: within one batch the dict is shared, so equal
+         * indices == equal values; duplicate values under different indices would only
+         * split a group, which partial aggregation permits (merged downstream by value) */
+        return (uint64_t)(uint32_t)((const int32_t *)c.data)[i];
     default:
         return 0;
     }

@@ -177,3 +177,34 @@ def test_partial_reduce_min_max():
         assert got[1] == direct["f_max"][r].as_py()
         assert got[2] == direct["i_min"][r].as_py()
         assert got[3] == direct["i_max"][r].as_py()
+
+
+def test_partial_reduce_dict_key():
+    import pyarrow as pa
+
+    rng = np.random.default_rng(41)
+    n = 200_000
+    nvals = 10
+    vals = [f"cat{v}".encode() for v in range(nvals)]
+    dby = np.frombuffer(b"".join(vals), dtype=np.uint8).copy()
+    doff = np.zeros(nvals + 1, dtype=np.int32)
+    doff[1:] = np.cumsum([len(v) for v in vals])
+    idx = rng.integers(0, nvals, n).astype(np.int32)
+    x = rng.normal(size=n)
+    batch = api.DeviceBatch([
+        {"dtype": "dict32", "data": idx, "dict_bytes": dby, "dict_offsets": doff,
+         "valid": None},
+        {"dtype": "f64", "data": x, "valid": None},
+    ])
+    res = api.partial_reduce(batch, [0], [(1, "sum_f64"), (None, "count")])
+    batch.free()
+    merged = merge_partials_mixed(res, int_ops={1})
+    tbl = pa.table({"k": idx, "v": x})
+    direct = tbl.group_by("k").aggregate([("v", "sum"), ([], "count_all")])
+    assert len(merged) == direct.num_rows
+    for r in range(direct.num_rows):
+        key = (direct["k"][r].as_py(), 0)
+        got = merged[key]
+        assert got[1] == direct["count_all"][r].as_py()
+        want = direct["v_sum"][r].as_py()
+        assert abs(got[0] - want) <= 1e-6 * max(abs(want), 1.0)
