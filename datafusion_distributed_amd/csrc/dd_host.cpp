@@ -1010,6 +1010,7 @@ extern "C" dd_status dd_exchanged_stats(const dd_exchanged *e, float *ms,
 
 struct dd_reducer {
     int32_t n_keys = 0, n_aggs = 0;
+    uint64_t *dict_hashes[DD_KMAX_KEYS] = {}; /* per dict32 KEY col (hash input) */
     int64_t n_rows = 0; /* output rows */
     uint64_t *keys = nullptr;
     uint32_t *keynull = nullptr;
@@ -1018,6 +1019,7 @@ struct dd_reducer {
     int32_t *meta = nullptr; /* device: agg_cols + agg_ops */
     float kernel_ms = 0;     /* hipEvent time of the reduce kernel alone */
     ~dd_reducer() {
+        for (auto &d : dict_hashes) (void)hipFree(d);
         (void)hipFree(keys);
         (void)hipFree(keynull);
         (void)hipFree(aggs);
@@ -1089,6 +1091,19 @@ extern "C" dd_status dd_partial_reduce_run(const dd_batch_desc *batch,
         delete r;
         return set_err(DD_ERR_HIP, m);
     };
+
+    /* dict32 key columns: the row hash reads precomputed per-value hashes */
+    for (int k = 0; k < n_keys; k++) {
+        const dd_col_desc &cd = batch->cols[key_cols[k]];
+        if (cd.dtype != DD_DT_DICT32) continue;
+        if (hipMalloc((void **)&r->dict_hashes[k], (size_t)cd.dict_n * 8 + 8) != hipSuccess)
+            return fail("dict hash alloc");
+        if (dd_launch_dict_hashes((const uint8_t *)cd.dict_bytes, cd.dict_offsets,
+                                  cd.dict_n, r->dict_hashes[k],
+                                  (hipStream_t)stream) != hipSuccess)
+            return fail("dict hash launch");
+        ka.cols[key_cols[k]].dict_hashes = r->dict_hashes[k];
+    }
     if (hipMalloc((void **)&r->keys, (size_t)max_rows * n_keys * 8) != hipSuccess ||
         hipMalloc((void **)&r->keynull, (size_t)max_rows * 4) != hipSuccess ||
         hipMalloc((void **)&r->aggs, (size_t)max_rows * n_aggs * 8) != hipSuccess ||
