@@ -14,6 +14,7 @@
 #include <cstdio>
 #include <cstring>
 #include <map>
+#include <memory>
 #include <mutex>
 #include <string>
 #include <vector>
@@ -580,8 +581,13 @@ extern "C" dd_status dd_execute_task(const dd_task_key *key, uint32_t part_lo, u
     if (part_lo > part_hi || part_hi > p->nparts)
         return set_err(DD_ERR_INVALID, "partition range out of bounds");
     if (!p->has_run) {
-        dd_status st = dd_partitioner_run(p, stream);
-        if (st != DD_OK) return st;
+        /* run under the cache lock: launches are async enqueues (cheap), and this keeps
+         * two concurrent execute_task calls on the same key from double-launching */
+        std::lock_guard<std::mutex> g(g_tasks_mu);
+        if (!p->has_run) {
+            dd_status st = dd_partitioner_run(p, stream);
+            if (st != DD_OK) return st;
+        }
     }
     *out = p;
     return DD_OK;
@@ -607,6 +613,7 @@ struct dd_comm {
 };
 
 extern "C" dd_status dd_comm_unique_id(void *bytes128) {
+    if (!bytes128) return set_err(DD_ERR_INVALID, "null argument");
     ncclUniqueId id;
     NCCL_TRY(ncclGetUniqueId(&id));
     memcpy(bytes128, &id, sizeof(id));
@@ -630,6 +637,32 @@ extern "C" dd_status dd_comm_init(const void *bytes128, int rank, int nranks, dd
 }
 
 extern "C" void dd_comm_destroy(dd_comm *c) { delete c; }
+
+/* allgather of the per-rank size matrix (exchange + coalesce); device staging buffers are
+ * freed on every path, including RCCL/HIP failure */
+static dd_status dd_meta_allgather(dd_comm *c, const std::vector<int64_t> &mine,
+                                   std::vector<int64_t> &all, hipStream_t s) {
+    const size_t meta_n = mine.size();
+    int64_t *d_in = nullptr, *d_all = nullptr;
+    HIP_TRY(hipMalloc(&d_in, meta_n * 8));
+    if (hipMalloc(&d_all, all.size() * 8) != hipSuccess) {
+        (void)hipFree(d_in);
+        return set_err(DD_ERR_HIP, "meta allgather alloc");
+    }
+    hipError_t he = hipMemcpyAsync(d_in, mine.data(), meta_n * 8, hipMemcpyHostToDevice, s);
+    ncclResult_t nr = ncclSuccess;
+    if (he == hipSuccess) nr = ncclAllGather(d_in, d_all, meta_n, ncclInt64, c->comm, s);
+    if (he == hipSuccess && nr == ncclSuccess) he = hipStreamSynchronize(s);
+    if (he == hipSuccess && nr == ncclSuccess)
+        he = hipMemcpy(all.data(), d_all, all.size() * 8, hipMemcpyDeviceToHost);
+    (void)hipFree(d_in);
+    (void)hipFree(d_all);
+    if (nr != ncclSuccess)
+        return set_err(DD_ERR_RCCL, std::string("size allgather: ") + ncclGetErrorString(nr));
+    if (he != hipSuccess)
+        return set_err(DD_ERR_HIP, std::string("size allgather: ") + hipGetErrorString(he));
+    return DD_OK;
+}
 
 /* ---------------- broadcast (dd_bcast) ---------------- */
 
@@ -697,21 +730,31 @@ extern "C" dd_status dd_broadcast_run(dd_comm *c, const dd_batch_desc *batch, in
     }
     int64_t *d_hdr = nullptr;
     HIP_TRY(hipMalloc(&d_hdr, sizeof(hdr)));
-    if (is_root)
-        HIP_TRY(hipMemcpyAsync(d_hdr, hdr, sizeof(hdr), hipMemcpyHostToDevice, s));
-    NCCL_TRY(ncclBroadcast(d_hdr, d_hdr, DD_BHDR, ncclInt64, root, c->comm, s));
-    HIP_TRY(hipStreamSynchronize(s));
-    HIP_TRY(hipMemcpy(hdr, d_hdr, sizeof(hdr), hipMemcpyDeviceToHost));
-    (void)hipFree(d_hdr);
+    {
+        /* merged checks so d_hdr is freed on every path */
+        hipError_t he = hipSuccess;
+        ncclResult_t nr = ncclSuccess;
+        if (is_root) he = hipMemcpyAsync(d_hdr, hdr, sizeof(hdr), hipMemcpyHostToDevice, s);
+        if (he == hipSuccess)
+            nr = ncclBroadcast(d_hdr, d_hdr, DD_BHDR, ncclInt64, root, c->comm, s);
+        if (he == hipSuccess && nr == ncclSuccess) he = hipStreamSynchronize(s);
+        if (he == hipSuccess && nr == ncclSuccess)
+            he = hipMemcpy(hdr, d_hdr, sizeof(hdr), hipMemcpyDeviceToHost);
+        (void)hipFree(d_hdr);
+        if (nr != ncclSuccess)
+            return set_err(DD_ERR_RCCL,
+                           std::string("header broadcast: ") + ncclGetErrorString(nr));
+        if (he != hipSuccess)
+            return set_err(DD_ERR_HIP,
+                           std::string("header broadcast: ") + hipGetErrorString(he));
+    }
 
-    auto b = new dd_bcast();
+    std::unique_ptr<dd_bcast> b_own(new dd_bcast()); /* freed on every error return */
+    dd_bcast *b = b_own.get();
     b->n_rows = hdr[0];
     b->n_cols = (int32_t)hdr[1];
     b->owns = !is_root;
-    auto fail = [&](const char *m) {
-        delete b;
-        return set_err(DD_ERR_HIP, m);
-    };
+    auto fail = [&](const char *m) { return set_err(DD_ERR_HIP, m); };
     for (int i = 0; i < b->n_cols; i++) {
         const int64_t *h = hdr + 2 + (size_t)i * 6;
         b->dtypes[i] = (int32_t)h[0];
@@ -759,7 +802,7 @@ extern "C" dd_status dd_broadcast_run(dd_comm *c, const dd_batch_desc *batch, in
     }
     NCCL_TRY(ncclGroupEnd());
     HIP_TRY(hipStreamSynchronize(s));
-    *out = b;
+    *out = b_own.release();
     return DD_OK;
 }
 
@@ -838,19 +881,14 @@ extern "C" dd_status dd_exchange_run(dd_comm *c, const dd_partitioner *p, void *
         for (uint32_t q = 0; q < p->nparts; q++)
             my_meta[(size_t)(1 + v) * p->nparts + q] = boff_h[v][q + 1] - boff_h[v][q];
     }
-    int64_t *d_meta_in = nullptr, *d_meta_all = nullptr;
-    HIP_TRY(hipMalloc(&d_meta_in, meta_n * 8));
-    HIP_TRY(hipMalloc(&d_meta_all, (size_t)R * meta_n * 8));
-    HIP_TRY(hipMemcpyAsync(d_meta_in, my_meta.data(), meta_n * 8, hipMemcpyHostToDevice, s));
-    NCCL_TRY(ncclAllGather(d_meta_in, d_meta_all, meta_n, ncclInt64, c->comm, s));
     std::vector<int64_t> all_meta((size_t)R * meta_n);
-    HIP_TRY(hipStreamSynchronize(s));
-    HIP_TRY(hipMemcpy(all_meta.data(), d_meta_all, (size_t)R * meta_n * 8,
-                      hipMemcpyDeviceToHost));
-    (void)hipFree(d_meta_in);
-    (void)hipFree(d_meta_all);
+    {
+        dd_status ms = dd_meta_allgather(c, my_meta, all_meta, s);
+        if (ms != DD_OK) return ms;
+    }
 
-    auto e = new dd_exchanged();
+    std::unique_ptr<dd_exchanged> e_own(new dd_exchanged()); /* freed on error returns */
+    dd_exchanged *e = e_own.get();
     e->alloc_stream = dd_pool_stream();
     e->n_cols = p->batch.n_cols;
     e->nranks = R;
@@ -869,10 +907,7 @@ extern "C" dd_status dd_exchange_run(dd_comm *c, const dd_partitioner *p, void *
         for (uint32_t q = 0; q < P; q++)
             recv_rows_per_producer[r] += e->row_counts[(size_t)r * P + q];
 
-    auto fail = [&](dd_status sc, const char *m) {
-        delete e;
-        return set_err(sc, m);
-    };
+    auto fail = [&](dd_status sc, const char *m) { return set_err(sc, m); };
 
     /* allocate receive buffers */
     std::vector<std::vector<int64_t>> recv_bytes_per_producer(p->batch.n_cols,
@@ -973,7 +1008,7 @@ extern "C" dd_status dd_exchange_run(dd_comm *c, const dd_partitioner *p, void *
     HIP_TRY(hipEventRecord(e->e1, s));
     HIP_TRY(hipStreamSynchronize(s));
     HIP_TRY(hipEventElapsedTime(&e->ms, e->e0, e->e1));
-    *out = e;
+    *out = e_own.release();
     return DD_OK;
 }
 
@@ -1087,7 +1122,10 @@ extern "C" dd_status dd_partial_reduce_run(const dd_batch_desc *batch,
     auto r = new dd_reducer();
     r->n_keys = n_keys;
     r->n_aggs = n_aggs;
+    hipEvent_t e0 = nullptr, e1 = nullptr;
     auto fail = [&](const char *m) {
+        if (e0) (void)hipEventDestroy(e0);
+        if (e1) (void)hipEventDestroy(e1);
         delete r;
         return set_err(DD_ERR_HIP, m);
     };
@@ -1120,7 +1158,6 @@ extern "C" dd_status dd_partial_reduce_run(const dd_batch_desc *batch,
     if (hipMemcpyAsync(r->meta, meta_h, sizeof(meta_h), hipMemcpyHostToDevice, s) !=
         hipSuccess)
         return fail("meta upload");
-    hipEvent_t e0 = nullptr, e1 = nullptr;
     (void)hipEventCreate(&e0);
     (void)hipEventCreate(&e1);
     (void)hipEventRecord(e0, s);
@@ -1133,6 +1170,7 @@ extern "C" dd_status dd_partial_reduce_run(const dd_batch_desc *batch,
     (void)hipEventElapsedTime(&r->kernel_ms, e0, e1);
     (void)hipEventDestroy(e0);
     (void)hipEventDestroy(e1);
+    e0 = e1 = nullptr; /* fail() must not double-destroy */
     uint64_t n_out = 0;
     if (hipMemcpy(&n_out, r->n_dev, 8, hipMemcpyDeviceToHost) != hipSuccess)
         return fail("n_out copy");
@@ -1194,17 +1232,11 @@ extern "C" dd_status dd_coalesce_run(dd_comm *c, const dd_partitioner *p,
         for (uint32_t q = 0; q < P; q++)
             my_meta[(size_t)(1 + v) * P + q] = boff_h[v][q + 1] - boff_h[v][q];
     }
-    int64_t *d_in = nullptr, *d_all = nullptr;
-    HIP_TRY(hipMalloc(&d_in, meta_n * 8));
-    HIP_TRY(hipMalloc(&d_all, (size_t)R * meta_n * 8));
-    HIP_TRY(hipMemcpyAsync(d_in, my_meta.data(), meta_n * 8, hipMemcpyHostToDevice, s));
-    NCCL_TRY(ncclAllGather(d_in, d_all, meta_n, ncclInt64, c->comm, s));
     std::vector<int64_t> all_meta((size_t)R * meta_n);
-    HIP_TRY(hipStreamSynchronize(s));
-    HIP_TRY(hipMemcpy(all_meta.data(), d_all, (size_t)R * meta_n * 8,
-                      hipMemcpyDeviceToHost));
-    (void)hipFree(d_in);
-    (void)hipFree(d_all);
+    {
+        dd_status ms = dd_meta_allgather(c, my_meta, all_meta, s);
+        if (ms != DD_OK) return ms;
+    }
 
     /* 2) my group as a consumer (ranks >= consumer_tasks consume nothing) */
     int gstart = 0, glen = 0;
@@ -1217,7 +1249,8 @@ extern "C" dd_status dd_coalesce_run(dd_comm *c, const dd_partitioner *p,
         if (c->rank >= s0 && c->rank < s0 + l0) my_consumer = t;
     }
 
-    auto e = new dd_exchanged();
+    std::unique_ptr<dd_exchanged> e_own(new dd_exchanged()); /* freed on error returns */
+    dd_exchanged *e = e_own.get();
     e->alloc_stream = dd_pool_stream();
     e->n_cols = p->batch.n_cols;
     e->nranks = glen; /* producers I consume */
@@ -1234,10 +1267,7 @@ extern "C" dd_status dd_coalesce_run(dd_comm *c, const dd_partitioner *p,
             e->total_rows += n;
         }
     }
-    auto fail = [&](dd_status sc, const char *m) {
-        delete e;
-        return set_err(sc, m);
-    };
+    auto fail = [&](dd_status sc, const char *m) { return set_err(sc, m); };
 
     std::vector<std::vector<int64_t>> recv_bytes(p->batch.n_cols,
                                                  std::vector<int64_t>(glen, 0));
@@ -1289,10 +1319,12 @@ extern "C" dd_status dd_coalesce_run(dd_comm *c, const dd_partitioner *p,
         const dd_kcol &kc = p->ka.cols[ci];
         /* sends (every rank has exactly one consumer) */
         if (my_consumer >= 0) {
+            int64_t sent = 0;
             if (kc.elem > 0) {
                 if (my_rows > 0)
                     NCCL_TRY(ncclSend(p->out_data[ci], my_rows * kc.elem, ncclUint8,
                                       my_consumer, c->comm, s));
+                sent = my_rows * kc.elem;
             } else {
                 int v = -1;
                 for (int vv = 0; vv < nvar; vv++)
@@ -1304,12 +1336,14 @@ extern "C" dd_status dd_coalesce_run(dd_comm *c, const dd_partitioner *p,
                 if (my_rows > 0)
                     NCCL_TRY(ncclSend(p->out_lengths[ci], my_rows * 4, ncclUint8,
                                       my_consumer, c->comm, s));
+                sent = my_b + my_rows * 4;
             }
-            if (kc.valid && my_rows > 0)
+            if (kc.valid && my_rows > 0) {
                 NCCL_TRY(ncclSend(p->out_valid[ci], my_rows, ncclUint8, my_consumer,
                                   c->comm, s));
-            if (my_consumer != c->rank)
-                e->egress += (kc.elem > 0 ? my_rows * kc.elem : 0);
+                sent += my_rows;
+            }
+            if (my_consumer != c->rank) e->egress += sent;
         }
         /* recvs */
         int64_t rrow0 = 0, rb0 = 0;
@@ -1338,7 +1372,7 @@ extern "C" dd_status dd_coalesce_run(dd_comm *c, const dd_partitioner *p,
     HIP_TRY(hipEventRecord(e->e1, s));
     HIP_TRY(hipStreamSynchronize(s));
     HIP_TRY(hipEventElapsedTime(&e->ms, e->e0, e->e1));
-    *out = e;
+    *out = e_own.release();
     return DD_OK;
 }
 
