@@ -233,7 +233,7 @@ class Partitioner:
                 ctypes.c_uint32(div), ctypes.byref(self.h)))
 
     def run(self, stream=None):
-        _check(lib().dd_partitioner_run(self.h, stream))
+        _check(lib().dd_partitioner_run(self.h, _stream_arg(stream)))
 
     def run_phase1(self, stream=None):
         """K1 hash+count + K2 scans (batch-pipelining split; see dd_shuffle.h)."""
@@ -316,15 +316,16 @@ class Comm:
 
     def exchange(self, part: Partitioner, stream=None):
         h = ctypes.c_void_p()
-        _check(lib().dd_exchange_run(self.h, part.h, stream, ctypes.byref(h)))
+        _check(lib().dd_exchange_run(self.h, part.h, _stream_arg(stream),
+                                     ctypes.byref(h)))
         return Exchanged(h, part, self)
 
     def coalesce(self, part: Partitioner, consumer_tasks: int, stream=None):
         """NetworkCoalesceExec data plane (dd_coalesce_run): this consumer rank receives
         the whole partitioned output of every producer rank in its contiguous group."""
         h = ctypes.c_void_p()
-        _check(lib().dd_coalesce_run(self.h, part.h, consumer_tasks, stream,
-                                     ctypes.byref(h)))
+        _check(lib().dd_coalesce_run(self.h, part.h, consumer_tasks,
+                                     _stream_arg(stream), ctypes.byref(h)))
         ex = Exchanged(h, part, self)
         return ex
 
@@ -333,7 +334,8 @@ class Comm:
         BroadcastExec/NetworkBroadcastExec data plane)."""
         h = ctypes.c_void_p()
         desc = ctypes.byref(batch.desc) if batch is not None else None
-        _check(lib().dd_broadcast_run(self.h, desc, root, stream, ctypes.byref(h)))
+        _check(lib().dd_broadcast_run(self.h, desc, root, _stream_arg(stream),
+                                      ctypes.byref(h)))
         return Broadcasted(h)
 
     def destroy(self):
