@@ -253,7 +253,11 @@ class Partitioner:
         _check(lib().dd_device_sync())
 
     def pids(self):
+        """Per-row partition ids, or None when the spec path recomputed hashes in-kernel
+        (no pid array exists; outputs + row_offsets fully define the partitioning)."""
         p = lib().dd_partitioner_pids(self.h)
+        if not p:
+            return None
         return _d2h(p, self.batch.n_rows * 4, np.uint32)
 
     def row_offsets(self):

@@ -312,10 +312,30 @@ extern "C" dd_status dd_partitioner_create(const dd_batch_desc *batch, const int
     }
     const int64_t nchunks = p->nchunks;
 
+    /* rhash: spec-path batches (wpb 16, all fixed, no validity) with integer/bool keys
+     * recompute the row hash in K3 from the already-preloaded register values — the pid
+     * array (4 B/row HBM write in K1 + read in K3) is skipped entirely. Must mirror the
+     * launcher's can_spec/can_spec8 conditions exactly (dd_launch_scatter_staged fails
+     * loudly if not). DD_RHASH=0 disables for A/B. */
+    if (p->staged && p->wpb == 16 && nvar == 0 && batch->n_cols <= DD_STAGE_MAXC) {
+        bool rhash = true;
+        for (int c = 0; c < batch->n_cols && rhash; c++)
+            if (batch->cols[c].validity || fixed_elem_size(batch->cols[c].dtype) == 0)
+                rhash = false;
+        for (int k = 0; k < n_keys && rhash; k++) {
+            const int dt = batch->cols[key_cols[k]].dtype;
+            if (!(dt == DD_DT_U8 || dt == DD_DT_BOOL || dt == DD_DT_I16 ||
+                  dt == DD_DT_I32 || dt == DD_DT_I64))
+                rhash = false; /* float needs canon, dict32 a table gather: pid path */
+        }
+        if (getenv("DD_RHASH") && atoi(getenv("DD_RHASH")) == 0) rhash = false;
+        ka.rhash = rhash ? 1 : 0;
+    }
+
     auto halloc = [&](void **ptr, size_t bytes) {
         return hipMalloc(ptr, bytes > 0 ? bytes : 1) == hipSuccess;
     };
-    bool ok = halloc((void **)&p->pid, (size_t)n * 4) &&
+    bool ok = (ka.rhash ? true : halloc((void **)&p->pid, (size_t)n * 4)) &&
               halloc((void **)&p->counts, (size_t)nchunks * P * 4) &&
               halloc((void **)&p->partials, (size_t)DD_SCAN_RANGES * P * 4) &&
               halloc((void **)&p->part_offsets, (size_t)(P + 1) * 8);

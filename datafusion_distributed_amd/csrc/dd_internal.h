@@ -48,6 +48,10 @@ struct dd_kargs {
     int32_t n_var;
     uint32_t pid_total; /* pid = (h % pid_total) >> pid_shift; nparts = pid_total >> shift */
     int32_t pid_shift;  /* 0 for plain partitioning; >0 = contiguous coarse buckets */
+    int32_t rhash;      /* staged spec path recomputes the row hash from preloaded
+                           registers in K3 (no pid array: K1 skips its store, K3 its
+                           load). Host gates: all-fixed no-validity batch, integer/bool
+                           keys, wpb==16 (DD_RHASH=0 disables). */
     int32_t key_idx[DD_KMAX_KEYS];
     int32_t var_idx[DD_KMAX_VAR];
     dd_kcol cols[DD_KMAX_COLS];

@@ -334,7 +334,7 @@ __global__ __launch_bounds__(BLOCK_THREADS) void k_hash_count_tile(
                                     ? (uint32_t)(h & (uint64_t)(tot - 1))
                                     : (uint32_t)(h % (uint64_t)tot);
                 pidu[u] = fine >> a.pid_shift;
-                pid_out[row] = pidu[u];
+                if (pid_out) pid_out[row] = pidu[u]; /* null when K3 recomputes (rhash) */
             }
         }
 #pragma unroll
@@ -394,8 +394,11 @@ __device__ __forceinline__ void dd_block_excl_scan(const uint32_t *vals, uint32_
 
 /* NC > 0: compile-time column count, no validity, no var (specialized fast path — the
  * generic runtime column loop with break-guards and validity checks measured ~8-15%
- * slower at the bench shape, tools/ablate_k3.cpp). NC == 0: generic. */
-template <int GMAX, int WPB, int MAXC, bool HASVAR, int NC = 0>
+ * slower at the bench shape, tools/ablate_k3.cpp). NC == 0: generic.
+ * RHASH: recompute the row hash from the preloaded register values instead of reading
+ * pid_in — the key columns are already in colv, so the pid array's HBM round trip
+ * (write in K1 + read here) disappears. Spec paths only (host gate: ka.rhash). */
+template <int GMAX, int WPB, int MAXC, bool HASVAR, int NC = 0, bool RHASH = false>
 __global__ __launch_bounds__(WPB * WAVE) void k_scatter_staged(
     dd_kargs a, int64_t tile_rows, uint32_t nparts, int nbits, const uint32_t *pid_in,
     const uint32_t *tile_off /* [nblocks][P] excl within partition */,
@@ -452,7 +455,7 @@ __global__ __launch_bounds__(WPB * WAVE) void k_scatter_staged(
                 pidr[g] = 0;
                 continue;
             }
-            pidr[g] = pid_in[row];
+            if constexpr (!RHASH) pidr[g] = pid_in[row];
 #pragma unroll
             for (int c = 0; c < (NC > 0 ? NC : MAXC); c++) {
                 if (NC == 0 && c >= a.n_cols) break;
@@ -478,6 +481,16 @@ __global__ __launch_bounds__(WPB * WAVE) void k_scatter_staged(
                     }
                 }
                 if (NC == 0 && col.valid) valv[g][c] = col.valid[row];
+            }
+            if constexpr (RHASH) {
+                /* same normative pid as K1 computed for the histogram, from the values
+                 * already in registers (dd_hash_device.h dd_row_hash_bits) */
+                const uint64_t h = dd_row_hash_bits(a, &colv[g][0]);
+                const uint32_t tot = a.pid_total;
+                uint32_t fine = ((tot & (tot - 1)) == 0)
+                                    ? (uint32_t)(h & (uint64_t)(tot - 1))
+                                    : (uint32_t)(h % (uint64_t)tot);
+                pidr[g] = fine >> a.pid_shift;
             }
         }
     };
@@ -815,6 +828,25 @@ __global__ void k4_part_boffsets(const uint64_t *out_off, const uint64_t *part_o
 
 /* ---------------- launchers (called from dd_host.cpp) ---------------- */
 
+/* one spec-path launch, fully typed; shared by the RHASH and pid variants (C++ linkage:
+ * templates cannot live inside the extern "C" block below) */
+template <int G, int C, int N, bool RH>
+static hipError_t dd_launch_spec1(const dd_kargs *a, dim3 grid, int64_t tile_rows,
+                                  uint32_t nparts, int nbits, const uint32_t *pid_in,
+                                  const uint32_t *tile_off, const uint64_t *part_offsets,
+                                  size_t lds_bytes, hipStream_t s) {
+    if (lds_bytes > 65536) {
+        hipError_t e =
+            hipFuncSetAttribute((const void *)k_scatter_staged<G, 16, C, false, N, RH>,
+                                hipFuncAttributeMaxDynamicSharedMemorySize, (int)lds_bytes);
+        if (e != hipSuccess) return e;
+    }
+    hipLaunchKernelGGL((k_scatter_staged<G, 16, C, false, N, RH>), grid, dim3(16 * WAVE),
+                       lds_bytes, s, *a, tile_rows, nparts, nbits, pid_in, tile_off,
+                       part_offsets);
+    return hipGetLastError();
+}
+
 extern "C" {
 
 hipError_t dd_launch_off64_to_off32(const uint64_t *off64, int64_t lo, int64_t n,
@@ -915,17 +947,12 @@ hipError_t dd_launch_scatter_staged(const dd_kargs *a, int64_t nblocks, int64_t 
         if (a->cols[c].valid || a->cols[c].elem == 0) can_spec = false;
 #define DD_SPEC(G, N)                                                                        \
     if (can_spec && gmax == G && a->n_cols == N) {                                           \
-        if (lds_bytes > 65536) {                                                             \
-            hipError_t e =                                                                   \
-                hipFuncSetAttribute((const void *)k_scatter_staged<G, 16, 4, false, N>,      \
-                                    hipFuncAttributeMaxDynamicSharedMemorySize,              \
-                                    (int)lds_bytes);                                         \
-            if (e != hipSuccess) return e;                                                   \
-        }                                                                                    \
-        hipLaunchKernelGGL((k_scatter_staged<G, 16, 4, false, N>), grid, dim3(16 * WAVE),    \
-                           lds_bytes, s, *a, tile_rows, nparts, nbits, pid_in, tile_off,     \
-                           part_offsets);                                                    \
-        return hipGetLastError();                                                            \
+        return a->rhash ? dd_launch_spec1<G, 4, N, true>(a, grid, tile_rows, nparts, nbits,  \
+                                                         pid_in, tile_off, part_offsets,    \
+                                                         lds_bytes, s)                       \
+                        : dd_launch_spec1<G, 4, N, false>(a, grid, tile_rows, nparts, nbits, \
+                                                          pid_in, tile_off, part_offsets,   \
+                                                          lds_bytes, s);                     \
     }
     DD_SPEC(2, 1)
     DD_SPEC(2, 2)
@@ -939,17 +966,12 @@ hipError_t dd_launch_scatter_staged(const dd_kargs *a, int64_t nblocks, int64_t 
     /* 5-8 fixed columns: same fast path at MAXC=8 */
 #define DD_SPEC8(G, N)                                                                       \
     if (can_spec8 && gmax == G && a->n_cols == N) {                                          \
-        if (lds_bytes > 65536) {                                                             \
-            hipError_t e =                                                                   \
-                hipFuncSetAttribute((const void *)k_scatter_staged<G, 16, 8, false, N>,      \
-                                    hipFuncAttributeMaxDynamicSharedMemorySize,              \
-                                    (int)lds_bytes);                                         \
-            if (e != hipSuccess) return e;                                                   \
-        }                                                                                    \
-        hipLaunchKernelGGL((k_scatter_staged<G, 16, 8, false, N>), grid, dim3(16 * WAVE),    \
-                           lds_bytes, s, *a, tile_rows, nparts, nbits, pid_in, tile_off,     \
-                           part_offsets);                                                    \
-        return hipGetLastError();                                                            \
+        return a->rhash ? dd_launch_spec1<G, 8, N, true>(a, grid, tile_rows, nparts, nbits,  \
+                                                         pid_in, tile_off, part_offsets,    \
+                                                         lds_bytes, s)                       \
+                        : dd_launch_spec1<G, 8, N, false>(a, grid, tile_rows, nparts, nbits, \
+                                                          pid_in, tile_off, part_offsets,   \
+                                                          lds_bytes, s);                     \
     }
     bool can_spec8 = !hasvar && wpb == 16 && a->n_cols >= 5 && a->n_cols <= 8;
     for (int c = 0; c < a->n_cols && can_spec8; c++)
@@ -963,6 +985,9 @@ hipError_t dd_launch_scatter_staged(const dd_kargs *a, int64_t nblocks, int64_t 
     DD_SPEC8(4, 7)
     DD_SPEC8(4, 8)
 #undef DD_SPEC8
+    /* rhash batches MUST take a spec path above (no pid array exists to read) — fail
+     * loudly rather than fall through to a pid-consuming variant */
+    if (a->rhash) return hipErrorInvalidValue;
 #define DD_CASE(G, W, C, V)                                                                  \
     if (gmax == G && wpb == W && maxc == C && hasvar == V) {                                 \
         if (lds_bytes > 65536) {                                                             \

@@ -113,7 +113,10 @@ dd_status dd_partitioner_wait_phase2(dd_partitioner *p, void *stream);
 void dd_partitioner_destroy(dd_partitioner *p);
 
 /* result accessors (pointers are device memory owned by the partitioner) */
-const uint32_t *dd_partitioner_pids(const dd_partitioner *p);      /* u32[n_rows] */
+/* u32[n_rows], or NULL when the partitioner recomputes hashes in-kernel (all-fixed
+ * no-validity batches with integer keys skip the pid array; partition membership is
+ * still fully defined by col_data + row_offsets) */
+const uint32_t *dd_partitioner_pids(const dd_partitioner *p);
 const void *dd_partitioner_col_data(const dd_partitioner *p, int32_t col); /* partition-major */
 const uint8_t *dd_partitioner_col_validity(const dd_partitioner *p, int32_t col);
 const uint32_t *dd_partitioner_col_lengths(const dd_partitioner *p, int32_t col); /* utf8 */
