@@ -249,6 +249,10 @@ def main():
     h2d_gbps = total_bytes_in / th2d / 1e9
 
     part = api.Partitioner(batch, key_idx, p_total)
+    if not part.has_pid_array():
+        # rhash spec path: K3 recomputes hashes from preloaded registers;
+        # the 4 B/row pid read is not part of its algorithmic bytes
+        k3_bytes -= 4 * rows
     comm = create_comm(rank, world) if world > 1 else None
     pipeline = world == 1 and args.pipeline and torch.cuda.is_available()
     if pipeline:

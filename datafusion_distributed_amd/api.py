@@ -252,6 +252,11 @@ class Partitioner:
     def sync(self):
         _check(lib().dd_device_sync())
 
+    def has_pid_array(self):
+        """Fast null-check of the device pid pointer (no download): False when the spec
+        path recomputes hashes in-kernel (rhash) and no pid array exists."""
+        return bool(lib().dd_partitioner_pids(self.h))
+
     def pids(self):
         """Per-row partition ids, or None when the spec path recomputed hashes in-kernel
         (no pid array exists; outputs + row_offsets fully define the partitioning)."""
