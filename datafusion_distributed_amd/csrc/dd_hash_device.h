@@ -94,20 +94,6 @@ __device__ __forceinline__ uint64_t dd_row_hash(const dd_kargs &a, int64_t i) {
     return h;
 }
 
-/* row hash from PRELOADED zero-extended register bits (staged scatter's RHASH path):
- * bits[c] holds column c's value zero-extended to u64, exactly what dd_value_hash_dev
- * would mix for the integer/bool dtypes the host gates in (no validity, no float/dict —
- * dd_host.cpp rhash eligibility). Combine identical to dd_row_hash with all rows valid. */
-__device__ __forceinline__ uint64_t dd_row_hash_bits(const dd_kargs &a,
-                                                     const uint64_t *bits) {
-    uint64_t h = 0;
-    for (int k = 0; k < a.n_keys; k++) {
-        uint64_t vh = dd_mix64(bits[a.key_idx[k]]);
-        h = h ^ (vh + 0x9e3779b97f4a7c15ULL + (h << 6) + (h >> 2));
-    }
-    return h;
-}
-
 /* ballot-multisplit: lanes with equal pid (among `act`); returns the equal-mask */
 __device__ __forceinline__ uint64_t dd_eq_mask(uint32_t pid, uint64_t act, int nbits) {
     uint64_t eq = act;

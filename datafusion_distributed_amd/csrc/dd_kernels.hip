@@ -455,7 +455,33 @@ __global__ __launch_bounds__(WPB * WAVE) void k_scatter_staged(
                 pidr[g] = 0;
                 continue;
             }
-            if constexpr (!RHASH) pidr[g] = pid_in[row];
+            if constexpr (!RHASH) {
+                pidr[g] = pid_in[row];
+            } else {
+                /* KEY loads issued first and hashed immediately: pidr then depends only
+                 * on these loads, not the whole preload — hashing after the full colv
+                 * loop made rank() wait on every column's load (measured +30% K3). The
+                 * colv loop below re-reads the key columns from the just-fetched line.
+                 * Same normative pid as K1 (dd_row_hash over always-valid int keys). */
+                uint64_t h = 0;
+                for (int k = 0; k < a.n_keys; k++) {
+                    const dd_kcol &col = a.cols[a.key_idx[k]];
+                    uint64_t bits = 0;
+                    switch (col.elem) {
+                    case 1: bits = ((const uint8_t *)col.data)[row]; break;
+                    case 2: bits = ((const uint16_t *)col.data)[row]; break;
+                    case 4: bits = ((const uint32_t *)col.data)[row]; break;
+                    case 8: bits = ((const uint64_t *)col.data)[row]; break;
+                    }
+                    uint64_t vh = dd_mix64(bits);
+                    h = h ^ (vh + 0x9e3779b97f4a7c15ULL + (h << 6) + (h >> 2));
+                }
+                const uint32_t tot = a.pid_total;
+                uint32_t fine = ((tot & (tot - 1)) == 0)
+                                    ? (uint32_t)(h & (uint64_t)(tot - 1))
+                                    : (uint32_t)(h % (uint64_t)tot);
+                pidr[g] = fine >> a.pid_shift;
+            }
 #pragma unroll
             for (int c = 0; c < (NC > 0 ? NC : MAXC); c++) {
                 if (NC == 0 && c >= a.n_cols) break;
@@ -481,16 +507,6 @@ __global__ __launch_bounds__(WPB * WAVE) void k_scatter_staged(
                     }
                 }
                 if (NC == 0 && col.valid) valv[g][c] = col.valid[row];
-            }
-            if constexpr (RHASH) {
-                /* same normative pid as K1 computed for the histogram, from the values
-                 * already in registers (dd_hash_device.h dd_row_hash_bits) */
-                const uint64_t h = dd_row_hash_bits(a, &colv[g][0]);
-                const uint32_t tot = a.pid_total;
-                uint32_t fine = ((tot & (tot - 1)) == 0)
-                                    ? (uint32_t)(h & (uint64_t)(tot - 1))
-                                    : (uint32_t)(h % (uint64_t)tot);
-                pidr[g] = fine >> a.pid_shift;
             }
         }
     };
