@@ -312,12 +312,18 @@ extern "C" dd_status dd_partitioner_create(const dd_batch_desc *batch, const int
     }
     const int64_t nchunks = p->nchunks;
 
-    /* rhash: spec-path batches (wpb 16, all fixed, no validity) with integer/bool keys
-     * recompute the row hash in K3 from the already-preloaded register values — the pid
-     * array (4 B/row HBM write in K1 + read in K3) is skipped entirely. Must mirror the
-     * launcher's can_spec/can_spec8 conditions exactly (dd_launch_scatter_staged fails
-     * loudly if not). DD_RHASH=0 disables for A/B. */
-    if (p->staged && p->wpb == 16 && nvar == 0 && batch->n_cols <= DD_STAGE_MAXC) {
+    /* rhash (opt-in, DD_RHASH=1): spec-path batches (wpb 16, all fixed, no validity)
+     * with integer/bool keys recompute the row hash in K3 from its own key loads and
+     * skip the pid array (4 B/row HBM write in K1 + read in K3). MEASURED NEGATIVE as a
+     * default at the bench shape (K3 1.65-1.70 ms vs 1.30 ms, same box, both the
+     * hash-after-preload and keys-first forms): the pid load's latency is fully hidden
+     * in the staged pipeline, while pidr's dependency on hashed key loads stalls the
+     * wave-synchronous rank() ballots — 0.24 GB less traffic does not buy back a 27%
+     * stall. Kept parity-tested for re-evaluation on future silicon.
+     * Must mirror the launcher's can_spec/can_spec8 conditions exactly
+     * (dd_launch_scatter_staged fails loudly if not). */
+    if (getenv("DD_RHASH") && atoi(getenv("DD_RHASH")) == 1 && p->staged &&
+        p->wpb == 16 && nvar == 0 && batch->n_cols <= DD_STAGE_MAXC) {
         bool rhash = true;
         for (int c = 0; c < batch->n_cols && rhash; c++)
             if (batch->cols[c].validity || fixed_elem_size(batch->cols[c].dtype) == 0)
@@ -328,7 +334,6 @@ extern "C" dd_status dd_partitioner_create(const dd_batch_desc *batch, const int
                   dt == DD_DT_I32 || dt == DD_DT_I64))
                 rhash = false; /* float needs canon, dict32 a table gather: pid path */
         }
-        if (getenv("DD_RHASH") && atoi(getenv("DD_RHASH")) == 0) rhash = false;
         ka.rhash = rhash ? 1 : 0;
     }
 
