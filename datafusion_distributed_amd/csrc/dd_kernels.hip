@@ -665,6 +665,258 @@ __global__ __launch_bounds__(WPB * WAVE) void k_scatter_staged(
     }
 }
 
+/* ================= HL: hidden-load scatter experiment (opt-in DD_K3_HL=1) ==========
+ * The plain spec kernel's ISA carries 194 `s_waitcnt vmcnt(0)` FULL drains and zero
+ * counted waits (llvm-objdump of k_scatter_staged<4,16,4,false,4>): every consumer wait
+ * also drains the in-flight flush stores, serializing stores against the next round's
+ * loads (SQ_WAIT 75 %, latency-bound). This clone hides the preload loads
+ * from hipcc in inline asm and counts the VMEM queue BY HAND (guide §5 trap 4b: "hide
+ * the register operand's loads in inline asm and count both queues by hand"):
+ *   - per round: 4 pid loads issued first, then 16 column loads, then (next iteration)
+ *     16 flush stores — vmcnt decrements in issue order, so
+ *   - place waits vmcnt(16): column loads retired, flush stores of the previous round
+ *     still in flight (never drained in-loop);
+ *   - rank runs AFTER the flush and waits vmcnt(32): pid loads retired, this round's
+ *     column loads + stores in flight.
+ * Loads are unconditional with the row clamped to the round end so the instruction
+ * count (and therefore the hand counts) is exact in ragged rounds; the flush is
+ * likewise padded (clamped index rewrites the same value — idempotent). Fixed shape:
+ * GMAX=4, WPB=16, 4 fixed columns, no validity, element sizes (8,8,8,4)/(8,8,8,8).
+ * Waitcnt asm ties the loaded values as "+v" operands so no use can be scheduled
+ * before the wait. Parity: tests/test_gpu_fuzz.py::test_hl_ab. */
+
+#define HL_LD64(dst, p) \
+    asm volatile("global_load_dwordx2 %0, %1, off" : "=v"(dst) : "v"(p))
+#define HL_LD32(dst, p) asm volatile("global_load_dword %0, %1, off" : "=v"(dst) : "v"(p))
+
+template <int E> struct hl_elem { using T = uint32_t; };
+template <> struct hl_elem<8> { using T = uint64_t; };
+
+template <int E, typename T>
+__device__ __forceinline__ void hl_load(T &dst, const void *base, uint32_t row) {
+    if constexpr (E == 8) {
+        HL_LD64(dst, (const uint64_t *)base + row);
+    } else {
+        static_assert(E == 4, "HL instantiations cover elem 4/8 only");
+        HL_LD32(dst, (const uint32_t *)base + row);
+    }
+}
+
+template <int E0, int E1, int E2, int E3>
+__global__ __launch_bounds__(16 * WAVE) void k_scatter_hl(
+    dd_kargs a, int64_t tile_rows, uint32_t nparts, int nbits, const uint32_t *pid_in,
+    const uint32_t *tile_off, const uint64_t *part_offsets) {
+    constexpr int WPB = 16, GMAX = 4;
+    constexpr int BT = WPB * WAVE;
+    constexpr int R = GMAX * BT;
+    constexpr int SEG = R / WPB;
+    extern __shared__ __attribute__((aligned(16))) char smem[];
+    char *ws = smem;
+    uint64_t *dstbase = (uint64_t *)ws;
+    ws += sizeof(uint64_t) * nparts;
+    uint32_t *seghist = (uint32_t *)ws;
+    ws += sizeof(uint32_t) * WPB * nparts;
+    uint32_t *roundcnt = (uint32_t *)ws;
+    ws += sizeof(uint32_t) * nparts;
+    uint32_t *round_off = (uint32_t *)ws;
+    ws += sizeof(uint32_t) * nparts;
+    uint32_t *scan_tmp = (uint32_t *)ws;
+    ws += sizeof(uint32_t) * BT;
+    uint32_t *dstg = (uint32_t *)ws;
+    ws += sizeof(uint32_t) * R;
+    char *const stage0 = ws;
+
+    const int tid = threadIdx.x;
+    const int wid = tid / WAVE;
+    const int lane = tid % WAVE;
+    uint32_t *myseg = seghist + (size_t)wid * nparts;
+    const uint64_t lt = ((uint64_t)1 << lane) - 1;
+
+    const int64_t tstart = (int64_t)blockIdx.x * tile_rows;
+    const int64_t tend = (tstart + tile_rows < a.n_rows) ? (tstart + tile_rows) : a.n_rows;
+
+    for (uint32_t p = tid; p < nparts; p += BT)
+        dstbase[p] = part_offsets[p] + tile_off[(size_t)blockIdx.x * nparts + p];
+    __syncthreads();
+
+    uint32_t pidr[GMAX], rankr[GMAX];
+    bool actr[GMAX];
+    typename hl_elem<E0>::T c0v[GMAX];
+    typename hl_elem<E1>::T c1v[GMAX];
+    typename hl_elem<E2>::T c2v[GMAX];
+    typename hl_elem<E3>::T c3v[GMAX];
+
+/* waits tie every value they guard as "+v" so uses cannot move above them */
+#define HL_WAIT_PID(N)                                                                       \
+    asm volatile("s_waitcnt vmcnt(" #N ")"                                                   \
+                 : "+v"(pidr[0]), "+v"(pidr[1]), "+v"(pidr[2]), "+v"(pidr[3])::"memory")
+#define HL_WAIT_COLV(N)                                                                      \
+    asm volatile("s_waitcnt vmcnt(" #N ")"                                                   \
+                 : "+v"(c0v[0]), "+v"(c0v[1]), "+v"(c0v[2]), "+v"(c0v[3]), "+v"(c1v[0]),     \
+                   "+v"(c1v[1]), "+v"(c1v[2]), "+v"(c1v[3]), "+v"(c2v[0]), "+v"(c2v[1]),     \
+                   "+v"(c2v[2]), "+v"(c2v[3]), "+v"(c3v[0]), "+v"(c3v[1]), "+v"(c3v[2]),     \
+                   "+v"(c3v[3])::"memory")
+
+    auto preload = [&](int64_t rstart, int64_t rend) {
+        const int64_t segstart = rstart + (int64_t)wid * SEG;
+        uint32_t rowc[GMAX];
+#pragma unroll
+        for (int g = 0; g < GMAX; g++) {
+            const int64_t row = segstart + g * WAVE + lane;
+            actr[g] = row < rend;
+            /* clamp: every load always issues so the hand counts stay exact */
+            rowc[g] = (uint32_t)(actr[g] ? row : rend - 1);
+        }
+#pragma unroll
+        for (int g = 0; g < GMAX; g++) HL_LD32(pidr[g], pid_in + rowc[g]);
+#pragma unroll
+        for (int g = 0; g < GMAX; g++) hl_load<E0>(c0v[g], a.cols[0].data, rowc[g]);
+#pragma unroll
+        for (int g = 0; g < GMAX; g++) hl_load<E1>(c1v[g], a.cols[1].data, rowc[g]);
+#pragma unroll
+        for (int g = 0; g < GMAX; g++) hl_load<E2>(c2v[g], a.cols[2].data, rowc[g]);
+#pragma unroll
+        for (int g = 0; g < GMAX; g++) hl_load<E3>(c3v[g], a.cols[3].data, rowc[g]);
+    };
+
+    auto rank = [&]() {
+        for (uint32_t p = lane; p < nparts; p += WAVE) myseg[p] = 0;
+#pragma unroll
+        for (int g = 0; g < GMAX; g++) {
+            const bool active = actr[g];
+            const uint32_t pid = pidr[g];
+            uint64_t act = __ballot(active);
+            uint32_t rk = 0;
+            if (active) {
+                uint64_t eq = dd_eq_mask(pid, act, nbits);
+                int leader = __ffsll((unsigned long long)eq) - 1;
+                uint32_t base = 0;
+                if (lane == leader) {
+                    base = myseg[pid];
+                    myseg[pid] = base + (uint32_t)__popcll((unsigned long long)eq);
+                }
+                base = (uint32_t)__shfl((int)base, leader);
+                rk = base + (uint32_t)__popcll((unsigned long long)(eq & lt));
+            }
+            rankr[g] = rk;
+        }
+    };
+
+    if (tstart < tend) {
+        const int64_t rend0 = (tstart + R < tend) ? (tstart + R) : tend;
+        preload(tstart, rend0);
+        HL_WAIT_PID(16); /* 16 column loads stay in flight */
+        rank();
+        HL_WAIT_COLV(0); /* tile prologue only: drain once */
+    }
+    __syncthreads();
+
+    for (int64_t rstart = tstart; rstart < tend; rstart += R) {
+        const int64_t rend = (rstart + R < tend) ? (rstart + R) : tend;
+        const int round_rows = (int)(rend - rstart);
+
+        { /* fused cross-wave + partition-offset scan (same as k_scatter_staged) */
+            const uint32_t span = (nparts + BT - 1) / BT;
+            const uint32_t plo = tid * span;
+            const uint32_t phi = (plo + span < nparts) ? plo + span : nparts;
+            uint32_t ssum = 0;
+            for (uint32_t p = plo; p < phi; p++) {
+                uint32_t run = 0;
+#pragma unroll
+                for (int w = 0; w < WPB; w++) {
+                    uint32_t v = seghist[(size_t)w * nparts + p];
+                    seghist[(size_t)w * nparts + p] = run;
+                    run += v;
+                }
+                roundcnt[p] = run;
+                ssum += run;
+            }
+            scan_tmp[tid] = ssum;
+            __syncthreads();
+            if (tid < WAVE) {
+                uint32_t carry = 0;
+                for (int k = 0; k < BT / WAVE; k++) {
+                    uint32_t v = scan_tmp[k * WAVE + tid];
+#pragma unroll
+                    for (int d = 1; d < WAVE; d <<= 1) {
+                        uint32_t u = (uint32_t)__shfl_up((int)v, d);
+                        if (tid >= d) v += u;
+                    }
+                    v += carry;
+                    scan_tmp[k * WAVE + tid] = v;
+                    carry = (uint32_t)__shfl((int)v, WAVE - 1);
+                }
+            }
+            __syncthreads();
+            uint32_t run = (tid > 0) ? scan_tmp[tid - 1] : 0;
+            for (uint32_t p = plo; p < phi; p++) {
+                round_off[p] = run;
+                run += roundcnt[p];
+            }
+            __syncthreads();
+        }
+
+        /* place: column loads retired at vmcnt(16) — the previous round's 16 flush
+         * stores are the only VMEM allowed to remain in flight */
+        HL_WAIT_COLV(16);
+#pragma unroll
+        for (int g = 0; g < GMAX; g++) {
+            if (!actr[g]) continue;
+            const uint32_t pid = pidr[g];
+            const uint32_t rank_r = myseg[pid] + rankr[g];
+            const uint32_t slot = round_off[pid] + rank_r;
+            dstg[slot] = (uint32_t)(dstbase[pid] + rank_r);
+            char *stage = stage0;
+            ((typename hl_elem<E0>::T *)stage)[slot] = c0v[g];
+            stage += (size_t)R * E0;
+            ((typename hl_elem<E1>::T *)stage)[slot] = c1v[g];
+            stage += (size_t)R * E1;
+            ((typename hl_elem<E2>::T *)stage)[slot] = c2v[g];
+            stage += (size_t)R * E2;
+            ((typename hl_elem<E3>::T *)stage)[slot] = c3v[g];
+        }
+        __syncthreads();
+
+        const bool more = rstart + R < tend;
+        if (more) {
+            const int64_t nrend = (rstart + 2 * R < tend) ? (rstart + 2 * R) : tend;
+            preload(rstart + R, nrend);
+        }
+
+        /* flush: padded to exactly GMAX iterations (clamped index rewrites the same
+         * value) so the store count per lane is always GMAX*4 = 16 */
+#pragma unroll
+        for (int u = 0; u < GMAX; u++) {
+            const int i = tid + u * BT;
+            const int ic = (i < round_rows) ? i : (round_rows - 1);
+            const uint64_t dst = dstg[ic];
+            char *stage = stage0;
+            ((typename hl_elem<E0>::T *)a.cols[0].out_data)[dst] =
+                ((const typename hl_elem<E0>::T *)stage)[ic];
+            stage += (size_t)R * E0;
+            ((typename hl_elem<E1>::T *)a.cols[1].out_data)[dst] =
+                ((const typename hl_elem<E1>::T *)stage)[ic];
+            stage += (size_t)R * E1;
+            ((typename hl_elem<E2>::T *)a.cols[2].out_data)[dst] =
+                ((const typename hl_elem<E2>::T *)stage)[ic];
+            stage += (size_t)R * E2;
+            ((typename hl_elem<E3>::T *)a.cols[3].out_data)[dst] =
+                ((const typename hl_elem<E3>::T *)stage)[ic];
+        }
+
+        if (more) {
+            /* pid loads retired at vmcnt(32): this round's 16 column loads + 16 flush
+             * stores stay in flight across the barrier */
+            HL_WAIT_PID(32);
+            rank();
+        }
+        for (uint32_t p = tid; p < nparts; p += BT) dstbase[p] += roundcnt[p];
+        __syncthreads();
+    }
+#undef HL_WAIT_PID
+#undef HL_WAIT_COLV
+}
+
 /* ================= K4: var-width bytes for the staged path =================
  * The staged scatter (v2) handles var columns' LENGTHS and a ROWID permutation as
  * synthetic fixed u32 columns (DD_KDT_VARLEN / DD_KDT_ROWID, set up by dd_host.cpp).
@@ -955,6 +1207,27 @@ hipError_t dd_launch_scatter_staged(const dd_kargs *a, int64_t nblocks, int64_t 
                                     const uint32_t *tile_off, const uint64_t *part_offsets,
                                     int gmax, int wpb, size_t lds_bytes, hipStream_t s) {
     dim3 grid((unsigned)nblocks);
+    if (a->hl) { /* hidden-load experiment (host gate: ka.hl, DD_K3_HL=1) */
+        if (wpb != 16 || gmax != 4 || a->n_cols != 4) return hipErrorInvalidValue;
+        const int e0 = a->cols[0].elem, e1 = a->cols[1].elem, e2 = a->cols[2].elem,
+                  e3 = a->cols[3].elem;
+#define DD_HL(A, B, C, D)                                                                    \
+    if (e0 == A && e1 == B && e2 == C && e3 == D) {                                          \
+        if (lds_bytes > 65536) {                                                             \
+            hipError_t e = hipFuncSetAttribute((const void *)k_scatter_hl<A, B, C, D>,       \
+                                               hipFuncAttributeMaxDynamicSharedMemorySize,   \
+                                               (int)lds_bytes);                              \
+            if (e != hipSuccess) return e;                                                   \
+        }                                                                                    \
+        hipLaunchKernelGGL((k_scatter_hl<A, B, C, D>), grid, dim3(16 * WAVE), lds_bytes, s,  \
+                           *a, tile_rows, nparts, nbits, pid_in, tile_off, part_offsets);    \
+        return hipGetLastError();                                                            \
+    }
+        DD_HL(8, 8, 8, 4)
+        DD_HL(8, 8, 8, 8)
+#undef DD_HL
+        return hipErrorInvalidValue;
+    }
     const int maxc = (a->n_cols <= 4) ? 4 : 8;
     const bool hasvar = a->n_var > 0;
     /* specialized fast path: wpb 16, <=4 fixed columns, no validity, no var */

@@ -98,3 +98,18 @@ def test_rhash_ab(monkeypatch):
         check_against_oracle(cols, keys, nparts)
         monkeypatch.delenv("DD_RHASH")
         check_against_oracle(cols, keys, nparts)
+
+def test_hl_ab(monkeypatch):
+    """Hidden-load scatter experiment (DD_K3_HL=1: inline-asm preload + hand-counted
+    s_waitcnt; dd_kernels.hip HL header) must match the oracle bit-exactly on its
+    gated shape (4 fixed cols, elems 8/8/8/{4,8}, no validity), including ragged last
+    rounds and non-pow2 P; then the default path on the same batch."""
+    rng = np.random.default_rng(777)
+    for n in [4096 * 3, 100_000, 250_001]:  # multiple of R, ragged, very ragged
+        cols = [random_col(rng, n, "i64", 0), random_col(rng, n, "f64", 0),
+                random_col(rng, n, "f64", 0), random_col(rng, n, "i32", 0)]
+        for keys, nparts in [([0], 128), ([3, 0], 100)]:
+            monkeypatch.setenv("DD_K3_HL", "1")
+            check_against_oracle(cols, keys, nparts)
+            monkeypatch.delenv("DD_K3_HL")
+            check_against_oracle(cols, keys, nparts)
