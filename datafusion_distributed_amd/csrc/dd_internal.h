@@ -52,10 +52,11 @@ struct dd_kargs {
                            registers in K3 (no pid array: K1 skips its store, K3 its
                            load). Host gates: all-fixed no-validity batch, integer/bool
                            keys, wpb==16 (DD_RHASH=0 disables). */
-    int32_t hl;         /* hidden-load scatter experiment (DD_K3_HL=1): preload loads in
-                           inline asm + hand-counted s_waitcnt so flush stores never
-                           drain mid-loop. gmax 4, wpb 16, 4 fixed cols, no validity,
-                           elems (8,8,8,4)/(8,8,8,8); mutually exclusive with rhash. */
+    int32_t hl;         /* hidden-load scatter (default on for its shape; DD_K3_HL=0
+                           reverts): preload loads in inline asm + hand-counted
+                           s_waitcnt so flush stores never drain mid-loop. gmax 4,
+                           wpb 16, 4 fixed cols, no validity, elems in {4,8};
+                           mutually exclusive with rhash. */
     int32_t key_idx[DD_KMAX_KEYS];
     int32_t var_idx[DD_KMAX_VAR];
     dd_kcol cols[DD_KMAX_COLS];

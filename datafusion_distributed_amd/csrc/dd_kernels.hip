@@ -680,10 +680,12 @@ __global__ __launch_bounds__(WPB * WAVE) void k_scatter_staged(
  *     column loads + stores in flight.
  * Loads are unconditional with the row clamped to the round end so the instruction
  * count (and therefore the hand counts) is exact in ragged rounds; the flush is
- * likewise padded (clamped index rewrites the same value — idempotent). Fixed shape:
- * GMAX=4, WPB=16, 4 fixed columns, no validity, element sizes (8,8,8,4)/(8,8,8,8).
- * Waitcnt asm ties the loaded values as "+v" operands so no use can be scheduled
- * before the wait. Parity: tests/test_gpu_fuzz.py::test_hl_ab. */
+ * likewise padded (clamped index rewrites the same value — idempotent). Gated shape:
+ * GMAX=4, WPB=16, 4 fixed columns, no validity, element sizes in {4,8}^4 (16
+ * instantiations). Waitcnt asm ties the loaded values as "+v" operands so no use can
+ * be scheduled before the wait. DEFAULT for its shape since the A/B: K3 1.015 vs
+ * 1.166 ms, headline +10.6 % on the same box (DD_K3_HL=0 reverts; profiles/).
+ * Parity: tests/test_gpu_fuzz.py::test_hl_ab. */
 
 #define HL_LD64(dst, p) \
     asm volatile("global_load_dwordx2 %0, %1, off" : "=v"(dst) : "v"(p))
@@ -1223,6 +1225,20 @@ hipError_t dd_launch_scatter_staged(const dd_kargs *a, int64_t nblocks, int64_t 
                            *a, tile_rows, nparts, nbits, pid_in, tile_off, part_offsets);    \
         return hipGetLastError();                                                            \
     }
+        DD_HL(4, 4, 4, 4)
+        DD_HL(4, 4, 4, 8)
+        DD_HL(4, 4, 8, 4)
+        DD_HL(4, 4, 8, 8)
+        DD_HL(4, 8, 4, 4)
+        DD_HL(4, 8, 4, 8)
+        DD_HL(4, 8, 8, 4)
+        DD_HL(4, 8, 8, 8)
+        DD_HL(8, 4, 4, 4)
+        DD_HL(8, 4, 4, 8)
+        DD_HL(8, 4, 8, 4)
+        DD_HL(8, 4, 8, 8)
+        DD_HL(8, 8, 4, 4)
+        DD_HL(8, 8, 4, 8)
         DD_HL(8, 8, 8, 4)
         DD_HL(8, 8, 8, 8)
 #undef DD_HL

@@ -100,16 +100,22 @@ def test_rhash_ab(monkeypatch):
         check_against_oracle(cols, keys, nparts)
 
 def test_hl_ab(monkeypatch):
-    """Hidden-load scatter experiment (DD_K3_HL=1: inline-asm preload + hand-counted
-    s_waitcnt; dd_kernels.hip HL header) must match the oracle bit-exactly on its
-    gated shape (4 fixed cols, elems 8/8/8/{4,8}, no validity), including ragged last
-    rounds and non-pow2 P; then the default path on the same batch."""
+    """Hidden-load scatter (default for its gated shape: inline-asm preload +
+    hand-counted s_waitcnt; dd_kernels.hip HL header) must match the oracle
+    bit-exactly — including ragged last rounds and non-pow2 P — and so must the plain
+    spec path on the same batch (DD_K3_HL=0). Also covers a mixed 4/8-elem shape."""
     rng = np.random.default_rng(777)
     for n in [4096 * 3, 100_000, 250_001]:  # multiple of R, ragged, very ragged
         cols = [random_col(rng, n, "i64", 0), random_col(rng, n, "f64", 0),
                 random_col(rng, n, "f64", 0), random_col(rng, n, "i32", 0)]
         for keys, nparts in [([0], 128), ([3, 0], 100)]:
-            monkeypatch.setenv("DD_K3_HL", "1")
+            monkeypatch.delenv("DD_K3_HL", raising=False)  # default: HL
             check_against_oracle(cols, keys, nparts)
-            monkeypatch.delenv("DD_K3_HL")
+            monkeypatch.setenv("DD_K3_HL", "0")  # plain spec path
             check_against_oracle(cols, keys, nparts)
+    n = 123_457
+    cols = [random_col(rng, n, "i32", 0), random_col(rng, n, "i64", 0),
+            random_col(rng, n, "f32", 0), random_col(rng, n, "f64", 0)]
+    monkeypatch.delenv("DD_K3_HL", raising=False)
+    check_against_oracle(cols, [0, 1], 128)
+    check_against_oracle(cols, [2, 3], 100)
