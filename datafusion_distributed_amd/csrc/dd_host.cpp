@@ -352,6 +352,25 @@ extern "C" dd_status dd_partitioner_create(const dd_batch_desc *batch, const int
         }
         if (okhl) ka.hl = 1;
     }
+    /* generalized HL (k_scatter_hlg): exact whitelisted shapes only — must mirror the
+     * launcher's DD_HLG table */
+    if (!ka.rhash && !ka.hl && !(getenv("DD_K3_HL") && atoi(getenv("DD_K3_HL")) == 0) &&
+        p->staged && p->wpb == 16) {
+        bool novalid = true;
+        for (int c = 0; c < batch->n_cols && novalid; c++)
+            if (batch->cols[c].validity) novalid = false;
+        auto elems_are = [&](std::initializer_list<int> want) {
+            if ((int)want.size() != batch->n_cols) return false;
+            int c = 0;
+            for (int w : want)
+                if (fixed_elem_size(batch->cols[c++].dtype) != w) return false;
+            return true;
+        };
+        if (novalid && nvar == 0 &&
+            ((p->gmax == 4 && elems_are({8, 8, 8, 4, 4})) ||
+             (p->gmax == 2 && elems_are({1, 1, 8, 8, 8, 8, 4}))))
+            ka.hl = 2;
+    }
 
     auto halloc = [&](void **ptr, size_t bytes) {
         return hipMalloc(ptr, bytes > 0 ? bytes : 1) == hipSuccess;

@@ -26,6 +26,8 @@
 #include <hip/hip_runtime.h>
 #include <stdint.h>
 
+#include <type_traits> /* integral_constant for the hlg compile-time column walk */
+
 #include "dd_internal.h"
 
 #define WAVE 64
@@ -919,6 +921,272 @@ __global__ __launch_bounds__(16 * WAVE) void k_scatter_hl(
 #undef HL_WAIT_COLV
 }
 
+/* ---- generalized hidden-load scatter (k_scatter_hlg): same discipline as
+ * k_scatter_hl for other column counts / widths / GMAX, with the wait counts computed
+ * at compile time: per round L = GMAX*NC column loads and L padded flush stores, so
+ *   place waits vmcnt(L)   (prev round's stores in flight),
+ *   rank  waits vmcnt(2L)  (this round's col loads + stores in flight; needs 2L <= 63).
+ * Ties are one tiny s_waitcnt per guarded value (the first does the real wait, the
+ * rest are satisfied single-cycle SALU ops) — that sidesteps variadic asm operand
+ * lists. Whitelisted instantiations only (launcher): the multikey bench shape
+ * (8,8,8,4,4)@G4 and the q1 shape (1,1,8,8,8,8,4)@G2. */
+
+template <int CNT, typename T>
+__device__ __forceinline__ void hl_tie_wait(T &v) {
+    asm volatile("s_waitcnt vmcnt(%1)" : "+v"(v) : "n"(CNT) : "memory");
+}
+
+template <int J, int NC, typename F>
+__device__ __forceinline__ void hl_for(F &&f) {
+    if constexpr (J < NC) {
+        f(std::integral_constant<int, J>{});
+        hl_for<J + 1, NC>(f);
+    }
+}
+
+template <int G, int... Es>
+__global__ __launch_bounds__(16 * WAVE) void k_scatter_hlg(
+    dd_kargs a, int64_t tile_rows, uint32_t nparts, int nbits, const uint32_t *pid_in,
+    const uint32_t *tile_off, const uint64_t *part_offsets) {
+    constexpr int WPB = 16, GMAX = G;
+    constexpr int BT = WPB * WAVE;
+    constexpr int R = GMAX * BT;
+    constexpr int SEG = R / WPB;
+    constexpr int NC = sizeof...(Es);
+    constexpr int EL[NC] = {Es...};
+    constexpr int L = GMAX * NC;
+    static_assert(2 * L <= 63, "vmcnt immediate is 6 bits");
+    /* compile-time storage split: 8-byte columns -> u64 slots, narrower -> u32 slots */
+    struct slots {
+        int idx[NC];
+        int n8, n4;
+        constexpr slots() : idx{}, n8(0), n4(0) {
+            for (int j = 0; j < NC; j++) idx[j] = (EL[j] == 8) ? n8++ : n4++;
+        }
+    };
+    constexpr slots S{};
+
+    extern __shared__ __attribute__((aligned(16))) char smem[];
+    char *ws = smem;
+    uint64_t *dstbase = (uint64_t *)ws;
+    ws += sizeof(uint64_t) * nparts;
+    uint32_t *seghist = (uint32_t *)ws;
+    ws += sizeof(uint32_t) * WPB * nparts;
+    uint32_t *roundcnt = (uint32_t *)ws;
+    ws += sizeof(uint32_t) * nparts;
+    uint32_t *round_off = (uint32_t *)ws;
+    ws += sizeof(uint32_t) * nparts;
+    uint32_t *scan_tmp = (uint32_t *)ws;
+    ws += sizeof(uint32_t) * BT;
+    uint32_t *dstg = (uint32_t *)ws;
+    ws += sizeof(uint32_t) * R;
+    char *const stage0 = ws;
+
+    const int tid = threadIdx.x;
+    const int wid = tid / WAVE;
+    const int lane = tid % WAVE;
+    uint32_t *myseg = seghist + (size_t)wid * nparts;
+    const uint64_t lt = ((uint64_t)1 << lane) - 1;
+
+    const int64_t tstart = (int64_t)blockIdx.x * tile_rows;
+    const int64_t tend = (tstart + tile_rows < a.n_rows) ? (tstart + tile_rows) : a.n_rows;
+
+    for (uint32_t p = tid; p < nparts; p += BT)
+        dstbase[p] = part_offsets[p] + tile_off[(size_t)blockIdx.x * nparts + p];
+    __syncthreads();
+
+    uint32_t pidr[GMAX], rankr[GMAX];
+    bool actr[GMAX];
+    uint64_t big[S.n8 > 0 ? S.n8 : 1][GMAX];
+    uint32_t sml[S.n4 > 0 ? S.n4 : 1][GMAX];
+
+    auto preload = [&](int64_t rstart, int64_t rend) {
+        const int64_t segstart = rstart + (int64_t)wid * SEG;
+        uint32_t rowc[GMAX];
+#pragma unroll
+        for (int g = 0; g < GMAX; g++) {
+            const int64_t row = segstart + g * WAVE + lane;
+            actr[g] = row < rend;
+            rowc[g] = (uint32_t)(actr[g] ? row : rend - 1); /* always issue (counts) */
+        }
+#pragma unroll
+        for (int g = 0; g < GMAX; g++) HL_LD32(pidr[g], pid_in + rowc[g]);
+        hl_for<0, NC>([&](auto jc) {
+            constexpr int J = jc.value;
+            const void *base = a.cols[J].data;
+#pragma unroll
+            for (int g = 0; g < GMAX; g++) {
+                if constexpr (EL[J] == 8) {
+                    HL_LD64(big[S.idx[J]][g], (const uint64_t *)base + rowc[g]);
+                } else if constexpr (EL[J] == 4) {
+                    HL_LD32(sml[S.idx[J]][g], (const uint32_t *)base + rowc[g]);
+                } else if constexpr (EL[J] == 2) {
+                    asm volatile("global_load_ushort %0, %1, off"
+                                 : "=v"(sml[S.idx[J]][g])
+                                 : "v"((const uint16_t *)base + rowc[g]));
+                } else {
+                    asm volatile("global_load_ubyte %0, %1, off"
+                                 : "=v"(sml[S.idx[J]][g])
+                                 : "v"((const uint8_t *)base + rowc[g]));
+                }
+            }
+        });
+    };
+
+    auto wait_pid = [&](auto cnt) {
+#pragma unroll
+        for (int g = 0; g < GMAX; g++) hl_tie_wait<cnt.value>(pidr[g]);
+    };
+    auto wait_colv = [&](auto cnt) {
+        hl_for<0, NC>([&](auto jc) {
+            constexpr int J = jc.value;
+#pragma unroll
+            for (int g = 0; g < GMAX; g++) {
+                if constexpr (EL[J] == 8) hl_tie_wait<cnt.value>(big[S.idx[J]][g]);
+                else hl_tie_wait<cnt.value>(sml[S.idx[J]][g]);
+            }
+        });
+    };
+
+    auto rank = [&]() {
+        for (uint32_t p = lane; p < nparts; p += WAVE) myseg[p] = 0;
+#pragma unroll
+        for (int g = 0; g < GMAX; g++) {
+            const bool active = actr[g];
+            const uint32_t pid = pidr[g];
+            uint64_t act = __ballot(active);
+            uint32_t rk = 0;
+            if (active) {
+                uint64_t eq = dd_eq_mask(pid, act, nbits);
+                int leader = __ffsll((unsigned long long)eq) - 1;
+                uint32_t base = 0;
+                if (lane == leader) {
+                    base = myseg[pid];
+                    myseg[pid] = base + (uint32_t)__popcll((unsigned long long)eq);
+                }
+                base = (uint32_t)__shfl((int)base, leader);
+                rk = base + (uint32_t)__popcll((unsigned long long)(eq & lt));
+            }
+            rankr[g] = rk;
+        }
+    };
+
+    if (tstart < tend) {
+        const int64_t rend0 = (tstart + R < tend) ? (tstart + R) : tend;
+        preload(tstart, rend0);
+        wait_pid(std::integral_constant<int, L>{});
+        rank();
+        wait_colv(std::integral_constant<int, 0>{}); /* prologue-only drain */
+    }
+    __syncthreads();
+
+    for (int64_t rstart = tstart; rstart < tend; rstart += R) {
+        const int64_t rend = (rstart + R < tend) ? (rstart + R) : tend;
+        const int round_rows = (int)(rend - rstart);
+
+        { /* fused cross-wave + partition-offset scan (same as k_scatter_staged) */
+            const uint32_t span = (nparts + BT - 1) / BT;
+            const uint32_t plo = tid * span;
+            const uint32_t phi = (plo + span < nparts) ? plo + span : nparts;
+            uint32_t ssum = 0;
+            for (uint32_t p = plo; p < phi; p++) {
+                uint32_t run = 0;
+#pragma unroll
+                for (int w = 0; w < WPB; w++) {
+                    uint32_t v = seghist[(size_t)w * nparts + p];
+                    seghist[(size_t)w * nparts + p] = run;
+                    run += v;
+                }
+                roundcnt[p] = run;
+                ssum += run;
+            }
+            scan_tmp[tid] = ssum;
+            __syncthreads();
+            if (tid < WAVE) {
+                uint32_t carry = 0;
+                for (int k = 0; k < BT / WAVE; k++) {
+                    uint32_t v = scan_tmp[k * WAVE + tid];
+#pragma unroll
+                    for (int d = 1; d < WAVE; d <<= 1) {
+                        uint32_t u = (uint32_t)__shfl_up((int)v, d);
+                        if (tid >= d) v += u;
+                    }
+                    v += carry;
+                    scan_tmp[k * WAVE + tid] = v;
+                    carry = (uint32_t)__shfl((int)v, WAVE - 1);
+                }
+            }
+            __syncthreads();
+            uint32_t run = (tid > 0) ? scan_tmp[tid - 1] : 0;
+            for (uint32_t p = plo; p < phi; p++) {
+                round_off[p] = run;
+                run += roundcnt[p];
+            }
+            __syncthreads();
+        }
+
+        wait_colv(std::integral_constant<int, L>{});
+#pragma unroll
+        for (int g = 0; g < GMAX; g++) {
+            if (!actr[g]) continue;
+            const uint32_t pid = pidr[g];
+            const uint32_t rank_r = myseg[pid] + rankr[g];
+            const uint32_t slot = round_off[pid] + rank_r;
+            dstg[slot] = (uint32_t)(dstbase[pid] + rank_r);
+            char *stage = stage0;
+            hl_for<0, NC>([&](auto jc) {
+                constexpr int J = jc.value;
+                if constexpr (EL[J] == 8)
+                    ((uint64_t *)stage)[slot] = big[S.idx[J]][g];
+                else if constexpr (EL[J] == 4)
+                    ((uint32_t *)stage)[slot] = sml[S.idx[J]][g];
+                else if constexpr (EL[J] == 2)
+                    ((uint16_t *)stage)[slot] = (uint16_t)sml[S.idx[J]][g];
+                else
+                    ((uint8_t *)stage)[slot] = (uint8_t)sml[S.idx[J]][g];
+                stage += (size_t)R * EL[J];
+            });
+        }
+        __syncthreads();
+
+        const bool more = rstart + R < tend;
+        if (more) {
+            const int64_t nrend = (rstart + 2 * R < tend) ? (rstart + 2 * R) : tend;
+            preload(rstart + R, nrend);
+        }
+
+        /* flush padded to GMAX iterations (clamped rewrite is idempotent): store count
+         * per lane is always L */
+#pragma unroll
+        for (int u = 0; u < GMAX; u++) {
+            const int i = tid + u * BT;
+            const int ic = (i < round_rows) ? i : (round_rows - 1);
+            const uint64_t dst = dstg[ic];
+            char *stage = stage0;
+            hl_for<0, NC>([&](auto jc) {
+                constexpr int J = jc.value;
+                void *out = a.cols[J].out_data;
+                if constexpr (EL[J] == 8)
+                    ((uint64_t *)out)[dst] = ((const uint64_t *)stage)[ic];
+                else if constexpr (EL[J] == 4)
+                    ((uint32_t *)out)[dst] = ((const uint32_t *)stage)[ic];
+                else if constexpr (EL[J] == 2)
+                    ((uint16_t *)out)[dst] = ((const uint16_t *)stage)[ic];
+                else
+                    ((uint8_t *)out)[dst] = ((const uint8_t *)stage)[ic];
+                stage += (size_t)R * EL[J];
+            });
+        }
+
+        if (more) {
+            wait_pid(std::integral_constant<int, 2 * L>{});
+            rank();
+        }
+        for (uint32_t p = tid; p < nparts; p += BT) dstbase[p] += roundcnt[p];
+        __syncthreads();
+    }
+}
+
 /* ================= K4: var-width bytes for the staged path =================
  * The staged scatter (v2) handles var columns' LENGTHS and a ROWID permutation as
  * synthetic fixed u32 columns (DD_KDT_VARLEN / DD_KDT_ROWID, set up by dd_host.cpp).
@@ -1209,7 +1477,35 @@ hipError_t dd_launch_scatter_staged(const dd_kargs *a, int64_t nblocks, int64_t 
                                     const uint32_t *tile_off, const uint64_t *part_offsets,
                                     int gmax, int wpb, size_t lds_bytes, hipStream_t s) {
     dim3 grid((unsigned)nblocks);
-    if (a->hl) { /* hidden-load experiment (host gate: ka.hl, DD_K3_HL=1) */
+    if (a->hl == 2) { /* generalized hidden-load (host whitelists the exact shapes) */
+        const int n = a->n_cols;
+        auto el = [&](int c) { return (int)a->cols[c].elem; };
+#define DD_HLG(GM, ...)                                                                      \
+    {                                                                                        \
+        const int want[] = {__VA_ARGS__};                                                    \
+        const int wn = (int)(sizeof(want) / sizeof(want[0]));                                \
+        bool m = (gmax == GM && wpb == 16 && n == wn);                                       \
+        for (int c = 0; c < wn && m; c++) m = el(c) == want[c];                              \
+        if (m) {                                                                             \
+            if (lds_bytes > 65536) {                                                         \
+                hipError_t e =                                                               \
+                    hipFuncSetAttribute((const void *)k_scatter_hlg<GM, __VA_ARGS__>,        \
+                                        hipFuncAttributeMaxDynamicSharedMemorySize,          \
+                                        (int)lds_bytes);                                     \
+                if (e != hipSuccess) return e;                                               \
+            }                                                                                \
+            hipLaunchKernelGGL((k_scatter_hlg<GM, __VA_ARGS__>), grid, dim3(16 * WAVE),      \
+                               lds_bytes, s, *a, tile_rows, nparts, nbits, pid_in,           \
+                               tile_off, part_offsets);                                      \
+            return hipGetLastError();                                                        \
+        }                                                                                    \
+    }
+        DD_HLG(4, 8, 8, 8, 4, 4)       /* multikey bench shape */
+        DD_HLG(2, 1, 1, 8, 8, 8, 8, 4) /* q1 shape */
+#undef DD_HLG
+        return hipErrorInvalidValue;
+    }
+    if (a->hl) { /* hidden-load, 4-column kernel (host gate: ka.hl) */
         if (wpb != 16 || gmax != 4 || a->n_cols != 4) return hipErrorInvalidValue;
         const int e0 = a->cols[0].elem, e1 = a->cols[1].elem, e2 = a->cols[2].elem,
                   e3 = a->cols[3].elem;

@@ -119,3 +119,23 @@ def test_hl_ab(monkeypatch):
     monkeypatch.delenv("DD_K3_HL", raising=False)
     check_against_oracle(cols, [0, 1], 128)
     check_against_oracle(cols, [2, 3], 100)
+
+def test_hlg_ab(monkeypatch):
+    """Generalized hidden-load scatter (k_scatter_hlg, whitelisted 5-col multikey and
+    7-col q1 shapes) vs the plain spec path (DD_K3_HL=0), both bit-exact vs the
+    oracle, including ragged rounds and non-pow2 P."""
+    rng = np.random.default_rng(999)
+    for n in [100_000, 250_001]:
+        mk = [random_col(rng, n, "i64", 0), random_col(rng, n, "f64", 0),
+              random_col(rng, n, "f64", 0), random_col(rng, n, "i32", 0),
+              random_col(rng, n, "i32", 0)]
+        q1 = [random_col(rng, n, "u8", 0), random_col(rng, n, "bool", 0),
+              random_col(rng, n, "f64", 0), random_col(rng, n, "f64", 0),
+              random_col(rng, n, "f64", 0), random_col(rng, n, "f64", 0),
+              random_col(rng, n, "i32", 0)]
+        for cols, keys, nparts in [(mk, [0, 4], 128), (mk, [0], 100),
+                                   (q1, [0, 1], 128), (q1, [0, 1], 7)]:
+            monkeypatch.delenv("DD_K3_HL", raising=False)  # default: hlg where gated
+            check_against_oracle(cols, keys, nparts)
+            monkeypatch.setenv("DD_K3_HL", "0")
+            check_against_oracle(cols, keys, nparts)
