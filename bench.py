@@ -118,8 +118,25 @@ def make_clickbench_userid(rows, seed):
     ], [0]
 
 
+def make_dictkey(rows, seed):
+    """SURVEY §8(d) microbench variant 'dictionary key': dict32(idx→utf8) key over the
+    q3 payload — exercises the precomputed per-distinct-value hash table (K0 dict pass)
+    with the scatter on 4-byte indices. 100k distinct values, ≈12 B avg."""
+    rng = np.random.default_rng(seed)
+    nvals = 100_000
+    lens = rng.integers(6, 20, nvals)
+    doff = np.zeros(nvals + 1, dtype=np.int32)
+    doff[1:] = np.cumsum(lens)
+    dbytes = rng.integers(97, 123, int(doff[-1]), dtype=np.int64).astype(np.uint8)
+    cols, _ = make_lineitem_q3(rows, seed)
+    cols[0] = {"dtype": "dict32", "data": rng.integers(0, nvals, rows).astype(np.int32),
+               "dict_bytes": dbytes, "dict_offsets": doff, "valid": None}
+    return cols, [0]
+
+
 WORKLOADS = {
     "tpch_sf10_lineitem_shuffle": (make_lineitem_q3, SF10_ROWS),
+    "dictkey_shuffle": (make_dictkey, 30_000_000),
     "tpch_sf10_multikey_shuffle": (make_lineitem_q3_multikey, SF10_ROWS),
     "tpch_sf10_utf8key_shuffle": (make_lineitem_utf8key, 30_000_000),
     "tpch_sf1_q1_repartition": (make_lineitem_q1, 6_001_215),
