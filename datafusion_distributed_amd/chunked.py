@@ -4,8 +4,12 @@ bytes per column — DESIGN.md §5) by row-chunking and device-side concatenatio
 Chunks are processed in input order and each partition's chunk segments are concatenated
 in chunk order, so the global result is exactly the stable partition of the whole input
 (bit-identical to running one oversized launch / the oracle). This is the harness-level
-orchestration of SURVEY §8(d)'s "chunked to fit" SF100/ClickBench configs; memory peak is
-final outputs + one in-flight chunk.
+orchestration of SURVEY §8(d)'s "chunked to fit" SF100/ClickBench configs. Memory peak:
+every chunk's PARTITIONED output stays resident until the global offsets are known (the
+final placement of chunk c's partition p depends on all chunks' counts), plus the final
+buffers and one in-flight INPUT chunk (input batches are freed as soon as their kernels
+complete) — i.e. ≈ 2× the total payload + one chunk; a count-only pre-pass would reduce
+it to final + one chunk at the cost of hashing twice (round-2 note).
 """
 
 import ctypes
@@ -97,6 +101,7 @@ def chunked_partition(cols, key_idx, nparts, max_chunk_rows=2**31,
         part = api.Partitioner(batch, key_idx, nparts)
         part.run()
         part.sync()
+        batch.free()  # input consumed; only the partitioned output must stay resident
         roff = part.row_offsets()
         rows_cp[ci] = roff[1:] - roff[:-1]
         boffs = {}
@@ -104,7 +109,7 @@ def chunked_partition(cols, key_idx, nparts, max_chunk_rows=2**31,
             bo = part.byte_offsets(i)
             bytes_cp[i][ci] = bo[1:] - bo[:-1]
             boffs[i] = bo
-        part_meta.append((batch, part, roff, boffs))
+        part_meta.append((part, roff, boffs))
 
     # global offsets: partition-major, chunk order within a partition (stable)
     part_rows = rows_cp.sum(axis=0)
@@ -114,7 +119,7 @@ def chunked_partition(cols, key_idx, nparts, max_chunk_rows=2**31,
              for i in bytes_cp}
 
     for ci in range(nchunks):
-        batch, part, roff, boffs = part_meta[ci]
+        part, roff, boffs = part_meta[ci]
         row_prior = rows_cp[:ci].sum(axis=0) if ci else np.zeros(nparts, dtype=np.int64)
         for p in range(nparts):
             gdst_row = int(grow[p] + row_prior[p])
@@ -137,7 +142,6 @@ def chunked_partition(cols, key_idx, nparts, max_chunk_rows=2**31,
                     d2d(final[i]["valid"], gdst_row,
                         L.dd_partitioner_col_validity(part.h, i), src_lo, nrows)
         part.destroy()
-        batch.free()
 
     # download for the harness caller
     out_cols = []
