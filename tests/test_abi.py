@@ -54,3 +54,16 @@ def test_batch_desc_layout_matches_c():
     assert ctypes.sizeof(api.ColDesc) == 8 * 8  # int32+pad, 3 ptrs, i64, 2 ptrs, i64
     assert ctypes.sizeof(api.BatchDesc) == 8 + 8 + api.DD_MAX_COLS * ctypes.sizeof(api.ColDesc)
     assert ctypes.sizeof(api.TaskKeyC) == 32
+
+
+def test_null_argument_guards():
+    """Null-pointer arguments return DD_ERR_INVALID with a message — never crash.
+    These guards run before any device work, so they are CPU-testable."""
+    L = api.lib()
+    assert L.dd_comm_unique_id(None) == 1  # DD_ERR_INVALID
+    out = ctypes.c_void_p()
+    assert L.dd_partitioner_create(None, None, 0, 0, ctypes.byref(out)) == 1
+    key = api.TaskKeyC(0, 0, 0, 0)
+    assert L.dd_execute_task(ctypes.byref(key), 0, 0, None, None) == 1
+    assert L.dd_set_plan(None, None, None, 0, 0) == 1
+    assert b"" != L.dd_last_error()  # message was set
