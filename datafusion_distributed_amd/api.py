@@ -457,8 +457,10 @@ def partial_reduce(batch: DeviceBatch, key_idx, aggs):
 
     aggs: list of (col_index_or_None, op) with op in {"sum_f64", "count", "sum_i64"}.
     Returns {"keys": u64[n, nk] canonical bits, "keynull": u32[n] per-key null bitmask,
-    "aggs": f64[n, na]} — integer aggregates (count / sum_i64) are bit-cast in the f64
-    slots (view with .view(np.int64)). May contain duplicate groups (partial semantics).
+    "aggs": f64[n, na], "nn": u64[n, na] non-null input counts} — integer aggregates
+    (count / sum_i64) are bit-cast in the f64 slots (view with .view(np.int64)). May
+    contain duplicate groups (partial semantics). The final merge sums "nn" per group and
+    emits NULL for an aggregate whose total non-null count is 0 (DataFusion semantics).
     """
     nk = len(key_idx)
     na = len(aggs)
@@ -474,11 +476,14 @@ def partial_reduce(batch: DeviceBatch, key_idx, aggs):
         keys = np.empty((n, nk), dtype=np.uint64)
         keynull = np.empty(n, dtype=np.uint32)
         vals = np.empty((n, na), dtype=np.float64)
+        nn = np.zeros((n, na), dtype=np.uint64)
         if n:
             _check(lib().dd_reducer_fetch(
                 h, keys.ctypes.data_as(ctypes.c_void_p),
                 keynull.ctypes.data_as(ctypes.c_void_p),
                 vals.ctypes.data_as(ctypes.c_void_p)))
-        return {"keys": keys, "keynull": keynull, "aggs": vals, "kernel_ms": kernel_ms}
+            _check(lib().dd_reducer_fetch_nn(h, nn.ctypes.data_as(ctypes.c_void_p)))
+        return {"keys": keys, "keynull": keynull, "aggs": vals, "nn": nn,
+                "kernel_ms": kernel_ms}
     finally:
         lib().dd_reducer_destroy(h)

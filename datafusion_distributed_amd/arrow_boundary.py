@@ -78,11 +78,17 @@ def partition_batches(part, lo, hi, batch_size=8192):
                     )
                 elif c["dtype"] == "dict32":
                     idx = pa.array(fixed[i][b0:b1], type=pa.int32(), mask=mask)
-                    doff = np.asarray(c["dict_offsets"], dtype=np.int64)
-                    dbuf = np.asarray(c["dict_bytes"], dtype=np.uint8).tobytes()
-                    values = pa.array(
-                        [dbuf[doff[k]:doff[k + 1]].decode("utf-8", "replace")
-                         for k in range(len(doff) - 1)])
+                    # Rebuild values from the raw offsets/bytes buffers: the device path
+                    # hashes/moves raw bytes, so round-tripping through str would silently
+                    # mutate (or mask) non-UTF8 dictionary bytes.
+                    doff = np.asarray(c["dict_offsets"], dtype=np.int32)
+                    dbuf = np.asarray(c["dict_bytes"], dtype=np.uint8)
+                    values = pa.StringArray.from_buffers(
+                        len(doff) - 1,
+                        pa.py_buffer(doff.tobytes()),
+                        pa.py_buffer(dbuf.tobytes()),
+                        None,
+                    )
                     arr = pa.DictionaryArray.from_arrays(idx, values)
                 elif c["dtype"] == "bool":
                     arr = pa.array(fixed[i][b0:b1].astype(bool), type=pa.bool_(), mask=mask)

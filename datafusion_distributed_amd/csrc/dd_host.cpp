@@ -13,6 +13,7 @@
 #include <cstdlib>
 #include <cstdio>
 #include <cstring>
+#include <atomic>
 #include <map>
 #include <memory>
 #include <mutex>
@@ -142,7 +143,7 @@ struct dd_partitioner {
     uint32_t *out_lengths[DD_KMAX_COLS] = {};
 
     hipEvent_t ev[5] = {}; /* 0 K1start 1 K1end 2 scans-end 3 K3end 4 K3start */
-    bool has_run = false;
+    std::atomic<bool> has_run{false}; /* release/acquire pairs with dd_execute_task */
 
     ~dd_partitioner() {
         (void)hipFree(pid);
@@ -217,8 +218,11 @@ extern "C" dd_status dd_partitioner_create(const dd_batch_desc *batch, const int
         kc.offsets = cd.offsets;
         if (cd.dtype == DD_DT_UTF8) {
             if (ka.n_var >= DD_KMAX_VAR) return fail(DD_ERR_UNSUPPORTED, "too many var columns");
-            if (cd.data_len < 0 || cd.data_len >= (int64_t)UINT32_MAX)
-                return fail(DD_ERR_UNSUPPORTED, "utf8 data >= 4 GB: chunk the batch");
+            /* Arrow utf8 offsets are int32: a column past INT32_MAX bytes has already
+             * overflowed its own offsets — reject rather than read garbage (chunked.py
+             * splits larger inputs before they get here) */
+            if (cd.data_len < 0 || cd.data_len > (int64_t)INT32_MAX)
+                return fail(DD_ERR_UNSUPPORTED, "utf8 data > 2 GB (int32 Arrow offsets): chunk the batch");
             ka.var_idx[ka.n_var++] = c;
         }
     }
@@ -520,7 +524,7 @@ extern "C" dd_status dd_partitioner_run_phase2(dd_partitioner *p, void *stream) 
                                   p->part_boffsets, p->lds_k3, s));
     }
     HIP_TRY(hipEventRecord(p->ev[3], s));
-    p->has_run = true;
+    p->has_run.store(true, std::memory_order_release);
     return DD_OK;
 }
 
@@ -605,8 +609,14 @@ struct TaskKeyCmp {
     }
 };
 
+/* Entries are shared_ptr so a concurrent dd_drop_task / dd_set_plan replacing the key
+ * cannot free the partitioner out from under an in-flight dd_execute_task (the execute
+ * call pins its own reference for the duration of the call — the reference worker keeps
+ * task state alive the same way via Arc'd entries in its TTI cache, task_data.rs:16-29).
+ * The raw pointer RETURNED to the caller is only guaranteed while the cache entry lives:
+ * see the lifetime note on dd_execute_task in dd_shuffle.h. */
 static std::mutex g_tasks_mu;
-static std::map<dd_task_key, dd_partitioner *, TaskKeyCmp> g_tasks;
+static std::map<dd_task_key, std::shared_ptr<dd_partitioner>, TaskKeyCmp> g_tasks;
 
 extern "C" dd_status dd_set_plan(const dd_task_key *key, const dd_batch_desc *batch,
                                  const int32_t *key_cols, int32_t n_keys,
@@ -615,36 +625,32 @@ extern "C" dd_status dd_set_plan(const dd_task_key *key, const dd_batch_desc *ba
     dd_partitioner *p = nullptr;
     dd_status st = dd_partitioner_create(batch, key_cols, n_keys, n_partitions, &p);
     if (st != DD_OK) return st;
+    std::shared_ptr<dd_partitioner> sp(p, dd_partitioner_destroy);
     std::lock_guard<std::mutex> g(g_tasks_mu);
-    auto it = g_tasks.find(*key);
-    if (it != g_tasks.end()) {
-        delete it->second;
-        it->second = p;
-    } else {
-        g_tasks[*key] = p;
-    }
+    g_tasks[*key] = std::move(sp); /* replaces; pinned in-flight executes keep the old one */
     return DD_OK;
 }
 
 extern "C" dd_status dd_execute_task(const dd_task_key *key, uint32_t part_lo, uint32_t part_hi,
                                      void *stream, dd_partitioner **out) {
     if (!key || !out) return set_err(DD_ERR_INVALID, "null argument");
-    dd_partitioner *p = nullptr;
+    std::shared_ptr<dd_partitioner> pin; /* keeps p alive for the whole call */
     {
         std::lock_guard<std::mutex> g(g_tasks_mu);
         auto it = g_tasks.find(*key);
         if (it == g_tasks.end())
             return set_err(DD_ERR_NOT_FOUND,
                            "unknown TaskKey (no plan was set: cf. impl_execute_task.rs:29-34)");
-        p = it->second;
+        pin = it->second;
     }
+    dd_partitioner *p = pin.get();
     if (part_lo > part_hi || part_hi > p->nparts)
         return set_err(DD_ERR_INVALID, "partition range out of bounds");
-    if (!p->has_run) {
+    if (!p->has_run.load(std::memory_order_acquire)) {
         /* run under the cache lock: launches are async enqueues (cheap), and this keeps
          * two concurrent execute_task calls on the same key from double-launching */
         std::lock_guard<std::mutex> g(g_tasks_mu);
-        if (!p->has_run) {
+        if (!p->has_run.load(std::memory_order_relaxed)) {
             dd_status st = dd_partitioner_run(p, stream);
             if (st != DD_OK) return st;
         }
@@ -654,10 +660,11 @@ extern "C" dd_status dd_execute_task(const dd_task_key *key, uint32_t part_lo, u
 }
 
 extern "C" dd_status dd_drop_task(const dd_task_key *key) {
+    std::shared_ptr<dd_partitioner> dead; /* destroy outside the lock */
     std::lock_guard<std::mutex> g(g_tasks_mu);
     auto it = g_tasks.find(*key);
     if (it == g_tasks.end()) return set_err(DD_ERR_NOT_FOUND, "unknown TaskKey");
-    delete it->second;
+    dead = std::move(it->second);
     g_tasks.erase(it);
     return DD_OK;
 }
@@ -1110,6 +1117,7 @@ struct dd_reducer {
     uint64_t *keys = nullptr;
     uint32_t *keynull = nullptr;
     double *aggs = nullptr;
+    uint64_t *nn = nullptr; /* [max_rows][n_aggs] non-null input counts */
     uint64_t *n_dev = nullptr;
     int32_t *meta = nullptr; /* device: agg_cols + agg_ops */
     float kernel_ms = 0;     /* hipEvent time of the reduce kernel alone */
@@ -1118,6 +1126,7 @@ struct dd_reducer {
         (void)hipFree(keys);
         (void)hipFree(keynull);
         (void)hipFree(aggs);
+        (void)hipFree(nn);
         (void)hipFree(n_dev);
         (void)hipFree(meta);
     }
@@ -1205,6 +1214,7 @@ extern "C" dd_status dd_partial_reduce_run(const dd_batch_desc *batch,
     if (hipMalloc((void **)&r->keys, (size_t)max_rows * n_keys * 8) != hipSuccess ||
         hipMalloc((void **)&r->keynull, (size_t)max_rows * 4) != hipSuccess ||
         hipMalloc((void **)&r->aggs, (size_t)max_rows * n_aggs * 8) != hipSuccess ||
+        hipMalloc((void **)&r->nn, (size_t)max_rows * n_aggs * 8) != hipSuccess ||
         hipMalloc((void **)&r->n_dev, 8) != hipSuccess ||
         hipMalloc((void **)&r->meta, 2 * 4 * 4) != hipSuccess)
         return fail("partial reduce alloc");
@@ -1223,7 +1233,7 @@ extern "C" dd_status dd_partial_reduce_run(const dd_batch_desc *batch,
     (void)hipEventRecord(e0, s);
     hipError_t e = dd_launch_partial_reduce(&ka, nblocks, chunk, n_aggs, r->meta,
                                             r->meta + 4, r->keys, r->keynull, r->aggs,
-                                            r->n_dev, s);
+                                            r->nn, r->n_dev, s);
     (void)hipEventRecord(e1, s);
     if (e != hipSuccess) return fail(hipGetErrorString(e));
     if (hipStreamSynchronize(s) != hipSuccess) return fail("sync");
@@ -1249,6 +1259,15 @@ extern "C" dd_status dd_reducer_fetch(const dd_reducer *r, uint64_t *host_keys,
     HIP_TRY(hipMemcpy(host_keynull, r->keynull, (size_t)r->n_rows * 4,
                       hipMemcpyDeviceToHost));
     HIP_TRY(hipMemcpy(host_aggs, r->aggs, (size_t)r->n_rows * r->n_aggs * 8,
+                      hipMemcpyDeviceToHost));
+    return DD_OK;
+}
+
+/* non-null input counts per output row per aggregate ([n_rows][n_aggs]): the final
+ * merge sums these; total 0 => the merged aggregate is NULL (DataFusion semantics for
+ * SUM/MIN/MAX over an all-null group) */
+extern "C" dd_status dd_reducer_fetch_nn(const dd_reducer *r, uint64_t *host_nn) {
+    HIP_TRY(hipMemcpy(host_nn, r->nn, (size_t)r->n_rows * r->n_aggs * 8,
                       hipMemcpyDeviceToHost));
     return DD_OK;
 }

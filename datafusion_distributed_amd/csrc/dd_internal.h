@@ -95,7 +95,7 @@ hipError_t dd_launch_partial_reduce(const dd_kargs *a, int64_t nblocks, int64_t 
                                     int n_aggs, const int32_t *agg_cols,
                                     const int32_t *agg_ops, uint64_t *out_keys,
                                     uint32_t *out_keynull, double *out_aggs,
-                                    uint64_t *out_n, hipStream_t s);
+                                    uint64_t *out_nn, uint64_t *out_n, hipStream_t s);
 }
 
 #endif

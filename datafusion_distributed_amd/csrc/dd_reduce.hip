@@ -16,8 +16,13 @@
  * singleton groups (sum = value, count = 1) — correct for any cardinality, fast for the
  * low-cardinality keys partial-reduce targets (q1: 6 groups).
  *
- * Round-1 coverage: fixed-width key columns (nullable), aggregate ops
- * sum(f64) / sum(i64) / count(*). */
+ * Coverage: fixed-width key columns (nullable), aggregate ops
+ * sum(f64) / sum(i64) / count(*) / min / max.
+ *
+ * NULL semantics: each aggregate carries a NON-NULL INPUT COUNT in the partial state
+ * (out_nn), so the downstream final merge can emit NULL for a group whose inputs were all
+ * NULL — matching DataFusion, where SUM/MIN/MAX over an all-null group is NULL, not the
+ * op identity. (COUNT counts every row it sees; its merge can never be NULL.) */
 
 #include <hip/hip_runtime.h>
 #include <stdint.h>
@@ -110,12 +115,14 @@ extern "C" __global__ __launch_bounds__(R_THREADS) void k_partial_reduce(
     const int32_t *agg_ops, uint64_t *out_keys /* [max_rows][n_keys] interleaved */,
     uint32_t *out_keynull /* [max_rows] bitmask */,
     double *out_aggs /* [max_rows][n_aggs] interleaved (i64 sums bit-cast) */,
+    uint64_t *out_nn /* [max_rows][n_aggs] non-null input counts */,
     uint64_t *out_n /* global row counter */) {
     __shared__ uint64_t t_hash[R_CAP];
     __shared__ uint32_t t_ready[R_CAP];
     __shared__ uint64_t t_keys[DD_KMAX_KEYS > 4 ? 4 : DD_KMAX_KEYS][R_CAP];
     __shared__ uint32_t t_null[R_CAP];
     __shared__ unsigned long long t_agg[4][R_CAP]; /* f64 or i64 state, bit pattern */
+    __shared__ uint32_t t_nn[4][R_CAP];            /* non-null input count per agg */
 
     const int nk = a.n_keys > 4 ? 4 : a.n_keys;
     for (int s = threadIdx.x; s < R_CAP; s += R_THREADS) {
@@ -123,7 +130,10 @@ extern "C" __global__ __launch_bounds__(R_THREADS) void k_partial_reduce(
         t_ready[s] = 0;
         t_null[s] = 0;
         for (int k = 0; k < nk; k++) t_keys[k][s] = 0;
-        for (int g = 0; g < n_aggs; g++) t_agg[g][s] = dd_agg_identity(agg_ops[g]);
+        for (int g = 0; g < n_aggs; g++) {
+            t_agg[g][s] = dd_agg_identity(agg_ops[g]);
+            t_nn[g][s] = 0;
+        }
     }
     __syncthreads();
 
@@ -190,6 +200,7 @@ extern "C" __global__ __launch_bounds__(R_THREADS) void k_partial_reduce(
                     const dd_kcol &c = a.cols[agg_cols[g]];
                     if (c.valid && !c.valid[row]) break;
                     atomicAdd((double *)&t_agg[g][slot], ((const double *)c.data)[row]);
+                    atomicAdd(&t_nn[g][slot], 1u);
                     break;
                 }
                 case DD_AGG_SUM_I64: {
@@ -197,6 +208,7 @@ extern "C" __global__ __launch_bounds__(R_THREADS) void k_partial_reduce(
                     if (c.valid && !c.valid[row]) break;
                     atomicAdd(&t_agg[g][slot],
                               (unsigned long long)((const uint64_t *)c.data)[row]);
+                    atomicAdd(&t_nn[g][slot], 1u);
                     break;
                 }
                 case DD_AGG_MIN_F64:
@@ -208,6 +220,7 @@ extern "C" __global__ __launch_bounds__(R_THREADS) void k_partial_reduce(
                         atomicMin(&t_agg[g][slot], (unsigned long long)k);
                     else
                         atomicMax(&t_agg[g][slot], (unsigned long long)k);
+                    atomicAdd(&t_nn[g][slot], 1u);
                     break;
                 }
                 case DD_AGG_MIN_I64:
@@ -219,10 +232,12 @@ extern "C" __global__ __launch_bounds__(R_THREADS) void k_partial_reduce(
                         atomicMin(&t_agg[g][slot], (unsigned long long)k);
                     else
                         atomicMax(&t_agg[g][slot], (unsigned long long)k);
+                    atomicAdd(&t_nn[g][slot], 1u);
                     break;
                 }
                 case DD_AGG_COUNT:
                     atomicAdd(&t_agg[g][slot], 1ull);
+                    atomicAdd(&t_nn[g][slot], 1u);
                     break;
                 }
             }
@@ -233,10 +248,13 @@ extern "C" __global__ __launch_bounds__(R_THREADS) void k_partial_reduce(
             out_keynull[o] = knull;
             for (int g = 0; g < n_aggs; g++) {
                 double v = 0;
+                uint64_t nn = 1;
                 switch (agg_ops[g]) {
                 case DD_AGG_SUM_F64: {
                     const dd_kcol &c = a.cols[agg_cols[g]];
-                    v = (c.valid && !c.valid[row]) ? 0.0 : ((const double *)c.data)[row];
+                    const bool null = c.valid && !c.valid[row];
+                    if (null) nn = 0;
+                    v = null ? 0.0 : ((const double *)c.data)[row];
                     out_aggs[o * n_aggs + g] = v;
                     break;
                 }
@@ -244,6 +262,7 @@ extern "C" __global__ __launch_bounds__(R_THREADS) void k_partial_reduce(
                 case DD_AGG_MAX_F64: {
                     const dd_kcol &c = a.cols[agg_cols[g]];
                     const bool null = c.valid && !c.valid[row];
+                    if (null) nn = 0;
                     out_aggs[o * n_aggs + g] =
                         null ? dd_f64_unkey(dd_agg_identity(agg_ops[g]))
                              : ((const double *)c.data)[row];
@@ -253,6 +272,7 @@ extern "C" __global__ __launch_bounds__(R_THREADS) void k_partial_reduce(
                 case DD_AGG_MAX_I64: {
                     const dd_kcol &c = a.cols[agg_cols[g]];
                     const bool null = c.valid && !c.valid[row];
+                    if (null) nn = 0;
                     out_aggs[o * n_aggs + g] = __longlong_as_double(
                         null ? dd_i64_unkey(dd_agg_identity(agg_ops[g]))
                              : ((const int64_t *)c.data)[row]);
@@ -260,9 +280,9 @@ extern "C" __global__ __launch_bounds__(R_THREADS) void k_partial_reduce(
                 }
                 case DD_AGG_SUM_I64: {
                     const dd_kcol &c = a.cols[agg_cols[g]];
-                    uint64_t iv = (c.valid && !c.valid[row])
-                                      ? 0
-                                      : ((const uint64_t *)c.data)[row];
+                    const bool null = c.valid && !c.valid[row];
+                    if (null) nn = 0;
+                    uint64_t iv = null ? 0 : ((const uint64_t *)c.data)[row];
                     out_aggs[o * n_aggs + g] = __longlong_as_double((long long)iv);
                     break;
                 }
@@ -270,6 +290,7 @@ extern "C" __global__ __launch_bounds__(R_THREADS) void k_partial_reduce(
                     out_aggs[o * n_aggs + g] = __longlong_as_double(1ll);
                     break;
                 }
+                out_nn[o * n_aggs + g] = nn;
             }
         }
     }
@@ -295,6 +316,7 @@ extern "C" __global__ __launch_bounds__(R_THREADS) void k_partial_reduce(
             default:
                 out_aggs[o * n_aggs + g] = __longlong_as_double((long long)st);
             }
+            out_nn[o * n_aggs + g] = (uint64_t)t_nn[g][s];
         }
     }
 }
@@ -304,9 +326,10 @@ extern "C" hipError_t dd_launch_partial_reduce(const dd_kargs *a, int64_t nblock
                                                const int32_t *agg_cols,
                                                const int32_t *agg_ops, uint64_t *out_keys,
                                                uint32_t *out_keynull, double *out_aggs,
-                                               uint64_t *out_n, hipStream_t s) {
+                                               uint64_t *out_nn, uint64_t *out_n,
+                                               hipStream_t s) {
     hipLaunchKernelGGL(k_partial_reduce, dim3((unsigned)nblocks), dim3(R_THREADS), 0, s, *a,
                        chunk_rows, n_aggs, agg_cols, agg_ops, out_keys, out_keynull,
-                       out_aggs, out_n);
+                       out_aggs, out_nn, out_n);
     return hipGetLastError();
 }

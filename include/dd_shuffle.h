@@ -113,9 +113,11 @@ dd_status dd_partitioner_wait_phase2(dd_partitioner *p, void *stream);
 void dd_partitioner_destroy(dd_partitioner *p);
 
 /* result accessors (pointers are device memory owned by the partitioner) */
-/* u32[n_rows], or NULL when the partitioner recomputes hashes in-kernel (all-fixed
- * no-validity batches with integer keys skip the pid array; partition membership is
- * still fully defined by col_data + row_offsets) */
+/* u32[n_rows] of per-row partition ids. NULL ONLY when the opt-in DD_RHASH=1 recompute
+ * path is active (all-fixed no-validity batches with integer keys, measured slower and
+ * OFF by default — DESIGN.md §9): shim authors must handle NULL for robustness but will
+ * not see it under default configuration; partition membership is always fully defined
+ * by col_data + row_offsets either way. */
 const uint32_t *dd_partitioner_pids(const dd_partitioner *p);
 const void *dd_partitioner_col_data(const dd_partitioner *p, int32_t col); /* partition-major */
 const uint8_t *dd_partitioner_col_validity(const dd_partitioner *p, int32_t col);
@@ -221,7 +223,12 @@ dd_status dd_set_plan(const dd_task_key *key, const dd_batch_desc *batch,
                       const int32_t *key_cols, int32_t n_keys, uint32_t n_partitions);
 /* Executes (or reuses) the cached task's partitioner and returns it; partition range
  * [part_lo, part_hi) mirrors ExecuteTaskRequest.target_partition_start/end. The returned
- * partitioner is owned by the cache; do not destroy. */
+ * partitioner is owned by the cache; do not destroy. Lifetime: concurrent
+ * set_plan/execute/drop calls on the same key are safe (entries are refcounted and an
+ * in-flight execute pins its entry), but the pointer *out refers to the cache entry as
+ * of this call — it is invalidated by a later dd_drop_task or dd_set_plan on the same
+ * key, so callers must not use it past either (the reference's worker has the same
+ * contract: task state lives until the TTI cache evicts it, task_data.rs:16-29). */
 dd_status dd_execute_task(const dd_task_key *key, uint32_t part_lo, uint32_t part_hi,
                           void *stream, dd_partitioner **out);
 dd_status dd_drop_task(const dd_task_key *key); /* task cleanup (stateful_data_cleanup) */
@@ -253,6 +260,10 @@ float dd_reducer_kernel_ms(const dd_reducer *r); /* reduce-kernel time of the la
  * host_aggs[n][n_aggs] (i64 sums / counts bit-cast into the double slot) */
 dd_status dd_reducer_fetch(const dd_reducer *r, uint64_t *host_keys, uint32_t *host_keynull,
                            double *host_aggs);
+/* host_nn[n][n_aggs] non-null input counts per aggregate: the final merge sums them and
+ * emits NULL when the total is 0 (DataFusion: SUM/MIN/MAX over an all-null group is NULL,
+ * not the op identity; COUNT counts rows and is never NULL) */
+dd_status dd_reducer_fetch_nn(const dd_reducer *r, uint64_t *host_nn);
 void dd_reducer_destroy(dd_reducer *r);
 
 /* ---------------- device helpers (harness convenience; not part of the seam) ------- */

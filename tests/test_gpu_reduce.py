@@ -22,18 +22,26 @@ def merge_partials(res, nk, na):
 
 
 def merge_partials_mixed(res, int_ops):
-    """Merge with integer slots bit-cast (count/sum_i64)."""
+    """Merge with integer slots bit-cast (count/sum_i64). NULL semantics: a non-count
+    aggregate whose total non-null input count (res["nn"]) is 0 merges to None — exactly
+    what DataFusion's mode=Final emits for SUM/MIN/MAX over an all-null group."""
     out = {}
+    nnsum = {}
     for i in range(len(res["keynull"])):
         key = tuple(int(x) for x in res["keys"][i]) + (int(res["keynull"][i]),)
         vals = []
         for g in range(res["aggs"].shape[1]):
             v = res["aggs"][i, g]
             vals.append(int(np.float64(v).view(np.int64)) if g in int_ops else float(v))
+        nn = [int(x) for x in res["nn"][i]]
         if key in out:
             out[key] = [a + b for a, b in zip(out[key], vals)]
+            nnsum[key] = [a + b for a, b in zip(nnsum[key], nn)]
         else:
             out[key] = vals
+            nnsum[key] = nn
+    for key, vals in out.items():
+        out[key] = [None if nnsum[key][g] == 0 else vals[g] for g in range(len(vals))]
     return out
 
 
@@ -134,8 +142,67 @@ def test_partial_reduce_null_keys_and_null_aggs():
         key = (0, 1) if kv is None else (kv, 0)
         got = merged[key]
         assert got[1] == direct["count_all"][i].as_py()
-        want = direct["v_sum"][i].as_py() or 0.0
-        assert abs(got[0] - want) <= 1e-6 * max(abs(want), 1.0)
+        want = direct["v_sum"][i].as_py()
+        if want is None:
+            assert got[0] is None
+        else:
+            assert abs(got[0] - want) <= 1e-6 * max(abs(want), 1.0)
+
+
+def test_partial_reduce_all_null_group_is_null():
+    """A group whose aggregate inputs are ALL NULL must merge to NULL, not the op
+    identity (DataFusion: SUM(v)=NULL, MIN(v)=NULL over an all-null group; COUNT(*)
+    still counts rows). Pinned vs pyarrow direct aggregation."""
+    import pyarrow as pa
+
+    rng = np.random.default_rng(41)
+    n = 50_000
+    k = rng.integers(0, 8, n, dtype=np.int64)
+    v = rng.normal(size=n)
+    vvalid = (k != 3).astype(np.uint8)  # group 3: every aggregate input is NULL
+    batch = api.DeviceBatch([
+        {"dtype": "i64", "data": k, "valid": None},
+        {"dtype": "f64", "data": v, "valid": vvalid},
+    ])
+    res = api.partial_reduce(batch, [0], [(1, "sum_f64"), (1, "min_f64"),
+                                          (None, "count")])
+    batch.free()
+    # ops-aware final merge (sum / min / count), NULL when total non-null count is 0
+    merged = {}
+    nnsum = {}
+    for i in range(len(res["keynull"])):
+        key = (int(res["keys"][i, 0]), int(res["keynull"][i]))
+        vals = [float(res["aggs"][i, 0]), float(res["aggs"][i, 1]),
+                int(np.float64(res["aggs"][i, 2]).view(np.int64))]
+        nn = [int(x) for x in res["nn"][i]]
+        if key in merged:
+            cur = merged[key]
+            # a partial with nn[g]==0 carries the op identity (NaN for min) — skip it
+            merged[key] = [cur[0] + vals[0],
+                           cur[1] if nn[1] == 0 else
+                           (vals[1] if nnsum[key][1] == 0 else min(cur[1], vals[1])),
+                           cur[2] + vals[2]]
+            nnsum[key] = [a + b for a, b in zip(nnsum[key], nn)]
+        else:
+            merged[key] = vals
+            nnsum[key] = nn
+    for key in merged:
+        merged[key] = [None if nnsum[key][g] == 0 else merged[key][g] for g in range(3)]
+
+    varr = pa.array([float(x) if vv else None for x, vv in zip(v, vvalid)])
+    tbl = pa.table({"k": pa.array(k), "v": varr})
+    direct = tbl.group_by("k").aggregate([("v", "sum"), ("v", "min"), ([], "count_all")])
+    assert len(merged) == direct.num_rows
+    for i in range(direct.num_rows):
+        kv = direct["k"][i].as_py()
+        got = merged[(kv, 0)]
+        for gi, col in ((0, "v_sum"), (1, "v_min")):
+            want = direct[col][i].as_py()
+            if want is None:
+                assert got[gi] is None, f"group {kv} {col}: want NULL, got {got[gi]}"
+            else:
+                assert abs(got[gi] - want) <= 1e-6 * max(abs(want), 1.0)
+        assert got[2] == direct["count_all"][i].as_py()
 
 
 def test_partial_reduce_min_max():
