@@ -125,6 +125,11 @@ struct dd_partitioner {
     bool staged = false; /* v2 path: block-tile + LDS-staged scatter (fixed-width only) */
     int gmax = 0;        /* v2 groups per wave per round */
     int wpb = 4;         /* v2 waves per block; rows per round R = gmax * wpb * 64 */
+    bool pre = false;    /* K3-P: precomputed round layout (k_scatter_pre) */
+    int64_t nrounds = 0, nseg_pad = 0; /* pre: globally-aligned rounds / padded segments */
+    int rpb = 1;                       /* pre: rounds per block */
+    int pre_nranges = 0;               /* pre: scan ranges (segment-granular counts) */
+    uint32_t sP2 = 0;                  /* pre: imgb row stride (u16, even) */
 
     /* device buffers (owned) */
     uint32_t *pid = nullptr;
@@ -134,6 +139,7 @@ struct dd_partitioner {
     uint32_t *bcounts = nullptr;         /* v1: [nvar][nchunks][P] */
     uint32_t *bpartials = nullptr;       /* v1: [nvar][RANGES][P] */
     uint64_t *part_boffsets = nullptr;   /* [nvar][P+1] */
+    uint16_t *imgb = nullptr;            /* pre: [nseg][sP2] LDS-image bases */
     uint32_t *src_row = nullptr;         /* staged-var: permutation out[slot] = input row */
     uint64_t *out_off[DD_KMAX_VAR] = {}; /* staged-var: Arrow byte offsets [n+1] per var */
     uint64_t *k4_partials = nullptr;     /* staged-var scan scratch */
@@ -147,6 +153,7 @@ struct dd_partitioner {
 
     ~dd_partitioner() {
         (void)hipFree(pid);
+        (void)hipFree(imgb);
         (void)hipFree(counts);
         (void)hipFree(partials);
         (void)hipFree(part_offsets);
@@ -314,6 +321,72 @@ extern "C" dd_status dd_partitioner_create(const dd_batch_desc *batch, const int
         p->chunk_rows = (n + nchunks - 1) / nchunks;
         if (p->chunk_rows < 1) p->chunk_rows = 1;
     }
+    /* K3-P (precomputed round layout, round 2 — DESIGN.md §13): replaces the in-kernel
+     * cross-wave scan / round_off scan / serial dstbase chain of the staged scatter with
+     * segment-granular precomputed bases (k_hash_count_seg + global-base scan fold +
+     * k_round_layout + k_scatter_pre). Default ON for its instantiated shapes;
+     * DD_K3_PRE=0 reverts to the HL/spec path for A/B. Mutually exclusive with rhash
+     * (pre reads the pid array). Shape whitelist mirrors dd_launch_scatter_pre. */
+    if (p->staged && p->wpb == 16 && nvar == 0 &&
+        !(getenv("DD_K3_PRE") && atoi(getenv("DD_K3_PRE")) == 0) &&
+        !(getenv("DD_RHASH") && atoi(getenv("DD_RHASH")) == 1)) {
+        bool fixed_ok = true;
+        int rowb = 0;
+        for (int c = 0; c < batch->n_cols && fixed_ok; c++) {
+            const int e = fixed_elem_size(batch->cols[c].dtype);
+            if (batch->cols[c].validity || e == 0) fixed_ok = false;
+            rowb += e;
+        }
+        auto pre_is = [&](std::initializer_list<int> want) {
+            if ((int)want.size() != batch->n_cols) return false;
+            int c = 0;
+            for (int w : want)
+                if (fixed_elem_size(batch->cols[c++].dtype) != w) return false;
+            return true;
+        };
+        int pgmax = 0;
+        if (fixed_ok) {
+            const uint32_t Pn = n_partitions;
+            bool all48 = batch->n_cols == 4;
+            for (int c = 0; c < batch->n_cols && all48; c++) {
+                const int e = fixed_elem_size(batch->cols[c].dtype);
+                if (e != 4 && e != 8) all48 = false;
+            }
+            if (Pn <= 128 && (all48 || pre_is({8, 8, 8, 4, 4})))
+                pgmax = 4;
+            else if (Pn <= 128 && pre_is({1, 1, 8, 8, 8, 8, 4}))
+                pgmax = 2;
+            else if (Pn <= 256 && pre_is({8, 8, 8, 4}))
+                pgmax = 4;
+            else if (Pn <= 512 && pre_is({8, 8, 8, 4}))
+                pgmax = 2;
+        }
+        if (pgmax > 0) {
+            const int wpb = 16;
+            const int64_t R = (int64_t)pgmax * wpb * 64;
+            const uint32_t sP2 = (n_partitions + 1) & ~1u;
+            const size_t lds = (size_t)R * rowb + 4 * (size_t)R +
+                               (size_t)wpb * (4 * (size_t)n_partitions + 2 * (size_t)sP2 +
+                                              2 * (size_t)n_partitions);
+            if (lds <= 163840) {
+                p->pre = true;
+                p->gmax = pgmax;
+                p->wpb = wpb;
+                p->lds_k3 = lds;
+                p->sP2 = sP2;
+                p->nrounds = (n + R - 1) / R;
+                if (p->nrounds < 1) p->nrounds = 1;
+                const int64_t nseg = p->nrounds * wpb;
+                p->nseg_pad = (nseg + 3) & ~3LL;
+                p->rpb = (int)((p->nrounds + 2047) / 2048);
+                if (p->rpb < 1) p->rpb = 1;
+                p->pre_nranges = 2048;
+                p->nchunks = p->nseg_pad;       /* counts rows (scan granularity) */
+                p->chunk_rows = (int64_t)pgmax * 64; /* = SEG (K1seg rows per wave) */
+                p->lds_k1 = (size_t)WAVES_PER_BLOCK_H * n_partitions * 4;
+            }
+        }
+    }
     const int64_t nchunks = p->nchunks;
 
     /* rhash (opt-in, DD_RHASH=1): spec-path batches (wpb 16, all fixed, no validity)
@@ -326,7 +399,7 @@ extern "C" dd_status dd_partitioner_create(const dd_batch_desc *batch, const int
      * stall. Kept parity-tested for re-evaluation on future silicon.
      * Must mirror the launcher's can_spec/can_spec8 conditions exactly
      * (dd_launch_scatter_staged fails loudly if not). */
-    if (getenv("DD_RHASH") && atoi(getenv("DD_RHASH")) == 1 && p->staged &&
+    if (getenv("DD_RHASH") && atoi(getenv("DD_RHASH")) == 1 && p->staged && !p->pre &&
         p->wpb == 16 && nvar == 0 && batch->n_cols <= DD_STAGE_MAXC) {
         bool rhash = true;
         for (int c = 0; c < batch->n_cols && rhash; c++)
@@ -347,7 +420,7 @@ extern "C" dd_status dd_partitioner_create(const dd_batch_desc *batch, const int
      * Gate must match the instantiated shapes exactly (4 fixed cols, elems 4/8, no
      * validity, gmax 4, wpb 16); mutually exclusive with rhash (HL reads the pid
      * array). */
-    if (!ka.rhash && !(getenv("DD_K3_HL") && atoi(getenv("DD_K3_HL")) == 0) &&
+    if (!ka.rhash && !p->pre && !(getenv("DD_K3_HL") && atoi(getenv("DD_K3_HL")) == 0) &&
         p->staged && p->wpb == 16 && p->gmax == 4 && nvar == 0 && batch->n_cols == 4) {
         bool okhl = true;
         for (int c = 0; c < 4; c++) {
@@ -358,7 +431,8 @@ extern "C" dd_status dd_partitioner_create(const dd_batch_desc *batch, const int
     }
     /* generalized HL (k_scatter_hlg): exact whitelisted shapes only — must mirror the
      * launcher's DD_HLG table */
-    if (!ka.rhash && !ka.hl && !(getenv("DD_K3_HL") && atoi(getenv("DD_K3_HL")) == 0) &&
+    if (!ka.rhash && !ka.hl && !p->pre &&
+        !(getenv("DD_K3_HL") && atoi(getenv("DD_K3_HL")) == 0) &&
         p->staged && p->wpb == 16) {
         bool novalid = true;
         for (int c = 0; c < batch->n_cols && novalid; c++)
@@ -381,8 +455,11 @@ extern "C" dd_status dd_partitioner_create(const dd_batch_desc *batch, const int
     };
     bool ok = (ka.rhash ? true : halloc((void **)&p->pid, (size_t)n * 4)) &&
               halloc((void **)&p->counts, (size_t)nchunks * P * 4) &&
-              halloc((void **)&p->partials, (size_t)DD_SCAN_RANGES * P * 4) &&
+              halloc((void **)&p->partials,
+                     (size_t)(p->pre ? p->pre_nranges : DD_SCAN_RANGES) * P * 4) &&
               halloc((void **)&p->part_offsets, (size_t)(P + 1) * 8);
+    if (ok && p->pre)
+        ok = halloc((void **)&p->imgb, (size_t)p->nseg_pad * p->sP2 * 2);
     if (ok && nvar > 0) {
         ok = halloc((void **)&p->part_boffsets, (size_t)nvar * (P + 1) * 8);
         if (p->staged) {
@@ -479,7 +556,10 @@ extern "C" dd_status dd_partitioner_run_phase1(dd_partitioner *p, void *stream) 
     if (!p) return set_err(DD_ERR_INVALID, "null partitioner");
     hipStream_t s = (hipStream_t)stream;
     HIP_TRY(hipEventRecord(p->ev[0], s));
-    if (p->staged) {
+    if (p->pre) {
+        HIP_TRY(dd_launch_hash_count_seg(&p->ka, p->nseg_pad, p->chunk_rows, p->nparts,
+                                         p->nbits, p->pid, p->counts, p->lds_k1, s));
+    } else if (p->staged) {
         HIP_TRY(dd_launch_hash_count_tile(&p->ka, p->nchunks, p->chunk_rows, p->nparts,
                                           p->nbits, p->pid, p->counts, p->lds_k1, s));
     } else {
@@ -487,14 +567,19 @@ extern "C" dd_status dd_partitioner_run_phase1(dd_partitioner *p, void *stream) 
                                      p->pid, p->counts, p->bcounts, p->lds_k1, s));
     }
     HIP_TRY(hipEventRecord(p->ev[1], s));
-    HIP_TRY(dd_launch_scan(p->counts, p->nchunks, p->nparts, DD_SCAN_RANGES, p->partials,
-                           p->part_offsets, s));
+    HIP_TRY(dd_launch_scan(p->counts, p->nchunks, p->nparts,
+                           p->pre ? p->pre_nranges : DD_SCAN_RANGES, p->partials,
+                           p->part_offsets, p->pre ? 1 : 0, s));
+    if (p->pre) {
+        HIP_TRY(dd_launch_round_layout(p->counts, p->part_offsets, p->nrounds, p->wpb,
+                                       p->nparts, p->sP2, p->imgb, s));
+    }
     if (!p->staged) {
         for (int v = 0; v < p->ka.n_var; v++) {
             HIP_TRY(dd_launch_scan(p->bcounts + (size_t)v * p->nchunks * p->nparts,
                                    p->nchunks, p->nparts, DD_SCAN_RANGES,
                                    p->bpartials + (size_t)v * DD_SCAN_RANGES * p->nparts,
-                                   p->part_boffsets + (size_t)v * (p->nparts + 1), s));
+                                   p->part_boffsets + (size_t)v * (p->nparts + 1), 0, s));
         }
     }
     HIP_TRY(hipEventRecord(p->ev[2], s));
@@ -505,7 +590,12 @@ extern "C" dd_status dd_partitioner_run_phase2(dd_partitioner *p, void *stream) 
     if (!p) return set_err(DD_ERR_INVALID, "null partitioner");
     hipStream_t s = (hipStream_t)stream;
     HIP_TRY(hipEventRecord(p->ev[4], s)); /* K3 start on the phase2 stream */
-    if (p->staged) {
+    if (p->pre) {
+        const int64_t nblocks = (p->nrounds + p->rpb - 1) / p->rpb;
+        HIP_TRY(dd_launch_scatter_pre(&p->ka, nblocks, p->nrounds, p->rpb, p->nparts,
+                                      p->nbits, p->pid, p->counts, p->imgb, p->sP2,
+                                      p->gmax, p->lds_k3, s));
+    } else if (p->staged) {
         HIP_TRY(dd_launch_scatter_staged(&p->ka, p->nchunks, p->chunk_rows, p->nparts,
                                          p->nbits, p->pid, p->counts, p->part_offsets,
                                          p->gmax, p->wpb, p->lds_k3, s));

@@ -70,7 +70,19 @@ hipError_t dd_launch_hash_count(const dd_kargs *a, int64_t nchunks, int64_t chun
                                 uint32_t *counts, uint32_t *bcounts, size_t lds_bytes,
                                 hipStream_t s);
 hipError_t dd_launch_scan(uint32_t *counts, int64_t nchunks, uint32_t nparts, int nranges,
-                          uint32_t *partials, uint64_t *part_offsets, hipStream_t s);
+                          uint32_t *partials, uint64_t *part_offsets, int fold_global,
+                          hipStream_t s);
+hipError_t dd_launch_hash_count_seg(const dd_kargs *a, int64_t nseg, int64_t seg_rows,
+                                    uint32_t nparts, int nbits, uint32_t *pid_out,
+                                    uint32_t *counts, size_t lds_bytes, hipStream_t s);
+hipError_t dd_launch_round_layout(const uint32_t *counts, const uint64_t *part_offsets,
+                                  int64_t nrounds, int wpb, uint32_t nparts, uint32_t sP2,
+                                  uint16_t *imgb, hipStream_t s);
+hipError_t dd_launch_scatter_pre(const dd_kargs *a, int64_t nblocks, int64_t nrounds,
+                                 int rpb, uint32_t nparts, int nbits,
+                                 const uint32_t *pid_in, const uint32_t *gbase,
+                                 const uint16_t *imgb, uint32_t sP2, int gmax,
+                                 size_t lds_bytes, hipStream_t s);
 hipError_t dd_launch_hash_count_tile(const dd_kargs *a, int64_t nblocks, int64_t tile_rows,
                                      uint32_t nparts, int nbits, uint32_t *pid_out,
                                      uint32_t *counts, size_t lds_bytes, hipStream_t s);

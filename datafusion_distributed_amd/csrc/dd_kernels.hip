@@ -149,7 +149,9 @@ __global__ void k_scan_combine(uint32_t *partials, int nranges, uint32_t nparts,
 }
 
 __global__ void k_scan_rewrite(uint32_t *counts, int64_t nchunks, uint32_t nparts,
-                               int nranges, const uint32_t *partials) {
+                               int nranges, const uint32_t *partials,
+                               const uint64_t *fold_offsets /* null, or part_offsets to
+                                                               make bases GLOBAL (pre) */) {
     const int64_t tid = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
     const int64_t total = (int64_t)nranges * nparts;
     if (tid >= total) return;
@@ -157,6 +159,7 @@ __global__ void k_scan_rewrite(uint32_t *counts, int64_t nchunks, uint32_t npart
     const uint32_t p = (uint32_t)(tid % nparts);
     const int64_t c0 = nchunks * r / nranges, c1 = nchunks * (r + 1) / nranges;
     uint32_t run = partials[(size_t)r * nparts + p];
+    if (fold_offsets) run += (uint32_t)fold_offsets[p];
     for (int64_t c = c0; c < c1; c++) {
         uint32_t v = counts[c * nparts + p];
         counts[c * nparts + p] = run;
@@ -1187,6 +1190,349 @@ __global__ __launch_bounds__(16 * WAVE) void k_scatter_hlg(
     }
 }
 
+/* ================= K3-P: precomputed-layout hidden-load scatter (round 2) ==========
+ * The HL kernel still spends ~0.4-0.5 ms/launch (of 1.1) on per-round machinery: the
+ * 16-deep serial cross-wave seghist scan, the fused partition-offset scan (3 of the 5
+ * barriers per round), and the serial dstbase accumulation chaining every round to the
+ * previous one (profiles/r01_k3_ablation.json: "residual ~0.5 ms"). This kernel removes
+ * ALL of it by precomputing the layout at WAVE-SEGMENT granularity in the K2 family:
+ *
+ *   - rounds are GLOBALLY aligned: round r = rows [r*R, (r+1)*R); segment s = r*WPB + w
+ *     = rows [r*R + w*SEG, +SEG). K1 (k_hash_count_seg) writes per-SEGMENT histograms.
+ *   - the K2 scan runs at segment granularity and REWRITES counts to global slot bases
+ *     (within-partition exclusive + part_offsets fold): gbase[s][p] = first output slot
+ *     of segment s's rows for partition p.
+ *   - k_round_layout precomputes imgb[s][p] (u16): the partition-major LDS-image base of
+ *     (segment, partition) within its round = round_off[p] + (gbase[s][p] - gbase[s0][p]).
+ *
+ * K3 then needs NO cross-wave communication outside the two image barriers: each wave
+ * hidden-loads its own gbase/imgb rows (wave-private LDS, program-ordered), ranks its
+ * rows intra-segment with a per-wave LDS counter, places at imgb[pid]+rank with global
+ * destination gbase[pid]+rank, and the flush is unchanged. 2 barriers per round instead
+ * of 5, no serial scan, no cross-round dependency, uniform full rounds everywhere except
+ * the single global tail. Hidden-load discipline identical to k_scatter_hl (issue order
+ * per round: G pid loads, L=G*NC column loads, NB base loads, L flush stores):
+ *   place waits vmcnt(L)      (prev round's stores in flight),
+ *   rank  waits vmcnt(2L+NB)  (this round's col+base loads + stores in flight).
+ * NBG/NBI are compile-time dword counts for the gbase/imgb row loads (host gates P to
+ * the instantiated tier); imgb rows are padded to even u16 (sP2) so dword loads stay
+ * aligned. Parity: bit-exact vs the oracle (tests/test_gpu_fuzz.py::test_pre_ab). */
+
+template <int G, int NBG, int NBI, int... Es>
+__global__ __launch_bounds__(16 * WAVE) void k_scatter_pre(
+    dd_kargs a, int64_t nrounds, int rpb, uint32_t nparts, int nbits,
+    const uint32_t *pid_in, const uint32_t *gbase /* [nseg][P] global slot bases */,
+    const uint16_t *imgb /* [nseg][sP2] LDS-image bases */, uint32_t sP2) {
+    constexpr int WPB = 16, GMAX = G;
+    constexpr int BT = WPB * WAVE;
+    constexpr int R = GMAX * BT;
+    constexpr int SEG = R / WPB;
+    constexpr int NC = sizeof...(Es);
+    constexpr int EL[NC] = {Es...};
+    constexpr int L = GMAX * NC;
+    constexpr int NB = NBG + NBI;
+    static_assert(2 * L + NB <= 63, "vmcnt immediate is 6 bits");
+    struct slots {
+        int idx[NC];
+        int n8, n4;
+        constexpr slots() : idx{}, n8(0), n4(0) {
+            for (int j = 0; j < NC; j++) idx[j] = (EL[j] == 8) ? n8++ : n4++;
+        }
+        constexpr int rowb() const {
+            int s = 0;
+            for (int j = 0; j < NC; j++) s += EL[j];
+            return s;
+        }
+    };
+    constexpr slots S{};
+
+    extern __shared__ __attribute__((aligned(16))) char smem[];
+    /* carve: stage | dstg u32[R] | per-wave gb u32[P] | per-wave ib u16[sP2] |
+     * per-wave ms u16[P] */
+    char *ws = smem;
+    char *const stage0 = ws;
+    ws += (size_t)R * S.rowb();
+    uint32_t *dstg = (uint32_t *)ws;
+    ws += sizeof(uint32_t) * R;
+    uint32_t *gb_all = (uint32_t *)ws;
+    ws += sizeof(uint32_t) * WPB * nparts;
+    uint16_t *ib_all = (uint16_t *)ws;
+    ws += sizeof(uint16_t) * WPB * sP2;
+    uint16_t *ms_all = (uint16_t *)ws;
+
+    const int tid = threadIdx.x;
+    const int wid = tid / WAVE;
+    const int lane = tid % WAVE;
+    uint32_t *gb = gb_all + (size_t)wid * nparts;
+    uint32_t *ib32 = (uint32_t *)(ib_all + (size_t)wid * sP2); /* dword view */
+    uint16_t *ib = ib_all + (size_t)wid * sP2;
+    uint16_t *ms = ms_all + (size_t)wid * nparts;
+    const uint64_t lt = ((uint64_t)1 << lane) - 1;
+
+    const int64_t r0 = (int64_t)blockIdx.x * rpb;
+    const int64_t r1 = (r0 + rpb < nrounds) ? r0 + rpb : nrounds;
+    if (r0 >= nrounds || a.n_rows == 0) return;
+
+    uint32_t pidr[GMAX], rankr[GMAX];
+    bool actr[GMAX];
+    uint64_t big[S.n8 > 0 ? S.n8 : 1][GMAX];
+    uint32_t sml[S.n4 > 0 ? S.n4 : 1][GMAX];
+    uint32_t baser[NB]; /* NBG gbase dwords then NBI imgb dwords */
+
+    const uint32_t ndw = (nparts + 1) / 2; /* imgb row dwords */
+
+    auto preload = [&](int64_t r) {
+        const int64_t rstart = r * R;
+        const int64_t rend = (rstart + R < a.n_rows) ? rstart + R : a.n_rows;
+        const int64_t segstart = rstart + (int64_t)wid * SEG;
+        const int64_t seg = r * WPB + wid;
+        uint32_t rowc[GMAX];
+#pragma unroll
+        for (int g = 0; g < GMAX; g++) {
+            const int64_t row = segstart + g * WAVE + lane;
+            actr[g] = row < rend;
+            rowc[g] = (uint32_t)(actr[g] ? row : rend - 1); /* always issue (counts) */
+        }
+#pragma unroll
+        for (int g = 0; g < GMAX; g++) HL_LD32(pidr[g], pid_in + rowc[g]);
+        hl_for<0, NC>([&](auto jc) {
+            constexpr int J = jc.value;
+            const void *base = a.cols[J].data;
+#pragma unroll
+            for (int g = 0; g < GMAX; g++) {
+                if constexpr (EL[J] == 8) {
+                    HL_LD64(big[S.idx[J]][g], (const uint64_t *)base + rowc[g]);
+                } else if constexpr (EL[J] == 4) {
+                    HL_LD32(sml[S.idx[J]][g], (const uint32_t *)base + rowc[g]);
+                } else if constexpr (EL[J] == 2) {
+                    asm volatile("global_load_ushort %0, %1, off"
+                                 : "=v"(sml[S.idx[J]][g])
+                                 : "v"((const uint16_t *)base + rowc[g]));
+                } else {
+                    asm volatile("global_load_ubyte %0, %1, off"
+                                 : "=v"(sml[S.idx[J]][g])
+                                 : "v"((const uint8_t *)base + rowc[g]));
+                }
+            }
+        });
+        /* base rows (wave-private): gbase u32[P] then imgb u16[sP2] as dwords */
+        const uint32_t *grow = gbase + (size_t)seg * nparts;
+#pragma unroll
+        for (int k = 0; k < NBG; k++) {
+            uint32_t idx = (uint32_t)lane + k * WAVE;
+            if (idx >= nparts) idx = nparts - 1; /* redundant re-load keeps counts exact */
+            HL_LD32(baser[k], grow + idx);
+        }
+        const uint32_t *irow = (const uint32_t *)(imgb + (size_t)seg * sP2);
+#pragma unroll
+        for (int k = 0; k < NBI; k++) {
+            uint32_t idx = (uint32_t)lane + k * WAVE;
+            if (idx >= ndw) idx = ndw - 1;
+            HL_LD32(baser[NBG + k], irow + idx);
+        }
+    };
+
+    auto wait_pid = [&](auto cnt) {
+#pragma unroll
+        for (int g = 0; g < GMAX; g++) hl_tie_wait<cnt.value>(pidr[g]);
+    };
+    auto wait_loads = [&](auto cnt) { /* columns + bases retired */
+        hl_for<0, NC>([&](auto jc) {
+            constexpr int J = jc.value;
+#pragma unroll
+            for (int g = 0; g < GMAX; g++) {
+                if constexpr (EL[J] == 8) hl_tie_wait<cnt.value>(big[S.idx[J]][g]);
+                else hl_tie_wait<cnt.value>(sml[S.idx[J]][g]);
+            }
+        });
+#pragma unroll
+        for (int k = 0; k < NB; k++) hl_tie_wait<cnt.value>(baser[k]);
+    };
+
+    auto rank = [&]() {
+        for (uint32_t p = lane; p < nparts; p += WAVE) ms[p] = 0;
+#pragma unroll
+        for (int g = 0; g < GMAX; g++) {
+            const bool active = actr[g];
+            const uint32_t pid = pidr[g];
+            uint64_t act = __ballot(active);
+            uint32_t rk = 0;
+            if (active) {
+                uint64_t eq = dd_eq_mask(pid, act, nbits);
+                int leader = __ffsll((unsigned long long)eq) - 1;
+                uint32_t base = 0;
+                if (lane == leader) {
+                    base = ms[pid];
+                    ms[pid] = (uint16_t)(base + (uint32_t)__popcll((unsigned long long)eq));
+                }
+                base = (uint32_t)__shfl((int)base, leader);
+                rk = base + (uint32_t)__popcll((unsigned long long)(eq & lt));
+            }
+            rankr[g] = rk;
+        }
+    };
+
+    /* prologue: first round's loads + rank, then a one-time drain */
+    preload(r0);
+    wait_pid(std::integral_constant<int, L + NB>{});
+    rank();
+    wait_loads(std::integral_constant<int, 0>{});
+    __syncthreads();
+
+    for (int64_t r = r0; r < r1; r++) {
+        const int64_t rstart = r * R;
+        const int round_rows =
+            (int)(((rstart + R < a.n_rows) ? rstart + R : a.n_rows) - rstart);
+
+        /* place: cols + bases retired at vmcnt(L) — only the previous round's flush
+         * stores stay in flight */
+        wait_loads(std::integral_constant<int, L>{});
+        {
+            /* base rows -> wave-private LDS (program-ordered with the reads below) */
+#pragma unroll
+            for (int k = 0; k < NBG; k++) {
+                uint32_t idx = (uint32_t)lane + k * WAVE;
+                if (idx < nparts) gb[idx] = baser[k];
+            }
+#pragma unroll
+            for (int k = 0; k < NBI; k++) {
+                uint32_t idx = (uint32_t)lane + k * WAVE;
+                if (idx < ndw) ib32[idx] = baser[NBG + k];
+            }
+        }
+#pragma unroll
+        for (int g = 0; g < GMAX; g++) {
+            if (!actr[g]) continue;
+            const uint32_t pid = pidr[g];
+            const uint32_t rk = rankr[g];
+            const uint32_t slot = (uint32_t)ib[pid] + rk;
+            dstg[slot] = gb[pid] + rk;
+            char *stage = stage0;
+            hl_for<0, NC>([&](auto jc) {
+                constexpr int J = jc.value;
+                if constexpr (EL[J] == 8)
+                    ((uint64_t *)stage)[slot] = big[S.idx[J]][g];
+                else if constexpr (EL[J] == 4)
+                    ((uint32_t *)stage)[slot] = sml[S.idx[J]][g];
+                else if constexpr (EL[J] == 2)
+                    ((uint16_t *)stage)[slot] = (uint16_t)sml[S.idx[J]][g];
+                else
+                    ((uint8_t *)stage)[slot] = (uint8_t)sml[S.idx[J]][g];
+                stage += (size_t)R * EL[J];
+            });
+        }
+        __syncthreads();
+
+        const bool more = r + 1 < r1;
+        if (more) preload(r + 1);
+
+        /* flush padded to GMAX iterations (clamped rewrite is idempotent) */
+#pragma unroll
+        for (int u = 0; u < GMAX; u++) {
+            const int i = tid + u * BT;
+            const int ic = (i < round_rows) ? i : (round_rows - 1);
+            const uint64_t dst = dstg[ic];
+            char *stage = stage0;
+            hl_for<0, NC>([&](auto jc) {
+                constexpr int J = jc.value;
+                void *out = a.cols[J].out_data;
+                if constexpr (EL[J] == 8)
+                    ((uint64_t *)out)[dst] = ((const uint64_t *)stage)[ic];
+                else if constexpr (EL[J] == 4)
+                    ((uint32_t *)out)[dst] = ((const uint32_t *)stage)[ic];
+                else if constexpr (EL[J] == 2)
+                    ((uint16_t *)out)[dst] = ((const uint16_t *)stage)[ic];
+                else
+                    ((uint8_t *)out)[dst] = ((const uint8_t *)stage)[ic];
+                stage += (size_t)R * EL[J];
+            });
+        }
+
+        if (more) {
+            wait_pid(std::integral_constant<int, 2 * L + NB>{});
+            rank();
+        }
+        __syncthreads();
+    }
+}
+
+/* K1 for the pre path: per-SEGMENT histograms (segment = one wave's SEG contiguous rows
+ * of a globally-aligned round). 4 segments per 256-thread block; per-wave LDS hist. */
+__global__ __launch_bounds__(BLOCK_THREADS) void k_hash_count_seg(
+    dd_kargs a, int64_t seg_rows, uint32_t nparts, int nbits, uint32_t *pid_out,
+    uint32_t *counts /* [nseg][P] */) {
+    extern __shared__ __attribute__((aligned(16))) char smem[];
+    const int wid = threadIdx.x / WAVE;
+    const int lane = threadIdx.x % WAVE;
+    const int64_t seg = (int64_t)blockIdx.x * WAVES_PER_BLOCK + wid;
+    uint32_t *hist = (uint32_t *)smem + (size_t)wid * nparts;
+
+    for (uint32_t p = lane; p < nparts; p += WAVE) hist[p] = 0;
+    /* single wave: LDS program order; no barrier */
+
+    const int64_t start = seg * seg_rows;
+    const int64_t end = (start + seg_rows < a.n_rows) ? start + seg_rows : a.n_rows;
+
+    for (int64_t base = start; base < end; base += 2 * WAVE) {
+        uint32_t pidu[2];
+        bool actu[2];
+#pragma unroll
+        for (int u = 0; u < 2; u++) {
+            const int64_t row = base + u * WAVE + lane;
+            actu[u] = row < end;
+            pidu[u] = 0;
+            if (actu[u]) {
+                uint64_t h = dd_row_hash(a, row);
+                const uint32_t tot = a.pid_total;
+                uint32_t fine = ((tot & (tot - 1)) == 0)
+                                    ? (uint32_t)(h & (uint64_t)(tot - 1))
+                                    : (uint32_t)(h % (uint64_t)tot);
+                pidu[u] = fine >> a.pid_shift;
+                pid_out[row] = pidu[u];
+            }
+        }
+#pragma unroll
+        for (int u = 0; u < 2; u++) {
+            uint64_t act = __ballot(actu[u]);
+            if (actu[u]) {
+                uint64_t eq = dd_eq_mask(pidu[u], act, nbits);
+                int leader = __ffsll((unsigned long long)eq) - 1;
+                if (lane == leader) hist[pidu[u]] += (uint32_t)__popcll((unsigned long long)eq);
+            }
+        }
+    }
+    for (uint32_t p = lane; p < nparts; p += WAVE)
+        counts[(size_t)seg * nparts + p] = hist[p];
+}
+
+/* K2d for the pre path: per-round LDS-image layout. counts must already be rewritten to
+ * GLOBAL slot bases (k_scan_rewrite with the part_offsets fold). One block per round:
+ *   roundcnt[p] = gbase[seg0_{r+1}][p] - gbase[seg0_r][p]   (tail: part_offsets[p+1])
+ *   round_off   = exclusive scan over p of roundcnt
+ *   imgb[s][p]  = round_off[p] + (gbase[s][p] - gbase[seg0_r][p])   (u16, <= R) */
+__global__ __launch_bounds__(256) void k_round_layout(
+    const uint32_t *counts, const uint64_t *part_offsets, int64_t nrounds, int wpb,
+    uint32_t nparts, uint32_t sP2, uint16_t *imgb) {
+    __shared__ uint32_t cnt[DD_MAX_P];
+    __shared__ uint32_t roff[DD_MAX_P];
+    __shared__ uint32_t tmp[256];
+    const int tid = threadIdx.x;
+    const int64_t r = blockIdx.x;
+    if (r >= nrounds) return;
+    const uint32_t *rb = counts + (size_t)r * wpb * nparts;
+    const uint32_t *re = (r + 1 < nrounds) ? counts + (size_t)(r + 1) * wpb * nparts : nullptr;
+    for (uint32_t p = tid; p < nparts; p += 256)
+        cnt[p] = (re ? re[p] : (uint32_t)part_offsets[p + 1]) - rb[p];
+    __syncthreads();
+    dd_block_excl_scan<256>(cnt, roff, nparts, tmp);
+    for (int w = 0; w < wpb; w++) {
+        const uint32_t *sr = counts + ((size_t)r * wpb + w) * nparts;
+        uint16_t *orow = imgb + ((size_t)r * wpb + w) * sP2;
+        for (uint32_t p = tid; p < nparts; p += 256)
+            orow[p] = (uint16_t)(roff[p] + (sr[p] - rb[p]));
+    }
+}
+
 /* ================= K4: var-width bytes for the staged path =================
  * The staged scatter (v2) handles var columns' LENGTHS and a ROWID permutation as
  * synthetic fixed u32 columns (DD_KDT_VARLEN / DD_KDT_ROWID, set up by dd_host.cpp).
@@ -1445,7 +1791,8 @@ hipError_t dd_launch_hash_count(const dd_kargs *a, int64_t nchunks, int64_t chun
 }
 
 hipError_t dd_launch_scan(uint32_t *counts, int64_t nchunks, uint32_t nparts, int nranges,
-                          uint32_t *partials, uint64_t *part_offsets, hipStream_t s) {
+                          uint32_t *partials, uint64_t *part_offsets, int fold_global,
+                          hipStream_t s) {
     int threads = 256;
     int64_t total = (int64_t)nranges * nparts;
     int blocks = (int)((total + threads - 1) / threads);
@@ -1454,8 +1801,87 @@ hipError_t dd_launch_scan(uint32_t *counts, int64_t nchunks, uint32_t nparts, in
     hipLaunchKernelGGL(k_scan_combine, dim3(1), dim3(256), 0, s, partials, nranges, nparts,
                        part_offsets);
     hipLaunchKernelGGL(k_scan_rewrite, dim3(blocks), dim3(threads), 0, s, counts, nchunks,
-                       nparts, nranges, partials);
+                       nparts, nranges, partials,
+                       fold_global ? (const uint64_t *)part_offsets : nullptr);
     return hipGetLastError();
+}
+
+hipError_t dd_launch_hash_count_seg(const dd_kargs *a, int64_t nseg, int64_t seg_rows,
+                                    uint32_t nparts, int nbits, uint32_t *pid_out,
+                                    uint32_t *counts, size_t lds_bytes, hipStream_t s) {
+    dim3 grid((unsigned)(nseg / WAVES_PER_BLOCK));
+    if (lds_bytes > 65536) {
+        hipError_t e = hipFuncSetAttribute((const void *)k_hash_count_seg,
+                                           hipFuncAttributeMaxDynamicSharedMemorySize,
+                                           (int)lds_bytes);
+        if (e != hipSuccess) return e;
+    }
+    hipLaunchKernelGGL(k_hash_count_seg, grid, dim3(BLOCK_THREADS), lds_bytes, s, *a,
+                       seg_rows, nparts, nbits, pid_out, counts);
+    return hipGetLastError();
+}
+
+hipError_t dd_launch_round_layout(const uint32_t *counts, const uint64_t *part_offsets,
+                                  int64_t nrounds, int wpb, uint32_t nparts, uint32_t sP2,
+                                  uint16_t *imgb, hipStream_t s) {
+    hipLaunchKernelGGL(k_round_layout, dim3((unsigned)nrounds), dim3(256), 0, s, counts,
+                       part_offsets, nrounds, wpb, nparts, sP2, imgb);
+    return hipGetLastError();
+}
+
+hipError_t dd_launch_scatter_pre(const dd_kargs *a, int64_t nblocks, int64_t nrounds,
+                                 int rpb, uint32_t nparts, int nbits,
+                                 const uint32_t *pid_in, const uint32_t *gbase,
+                                 const uint16_t *imgb, uint32_t sP2, int gmax,
+                                 size_t lds_bytes, hipStream_t s) {
+    dim3 grid((unsigned)nblocks);
+    const int n = a->n_cols;
+    auto el = [&](int c) { return (int)a->cols[c].elem; };
+#define DD_PRE(GM, NBG_, NBI_, ...)                                                          \
+    {                                                                                        \
+        const int want[] = {__VA_ARGS__};                                                    \
+        const int wn = (int)(sizeof(want) / sizeof(want[0]));                                \
+        bool m = (gmax == GM && n == wn && nparts <= (uint32_t)(NBG_ * WAVE));               \
+        for (int c = 0; c < wn && m; c++) m = el(c) == want[c];                              \
+        if (m) {                                                                             \
+            if (lds_bytes > 65536) {                                                         \
+                hipError_t e = hipFuncSetAttribute(                                          \
+                    (const void *)k_scatter_pre<GM, NBG_, NBI_, __VA_ARGS__>,                \
+                    hipFuncAttributeMaxDynamicSharedMemorySize, (int)lds_bytes);             \
+                if (e != hipSuccess) return e;                                               \
+            }                                                                                \
+            hipLaunchKernelGGL((k_scatter_pre<GM, NBG_, NBI_, __VA_ARGS__>), grid,           \
+                               dim3(16 * WAVE), lds_bytes, s, *a, nrounds, rpb, nparts,      \
+                               nbits, pid_in, gbase, imgb, sP2);                             \
+            return hipGetLastError();                                                        \
+        }                                                                                    \
+    }
+    /* P <= 128 tier (NBG=2, NBI=1): all {4,8}^4 combos + the multikey and q1 shapes */
+    DD_PRE(4, 2, 1, 4, 4, 4, 4)
+    DD_PRE(4, 2, 1, 4, 4, 4, 8)
+    DD_PRE(4, 2, 1, 4, 4, 8, 4)
+    DD_PRE(4, 2, 1, 4, 4, 8, 8)
+    DD_PRE(4, 2, 1, 4, 8, 4, 4)
+    DD_PRE(4, 2, 1, 4, 8, 4, 8)
+    DD_PRE(4, 2, 1, 4, 8, 8, 4)
+    DD_PRE(4, 2, 1, 4, 8, 8, 8)
+    DD_PRE(4, 2, 1, 8, 4, 4, 4)
+    DD_PRE(4, 2, 1, 8, 4, 4, 8)
+    DD_PRE(4, 2, 1, 8, 4, 8, 4)
+    DD_PRE(4, 2, 1, 8, 4, 8, 8)
+    DD_PRE(4, 2, 1, 8, 8, 4, 4)
+    DD_PRE(4, 2, 1, 8, 8, 4, 8)
+    DD_PRE(4, 2, 1, 8, 8, 8, 4)
+    DD_PRE(4, 2, 1, 8, 8, 8, 8)
+    DD_PRE(4, 2, 1, 8, 8, 8, 4, 4)       /* multikey bench shape */
+    DD_PRE(2, 2, 1, 1, 1, 8, 8, 8, 8, 4) /* q1 shape */
+    /* 129..256 tier (NBG=4, NBI=2): headline + multikey */
+    DD_PRE(4, 4, 2, 8, 8, 8, 4)
+    DD_PRE(4, 4, 2, 8, 8, 8, 4, 4)
+    /* 257..512 tier (NBG=8, NBI=4, G=2 for LDS): headline shape */
+    DD_PRE(2, 8, 4, 8, 8, 8, 4)
+#undef DD_PRE
+    return hipErrorInvalidValue;
 }
 
 hipError_t dd_launch_hash_count_tile(const dd_kargs *a, int64_t nblocks, int64_t tile_rows,

@@ -109,11 +109,13 @@ def test_hl_ab(monkeypatch):
         cols = [random_col(rng, n, "i64", 0), random_col(rng, n, "f64", 0),
                 random_col(rng, n, "f64", 0), random_col(rng, n, "i32", 0)]
         for keys, nparts in [([0], 128), ([3, 0], 100)]:
-            monkeypatch.delenv("DD_K3_HL", raising=False)  # default: HL
+            monkeypatch.setenv("DD_K3_PRE", "0")  # force the HL tier (pre outranks it)
+            monkeypatch.delenv("DD_K3_HL", raising=False)  # default-within-tier: HL
             check_against_oracle(cols, keys, nparts)
             monkeypatch.setenv("DD_K3_HL", "0")  # plain spec path
             check_against_oracle(cols, keys, nparts)
     n = 123_457
+    monkeypatch.setenv("DD_K3_PRE", "0")
     cols = [random_col(rng, n, "i32", 0), random_col(rng, n, "i64", 0),
             random_col(rng, n, "f32", 0), random_col(rng, n, "f64", 0)]
     monkeypatch.delenv("DD_K3_HL", raising=False)
@@ -135,7 +137,43 @@ def test_hlg_ab(monkeypatch):
               random_col(rng, n, "i32", 0)]
         for cols, keys, nparts in [(mk, [0, 4], 128), (mk, [0], 100),
                                    (q1, [0, 1], 128), (q1, [0, 1], 7)]:
-            monkeypatch.delenv("DD_K3_HL", raising=False)  # default: hlg where gated
+            monkeypatch.setenv("DD_K3_PRE", "0")  # force the hlg tier (pre outranks it)
+            monkeypatch.delenv("DD_K3_HL", raising=False)  # default-within-tier: hlg
             check_against_oracle(cols, keys, nparts)
             monkeypatch.setenv("DD_K3_HL", "0")
             check_against_oracle(cols, keys, nparts)
+
+
+def test_pre_ab(monkeypatch):
+    '''K3-P precomputed-layout scatter (k_scatter_pre, the round-2 default for its
+    instantiated shapes; dd_kernels.hip K3-P header) must match the oracle bit-exactly
+    across every NBG tier (P<=128 / <=256 / <=512), ragged and tiny inputs, non-pow2 P,
+    and the multikey/q1 shapes; the same batches through DD_K3_PRE=0 (HL/spec tier)
+    must agree too.'''
+    rng = np.random.default_rng(20_24)
+    for n in [4096 * 3, 100_000, 250_001, 65, 1]:  # round-multiple, ragged, tiny
+        cols = [random_col(rng, n, "i64", 0), random_col(rng, n, "f64", 0),
+                random_col(rng, n, "f64", 0), random_col(rng, n, "i32", 0)]
+        for keys, nparts in [([0], 128), ([0], 100), ([0], 1), ([3, 0], 256),
+                             ([0], 200), ([0], 512), ([0], 300)]:
+            monkeypatch.delenv("DD_K3_PRE", raising=False)  # default: pre
+            check_against_oracle(cols, keys, nparts)
+            monkeypatch.setenv("DD_K3_PRE", "0")
+            check_against_oracle(cols, keys, nparts)
+    # mixed 4/8 shapes (tier 1) + multikey + q1 shapes through the pre path
+    n = 123_457
+    monkeypatch.delenv("DD_K3_PRE", raising=False)
+    mixed = [random_col(rng, n, "i32", 0), random_col(rng, n, "i64", 0),
+             random_col(rng, n, "f32", 0), random_col(rng, n, "f64", 0)]
+    check_against_oracle(mixed, [0, 1], 128)
+    check_against_oracle(mixed, [2, 3], 97)
+    mk = [random_col(rng, n, "i64", 0), random_col(rng, n, "f64", 0),
+          random_col(rng, n, "f64", 0), random_col(rng, n, "i32", 0),
+          random_col(rng, n, "i32", 0)]
+    check_against_oracle(mk, [0, 4], 128)
+    q1 = [random_col(rng, n, "u8", 0), random_col(rng, n, "bool", 0),
+          random_col(rng, n, "f64", 0), random_col(rng, n, "f64", 0),
+          random_col(rng, n, "f64", 0), random_col(rng, n, "f64", 0),
+          random_col(rng, n, "i32", 0)]
+    check_against_oracle(q1, [0, 1], 128)
+    check_against_oracle(q1, [0, 1], 7)
