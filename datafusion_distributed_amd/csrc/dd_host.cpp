@@ -140,6 +140,7 @@ struct dd_partitioner {
     uint32_t *bpartials = nullptr;       /* v1: [nvar][RANGES][P] */
     uint64_t *part_boffsets = nullptr;   /* [nvar][P+1] */
     uint16_t *imgb = nullptr;            /* pre: [nseg][sP2] LDS-image bases */
+    uint32_t *partials2 = nullptr;       /* pre: second-level scan partials [64][P] */
     uint32_t *src_row = nullptr;         /* staged-var: permutation out[slot] = input row */
     uint64_t *out_off[DD_KMAX_VAR] = {}; /* staged-var: Arrow byte offsets [n+1] per var */
     uint64_t *k4_partials = nullptr;     /* staged-var scan scratch */
@@ -154,6 +155,7 @@ struct dd_partitioner {
     ~dd_partitioner() {
         (void)hipFree(pid);
         (void)hipFree(imgb);
+        (void)hipFree(partials2);
         (void)hipFree(counts);
         (void)hipFree(partials);
         (void)hipFree(part_offsets);
@@ -362,6 +364,10 @@ extern "C" dd_status dd_partitioner_create(const dd_batch_desc *batch, const int
                 pgmax = 2;
         }
         if (pgmax > 0) {
+            if (const char *e = getenv("DD_PRE_GMAX")) { /* experiment knob */
+                int v = atoi(e);
+                if (v == 2 || v == 4) pgmax = v;
+            }
             const int wpb = 16;
             const int64_t R = (int64_t)pgmax * wpb * 64;
             const uint32_t sP2 = (n_partitions + 1) & ~1u;
@@ -380,6 +386,10 @@ extern "C" dd_status dd_partitioner_create(const dd_batch_desc *batch, const int
                 p->nseg_pad = (nseg + 3) & ~3LL;
                 p->rpb = (int)((p->nrounds + 2047) / 2048);
                 if (p->rpb < 1) p->rpb = 1;
+                if (const char *e = getenv("DD_PRE_RPB")) { /* experiment knob */
+                    int v = atoi(e);
+                    if (v >= 1 && v <= 256) p->rpb = v;
+                }
                 p->pre_nranges = 2048;
                 p->nchunks = p->nseg_pad;       /* counts rows (scan granularity) */
                 p->chunk_rows = (int64_t)pgmax * 64; /* = SEG (K1seg rows per wave) */
@@ -459,7 +469,8 @@ extern "C" dd_status dd_partitioner_create(const dd_batch_desc *batch, const int
                      (size_t)(p->pre ? p->pre_nranges : DD_SCAN_RANGES) * P * 4) &&
               halloc((void **)&p->part_offsets, (size_t)(P + 1) * 8);
     if (ok && p->pre)
-        ok = halloc((void **)&p->imgb, (size_t)p->nseg_pad * p->sP2 * 2);
+        ok = halloc((void **)&p->imgb, (size_t)p->nseg_pad * p->sP2 * 2) &&
+             halloc((void **)&p->partials2, (size_t)64 * P * 4);
     if (ok && nvar > 0) {
         ok = halloc((void **)&p->part_boffsets, (size_t)nvar * (P + 1) * 8);
         if (p->staged) {
@@ -567,9 +578,13 @@ extern "C" dd_status dd_partitioner_run_phase1(dd_partitioner *p, void *stream) 
                                      p->pid, p->counts, p->bcounts, p->lds_k1, s));
     }
     HIP_TRY(hipEventRecord(p->ev[1], s));
-    HIP_TRY(dd_launch_scan(p->counts, p->nchunks, p->nparts,
-                           p->pre ? p->pre_nranges : DD_SCAN_RANGES, p->partials,
-                           p->part_offsets, p->pre ? 1 : 0, s));
+    if (p->pre) {
+        HIP_TRY(dd_launch_scan_deep(p->counts, p->nchunks, p->nparts, p->pre_nranges, 64,
+                                    p->partials, p->partials2, p->part_offsets, 1, s));
+    } else {
+        HIP_TRY(dd_launch_scan(p->counts, p->nchunks, p->nparts, DD_SCAN_RANGES,
+                               p->partials, p->part_offsets, 0, s));
+    }
     if (p->pre) {
         HIP_TRY(dd_launch_round_layout(p->counts, p->part_offsets, p->nrounds, p->wpb,
                                        p->nparts, p->sP2, p->imgb, s));

@@ -72,6 +72,9 @@ hipError_t dd_launch_hash_count(const dd_kargs *a, int64_t nchunks, int64_t chun
 hipError_t dd_launch_scan(uint32_t *counts, int64_t nchunks, uint32_t nparts, int nranges,
                           uint32_t *partials, uint64_t *part_offsets, int fold_global,
                           hipStream_t s);
+hipError_t dd_launch_scan_deep(uint32_t *counts, int64_t nchunks, uint32_t nparts,
+                               int nr1, int nr2, uint32_t *partials, uint32_t *partials2,
+                               uint64_t *part_offsets, int fold_global, hipStream_t s);
 hipError_t dd_launch_hash_count_seg(const dd_kargs *a, int64_t nseg, int64_t seg_rows,
                                     uint32_t nparts, int nbits, uint32_t *pid_out,
                                     uint32_t *counts, size_t lds_bytes, hipStream_t s);

@@ -1806,6 +1806,29 @@ hipError_t dd_launch_scan(uint32_t *counts, int64_t nchunks, uint32_t nparts, in
     return hipGetLastError();
 }
 
+/* two-level scan for segment-granular counts (pre path): k_scan_combine walking 2048
+ * ranges serially in one block measured 0.6 ms (latency-bound); recurse instead:
+ * counts -> partials[nr1] -> partials2[nr2] -> combine(nr2) -> rewrite back up. */
+hipError_t dd_launch_scan_deep(uint32_t *counts, int64_t nchunks, uint32_t nparts,
+                               int nr1, int nr2, uint32_t *partials, uint32_t *partials2,
+                               uint64_t *part_offsets, int fold_global, hipStream_t s) {
+    int threads = 256;
+    int b1 = (int)(((int64_t)nr1 * nparts + threads - 1) / threads);
+    int b2 = (int)(((int64_t)nr2 * nparts + threads - 1) / threads);
+    hipLaunchKernelGGL(k_scan_partial, dim3(b1), dim3(threads), 0, s, counts, nchunks,
+                       nparts, nr1, partials);
+    hipLaunchKernelGGL(k_scan_partial, dim3(b2), dim3(threads), 0, s, partials, (int64_t)nr1,
+                       nparts, nr2, partials2);
+    hipLaunchKernelGGL(k_scan_combine, dim3(1), dim3(256), 0, s, partials2, nr2, nparts,
+                       part_offsets);
+    hipLaunchKernelGGL(k_scan_rewrite, dim3(b2), dim3(threads), 0, s, partials, (int64_t)nr1,
+                       nparts, nr2, partials2, nullptr);
+    hipLaunchKernelGGL(k_scan_rewrite, dim3(b1), dim3(threads), 0, s, counts, nchunks,
+                       nparts, nr1, partials,
+                       fold_global ? (const uint64_t *)part_offsets : nullptr);
+    return hipGetLastError();
+}
+
 hipError_t dd_launch_hash_count_seg(const dd_kargs *a, int64_t nseg, int64_t seg_rows,
                                     uint32_t nparts, int nbits, uint32_t *pid_out,
                                     uint32_t *counts, size_t lds_bytes, hipStream_t s) {
@@ -1874,6 +1897,10 @@ hipError_t dd_launch_scatter_pre(const dd_kargs *a, int64_t nblocks, int64_t nro
     DD_PRE(4, 2, 1, 8, 8, 8, 4)
     DD_PRE(4, 2, 1, 8, 8, 8, 8)
     DD_PRE(4, 2, 1, 8, 8, 8, 4, 4)       /* multikey bench shape */
+    DD_PRE(2, 2, 1, 8, 8, 8, 4)          /* G2 experiment (2 blocks/CU) */
+    DD_PRE(2, 2, 1, 8, 8, 8, 8)
+    DD_PRE(2, 2, 1, 4, 4, 4, 4)
+    DD_PRE(2, 2, 1, 8, 8, 8, 4, 4)
     DD_PRE(2, 2, 1, 1, 1, 8, 8, 8, 8, 4) /* q1 shape */
     /* 129..256 tier (NBG=4, NBI=2): headline + multikey */
     DD_PRE(4, 4, 2, 8, 8, 8, 4)
