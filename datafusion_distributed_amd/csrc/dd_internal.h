@@ -52,6 +52,7 @@ struct dd_kargs {
                            registers in K3 (no pid array: K1 skips its store, K3 its
                            load). Host gates: all-fixed no-validity batch, integer/bool
                            keys, wpb==16 (DD_RHASH=0 disables). */
+    int32_t nt;         /* pre path: non-temporal flush stores (experiment knob) */
     int32_t hl;         /* hidden-load scatter (default on for its shape; DD_K3_HL=0
                            reverts): preload loads in inline asm + hand-counted
                            s_waitcnt so flush stores never drain mid-loop. gmax 4,

@@ -1426,7 +1426,9 @@ __global__ __launch_bounds__(16 * WAVE) void k_scatter_pre(
         const bool more = r + 1 < r1;
         if (more) preload(r + 1);
 
-        /* flush padded to GMAX iterations (clamped rewrite is idempotent) */
+        /* flush padded to GMAX iterations (clamped rewrite is idempotent); a.nt=1
+         * uses non-temporal stores (partition-major output is write-once, never
+         * re-read on this GPU — bypassing L2 leaves it to the read streams) */
 #pragma unroll
         for (int u = 0; u < GMAX; u++) {
             const int i = tid + u * BT;
@@ -1436,14 +1438,23 @@ __global__ __launch_bounds__(16 * WAVE) void k_scatter_pre(
             hl_for<0, NC>([&](auto jc) {
                 constexpr int J = jc.value;
                 void *out = a.cols[J].out_data;
-                if constexpr (EL[J] == 8)
-                    ((uint64_t *)out)[dst] = ((const uint64_t *)stage)[ic];
-                else if constexpr (EL[J] == 4)
-                    ((uint32_t *)out)[dst] = ((const uint32_t *)stage)[ic];
-                else if constexpr (EL[J] == 2)
-                    ((uint16_t *)out)[dst] = ((const uint16_t *)stage)[ic];
-                else
-                    ((uint8_t *)out)[dst] = ((const uint8_t *)stage)[ic];
+                if constexpr (EL[J] == 8) {
+                    uint64_t v = ((const uint64_t *)stage)[ic];
+                    if (a.nt) __builtin_nontemporal_store(v, (uint64_t *)out + dst);
+                    else ((uint64_t *)out)[dst] = v;
+                } else if constexpr (EL[J] == 4) {
+                    uint32_t v = ((const uint32_t *)stage)[ic];
+                    if (a.nt) __builtin_nontemporal_store(v, (uint32_t *)out + dst);
+                    else ((uint32_t *)out)[dst] = v;
+                } else if constexpr (EL[J] == 2) {
+                    uint16_t v = ((const uint16_t *)stage)[ic];
+                    if (a.nt) __builtin_nontemporal_store(v, (uint16_t *)out + dst);
+                    else ((uint16_t *)out)[dst] = v;
+                } else {
+                    uint8_t v = ((const uint8_t *)stage)[ic];
+                    if (a.nt) __builtin_nontemporal_store(v, (uint8_t *)out + dst);
+                    else ((uint8_t *)out)[dst] = v;
+                }
                 stage += (size_t)R * EL[J];
             });
         }
