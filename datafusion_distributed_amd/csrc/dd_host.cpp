@@ -139,7 +139,7 @@ struct dd_partitioner {
     uint32_t *bcounts = nullptr;         /* v1: [nvar][nchunks][P] */
     uint32_t *bpartials = nullptr;       /* v1: [nvar][RANGES][P] */
     uint64_t *part_boffsets = nullptr;   /* [nvar][P+1] */
-    uint16_t *imgb = nullptr;            /* pre: [nseg][sP2] LDS-image bases */
+    uint16_t *imgb = nullptr;            /* pre: [nrounds][sP2] round image bases (roff) */
     uint32_t *partials2 = nullptr;       /* pre: second-level scan partials [64][P] */
     uint32_t *src_row = nullptr;         /* staged-var: permutation out[slot] = input row */
     uint64_t *out_off[DD_KMAX_VAR] = {}; /* staged-var: Arrow byte offsets [n+1] per var */
@@ -470,7 +470,7 @@ extern "C" dd_status dd_partitioner_create(const dd_batch_desc *batch, const int
                      (size_t)(p->pre ? p->pre_nranges : DD_SCAN_RANGES) * P * 4) &&
               halloc((void **)&p->part_offsets, (size_t)(P + 1) * 8);
     if (ok && p->pre)
-        ok = halloc((void **)&p->imgb, (size_t)p->nseg_pad * p->sP2 * 2) &&
+        ok = halloc((void **)&p->imgb, (size_t)p->nrounds * p->sP2 * 2) &&
              halloc((void **)&p->partials2, (size_t)64 * P * 4);
     if (ok && nvar > 0) {
         ok = halloc((void **)&p->part_boffsets, (size_t)nvar * (P + 1) * 8);

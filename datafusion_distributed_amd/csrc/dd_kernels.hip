@@ -1202,27 +1202,33 @@ __global__ __launch_bounds__(16 * WAVE) void k_scatter_hlg(
  *   - the K2 scan runs at segment granularity and REWRITES counts to global slot bases
  *     (within-partition exclusive + part_offsets fold): gbase[s][p] = first output slot
  *     of segment s's rows for partition p.
- *   - k_round_layout precomputes imgb[s][p] (u16): the partition-major LDS-image base of
- *     (segment, partition) within its round = round_off[p] + (gbase[s][p] - gbase[s0][p]).
+ *   - k_round_roff precomputes roff[r][p] (u16, ~4 KB/round): the partition-major
+ *     LDS-image base of partition p within round r (exclusive scan of the round's
+ *     partition counts) — reading only the ROUND-BOUNDARY rows of counts (a per-segment
+ *     imgb table was measured first: +120 MB of K2 traffic for no K3 gain).
  *
- * K3 then needs NO cross-wave communication outside the two image barriers: each wave
- * hidden-loads its own gbase/imgb rows (wave-private LDS, program-ordered), ranks its
- * rows intra-segment with a per-wave LDS counter, places at imgb[pid]+rank with global
- * destination gbase[pid]+rank, and the flush is unchanged. 2 barriers per round instead
- * of 5, no serial scan, no cross-round dependency, uniform full rounds everywhere except
- * the single global tail. Hidden-load discipline identical to k_scatter_hl (issue order
- * per round: G pid loads, L=G*NC column loads, NB base loads, L flush stores):
- *   place waits vmcnt(L)      (prev round's stores in flight),
- *   rank  waits vmcnt(2L+NB)  (this round's col+base loads + stores in flight).
- * NBG/NBI are compile-time dword counts for the gbase/imgb row loads (host gates P to
- * the instantiated tier); imgb rows are padded to even u16 (sP2) so dword loads stay
+ * K3 then needs NO cross-wave LDS traffic outside the two image barriers: each wave
+ * hidden-loads its own gbase row + the round's roff row (wave-private LDS), ranks its
+ * rows intra-segment with a per-wave LDS counter, and places at
+ *   img  = roff[pid] + (gbase_w[pid] - gbase_w0[pid]) + rank    (w0 = wave 0's row,
+ *                                     cross-wave LDS READ ordered by the round barrier)
+ *   gdst = gbase_w[pid] + rank.
+ * 2 barriers per round instead of 5, no serial scan, no cross-round dependency, uniform
+ * full rounds everywhere except the single global tail; one round per BLOCK by default
+ * (rpb=1 measured best — cross-block overlap beats in-block software pipelining here).
+ * Hidden-load discipline identical to k_scatter_hl (issue order per round: G pid loads,
+ * L=G*NC column loads, NB base loads, L flush stores); a single counted wait per round:
+ *   wait vmcnt(L) after the flush (pid+cols+bases retired, stores stay in flight),
+ * then the base rows are written to LDS and rank runs before the round barrier.
+ * NBG/NBI are compile-time dword counts for the gbase/roff row loads (host gates P to
+ * the instantiated tier); roff rows are padded to even u16 (sP2) so dword loads stay
  * aligned. Parity: bit-exact vs the oracle (tests/test_gpu_fuzz.py::test_pre_ab). */
 
 template <int G, int NBG, int NBI, int... Es>
 __global__ __launch_bounds__(16 * WAVE) void k_scatter_pre(
     dd_kargs a, int64_t nrounds, int rpb, uint32_t nparts, int nbits,
     const uint32_t *pid_in, const uint32_t *gbase /* [nseg][P] global slot bases */,
-    const uint16_t *imgb /* [nseg][sP2] LDS-image bases */, uint32_t sP2) {
+    const uint16_t *rofftab /* [nrounds][sP2] round image bases */, uint32_t sP2) {
     constexpr int WPB = 16, GMAX = G;
     constexpr int BT = WPB * WAVE;
     constexpr int R = GMAX * BT;
@@ -1247,7 +1253,7 @@ __global__ __launch_bounds__(16 * WAVE) void k_scatter_pre(
     constexpr slots S{};
 
     extern __shared__ __attribute__((aligned(16))) char smem[];
-    /* carve: stage | dstg u32[R] | per-wave gb u32[P] | per-wave ib u16[sP2] |
+    /* carve: stage | dstg u32[R] | per-wave gb u32[P] | per-wave rf u16[sP2] |
      * per-wave ms u16[P] */
     char *ws = smem;
     char *const stage0 = ws;
@@ -1264,8 +1270,9 @@ __global__ __launch_bounds__(16 * WAVE) void k_scatter_pre(
     const int wid = tid / WAVE;
     const int lane = tid % WAVE;
     uint32_t *gb = gb_all + (size_t)wid * nparts;
+    uint32_t *gb0 = gb_all; /* wave 0's row == the round's first-segment bases */
     uint32_t *ib32 = (uint32_t *)(ib_all + (size_t)wid * sP2); /* dword view */
-    uint16_t *ib = ib_all + (size_t)wid * sP2;
+    uint16_t *rf = ib_all + (size_t)wid * sP2;
     uint16_t *ms = ms_all + (size_t)wid * nparts;
     const uint64_t lt = ((uint64_t)1 << lane) - 1;
 
@@ -1277,9 +1284,9 @@ __global__ __launch_bounds__(16 * WAVE) void k_scatter_pre(
     bool actr[GMAX];
     uint64_t big[S.n8 > 0 ? S.n8 : 1][GMAX];
     uint32_t sml[S.n4 > 0 ? S.n4 : 1][GMAX];
-    uint32_t baser[NB]; /* NBG gbase dwords then NBI imgb dwords */
+    uint32_t baser[NB]; /* NBG gbase dwords then NBI roff dwords */
 
-    const uint32_t ndw = (nparts + 1) / 2; /* imgb row dwords */
+    const uint32_t ndw = (nparts + 1) / 2; /* roff row dwords */
 
     auto preload = [&](int64_t r) {
         const int64_t rstart = r * R;
@@ -1315,7 +1322,9 @@ __global__ __launch_bounds__(16 * WAVE) void k_scatter_pre(
                 }
             }
         });
-        /* base rows (wave-private): gbase u32[P] then imgb u16[sP2] as dwords */
+        /* base rows (wave-private): gbase u32[P] for MY segment, then the round's roff
+         * u16[sP2] as dwords (same row for all 16 waves; the table is ~4 KB/round and
+         * L2-hot, so the duplicate loads are free) */
         const uint32_t *grow = gbase + (size_t)seg * nparts;
 #pragma unroll
         for (int k = 0; k < NBG; k++) {
@@ -1323,7 +1332,7 @@ __global__ __launch_bounds__(16 * WAVE) void k_scatter_pre(
             if (idx >= nparts) idx = nparts - 1; /* redundant re-load keeps counts exact */
             HL_LD32(baser[k], grow + idx);
         }
-        const uint32_t *irow = (const uint32_t *)(imgb + (size_t)seg * sP2);
+        const uint32_t *irow = (const uint32_t *)(rofftab + (size_t)r * sP2);
 #pragma unroll
         for (int k = 0; k < NBI; k++) {
             uint32_t idx = (uint32_t)lane + k * WAVE;
@@ -1332,11 +1341,9 @@ __global__ __launch_bounds__(16 * WAVE) void k_scatter_pre(
         }
     };
 
-    auto wait_pid = [&](auto cnt) {
+    auto wait_all = [&](auto cnt) { /* pid + columns + bases retired */
 #pragma unroll
         for (int g = 0; g < GMAX; g++) hl_tie_wait<cnt.value>(pidr[g]);
-    };
-    auto wait_loads = [&](auto cnt) { /* columns + bases retired */
         hl_for<0, NC>([&](auto jc) {
             constexpr int J = jc.value;
 #pragma unroll
@@ -1347,6 +1354,20 @@ __global__ __launch_bounds__(16 * WAVE) void k_scatter_pre(
         });
 #pragma unroll
         for (int k = 0; k < NB; k++) hl_tie_wait<cnt.value>(baser[k]);
+    };
+
+    auto write_base_rows = [&]() { /* wave-private LDS; ordered for OTHER waves (gb0
+                                      readers) by the round barrier that follows */
+#pragma unroll
+        for (int k = 0; k < NBG; k++) {
+            uint32_t idx = (uint32_t)lane + k * WAVE;
+            if (idx < nparts) gb[idx] = baser[k];
+        }
+#pragma unroll
+        for (int k = 0; k < NBI; k++) {
+            uint32_t idx = (uint32_t)lane + k * WAVE;
+            if (idx < ndw) ib32[idx] = baser[NBG + k];
+        }
     };
 
     auto rank = [&]() {
@@ -1372,11 +1393,11 @@ __global__ __launch_bounds__(16 * WAVE) void k_scatter_pre(
         }
     };
 
-    /* prologue: first round's loads + rank, then a one-time drain */
+    /* prologue: first round's loads, base rows, rank; barrier orders gb0 for place */
     preload(r0);
-    wait_pid(std::integral_constant<int, L + NB>{});
+    wait_all(std::integral_constant<int, 0>{});
+    write_base_rows();
     rank();
-    wait_loads(std::integral_constant<int, 0>{});
     __syncthreads();
 
     for (int64_t r = r0; r < r1; r++) {
@@ -1384,29 +1405,16 @@ __global__ __launch_bounds__(16 * WAVE) void k_scatter_pre(
         const int round_rows =
             (int)(((rstart + R < a.n_rows) ? rstart + R : a.n_rows) - rstart);
 
-        /* place: cols + bases retired at vmcnt(L) — only the previous round's flush
-         * stores stay in flight */
-        wait_loads(std::integral_constant<int, L>{});
-        {
-            /* base rows -> wave-private LDS (program-ordered with the reads below) */
-#pragma unroll
-            for (int k = 0; k < NBG; k++) {
-                uint32_t idx = (uint32_t)lane + k * WAVE;
-                if (idx < nparts) gb[idx] = baser[k];
-            }
-#pragma unroll
-            for (int k = 0; k < NBI; k++) {
-                uint32_t idx = (uint32_t)lane + k * WAVE;
-                if (idx < ndw) ib32[idx] = baser[NBG + k];
-            }
-        }
+        /* place: everything this round needs was waited on and LDS-staged before the
+         * preceding barrier; the only VMEM in flight is the previous round's stores */
 #pragma unroll
         for (int g = 0; g < GMAX; g++) {
             if (!actr[g]) continue;
             const uint32_t pid = pidr[g];
             const uint32_t rk = rankr[g];
-            const uint32_t slot = (uint32_t)ib[pid] + rk;
-            dstg[slot] = gb[pid] + rk;
+            const uint32_t gd = gb[pid] + rk;
+            const uint32_t slot = (uint32_t)rf[pid] + (gd - gb0[pid]);
+            dstg[slot] = gd;
             char *stage = stage0;
             hl_for<0, NC>([&](auto jc) {
                 constexpr int J = jc.value;
@@ -1460,7 +1468,10 @@ __global__ __launch_bounds__(16 * WAVE) void k_scatter_pre(
         }
 
         if (more) {
-            wait_pid(std::integral_constant<int, 2 * L + NB>{});
+            /* one counted wait per round: pid+cols+bases of round r+1 retired, the L
+             * flush stores of round r stay in flight (never drained in-loop) */
+            wait_all(std::integral_constant<int, L>{});
+            write_base_rows();
             rank();
         }
         __syncthreads();
@@ -1516,14 +1527,14 @@ __global__ __launch_bounds__(BLOCK_THREADS) void k_hash_count_seg(
         counts[(size_t)seg * nparts + p] = hist[p];
 }
 
-/* K2d for the pre path: per-round LDS-image layout. counts must already be rewritten to
- * GLOBAL slot bases (k_scan_rewrite with the part_offsets fold). One block per round:
- *   roundcnt[p] = gbase[seg0_{r+1}][p] - gbase[seg0_r][p]   (tail: part_offsets[p+1])
- *   round_off   = exclusive scan over p of roundcnt
- *   imgb[s][p]  = round_off[p] + (gbase[s][p] - gbase[seg0_r][p])   (u16, <= R) */
-__global__ __launch_bounds__(256) void k_round_layout(
+/* K2d for the pre path: per-round image bases. counts must already be rewritten to
+ * GLOBAL slot bases (k_scan_rewrite with the part_offsets fold). One block per round,
+ * touching only the ROUND-BOUNDARY rows of counts:
+ *   roundcnt[p]  = gbase[seg0_{r+1}][p] - gbase[seg0_r][p]   (tail: part_offsets[p+1])
+ *   roff[r][p]   = exclusive scan over p of roundcnt          (u16, <= R) */
+__global__ __launch_bounds__(256) void k_round_roff(
     const uint32_t *counts, const uint64_t *part_offsets, int64_t nrounds, int wpb,
-    uint32_t nparts, uint32_t sP2, uint16_t *imgb) {
+    uint32_t nparts, uint32_t sP2, uint16_t *rofftab) {
     __shared__ uint32_t cnt[DD_MAX_P];
     __shared__ uint32_t roff[DD_MAX_P];
     __shared__ uint32_t tmp[256];
@@ -1536,12 +1547,8 @@ __global__ __launch_bounds__(256) void k_round_layout(
         cnt[p] = (re ? re[p] : (uint32_t)part_offsets[p + 1]) - rb[p];
     __syncthreads();
     dd_block_excl_scan<256>(cnt, roff, nparts, tmp);
-    for (int w = 0; w < wpb; w++) {
-        const uint32_t *sr = counts + ((size_t)r * wpb + w) * nparts;
-        uint16_t *orow = imgb + ((size_t)r * wpb + w) * sP2;
-        for (uint32_t p = tid; p < nparts; p += 256)
-            orow[p] = (uint16_t)(roff[p] + (sr[p] - rb[p]));
-    }
+    uint16_t *orow = rofftab + (size_t)r * sP2;
+    for (uint32_t p = tid; p < nparts; p += 256) orow[p] = (uint16_t)roff[p];
 }
 
 /* ================= K4: var-width bytes for the staged path =================
@@ -1857,16 +1864,16 @@ hipError_t dd_launch_hash_count_seg(const dd_kargs *a, int64_t nseg, int64_t seg
 
 hipError_t dd_launch_round_layout(const uint32_t *counts, const uint64_t *part_offsets,
                                   int64_t nrounds, int wpb, uint32_t nparts, uint32_t sP2,
-                                  uint16_t *imgb, hipStream_t s) {
-    hipLaunchKernelGGL(k_round_layout, dim3((unsigned)nrounds), dim3(256), 0, s, counts,
-                       part_offsets, nrounds, wpb, nparts, sP2, imgb);
+                                  uint16_t *rofftab, hipStream_t s) {
+    hipLaunchKernelGGL(k_round_roff, dim3((unsigned)nrounds), dim3(256), 0, s, counts,
+                       part_offsets, nrounds, wpb, nparts, sP2, rofftab);
     return hipGetLastError();
 }
 
 hipError_t dd_launch_scatter_pre(const dd_kargs *a, int64_t nblocks, int64_t nrounds,
                                  int rpb, uint32_t nparts, int nbits,
                                  const uint32_t *pid_in, const uint32_t *gbase,
-                                 const uint16_t *imgb, uint32_t sP2, int gmax,
+                                 const uint16_t *rofftab, uint32_t sP2, int gmax,
                                  size_t lds_bytes, hipStream_t s) {
     dim3 grid((unsigned)nblocks);
     const int n = a->n_cols;
@@ -1886,7 +1893,7 @@ hipError_t dd_launch_scatter_pre(const dd_kargs *a, int64_t nblocks, int64_t nro
             }                                                                                \
             hipLaunchKernelGGL((k_scatter_pre<GM, NBG_, NBI_, __VA_ARGS__>), grid,           \
                                dim3(16 * WAVE), lds_bytes, s, *a, nrounds, rpb, nparts,      \
-                               nbits, pid_in, gbase, imgb, sP2);                             \
+                               nbits, pid_in, gbase, rofftab, sP2);                          \
             return hipGetLastError();                                                        \
         }                                                                                    \
     }
