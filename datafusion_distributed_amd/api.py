@@ -258,11 +258,14 @@ class Partitioner:
         return bool(lib().dd_partitioner_pids(self.h))
 
     def pids(self):
-        """Per-row partition ids, or None when the spec path recomputed hashes in-kernel
-        (no pid array exists; outputs + row_offsets fully define the partitioning)."""
+        """Per-row partition ids (u32; widened from u8 when the precomputed-layout path
+        stores byte pids), or None when the spec path recomputed hashes in-kernel (no
+        pid array exists; outputs + row_offsets fully define the partitioning)."""
         p = lib().dd_partitioner_pids(self.h)
         if not p:
             return None
+        if lib().dd_partitioner_pid_elem(self.h) == 1:
+            return _d2h(p, self.batch.n_rows, np.uint8).astype(np.uint32)
         return _d2h(p, self.batch.n_rows * 4, np.uint32)
 
     def row_offsets(self):

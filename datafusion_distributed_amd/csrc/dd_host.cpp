@@ -386,7 +386,13 @@ extern "C" dd_status dd_partitioner_create(const dd_batch_desc *batch, const int
                 p->nseg_pad = (nseg + 3) & ~3LL;
                 p->rpb = 1; /* one round per block measured best (0.954 vs 1.079 ms
                                at ceil(nrounds/2048)=8 on the bench shape) */
-                ka.nt = (getenv("DD_PRE_NT") && atoi(getenv("DD_PRE_NT")) == 1) ? 1 : 0;
+                /* non-temporal flush stores: +4.5% K3 measured (DD_PRE_NT=0 reverts) */
+                ka.nt = (getenv("DD_PRE_NT") && atoi(getenv("DD_PRE_NT")) == 0) ? 0 : 1;
+                /* u8 pid array when P fits: 3 B/row less HBM write (K1) + read (K3);
+                   DD_PID8=0 reverts */
+                if (n_partitions <= 256 &&
+                    !(getenv("DD_PID8") && atoi(getenv("DD_PID8")) == 0))
+                    ka.pid8 = 1;
                 if (const char *e = getenv("DD_PRE_RPB")) { /* experiment knob */
                     int v = atoi(e);
                     if (v >= 1 && v <= 256) p->rpb = v;
@@ -464,7 +470,8 @@ extern "C" dd_status dd_partitioner_create(const dd_batch_desc *batch, const int
     auto halloc = [&](void **ptr, size_t bytes) {
         return hipMalloc(ptr, bytes > 0 ? bytes : 1) == hipSuccess;
     };
-    bool ok = (ka.rhash ? true : halloc((void **)&p->pid, (size_t)n * 4)) &&
+    bool ok = (ka.rhash ? true
+                        : halloc((void **)&p->pid, (size_t)n * (ka.pid8 ? 1 : 4))) &&
               halloc((void **)&p->counts, (size_t)nchunks * P * 4) &&
               halloc((void **)&p->partials,
                      (size_t)(p->pre ? p->pre_nranges : DD_SCAN_RANGES) * P * 4) &&
@@ -569,8 +576,10 @@ extern "C" dd_status dd_partitioner_run_phase1(dd_partitioner *p, void *stream) 
     hipStream_t s = (hipStream_t)stream;
     HIP_TRY(hipEventRecord(p->ev[0], s));
     if (p->pre) {
+        HIP_TRY(hipMemsetAsync(p->partials, 0, (size_t)p->pre_nranges * p->nparts * 4, s));
         HIP_TRY(dd_launch_hash_count_seg(&p->ka, p->nseg_pad, p->chunk_rows, p->nparts,
-                                         p->nbits, p->pid, p->counts, p->lds_k1, s));
+                                         p->nbits, p->pid, p->counts, p->partials,
+                                         p->pre_nranges, p->lds_k1, s));
     } else if (p->staged) {
         HIP_TRY(dd_launch_hash_count_tile(&p->ka, p->nchunks, p->chunk_rows, p->nparts,
                                           p->nbits, p->pid, p->counts, p->lds_k1, s));
@@ -651,6 +660,10 @@ extern "C" dd_status dd_partitioner_run(dd_partitioner *p, void *stream) {
 }
 
 extern "C" void dd_partitioner_destroy(dd_partitioner *p) { delete p; }
+
+extern "C" int32_t dd_partitioner_pid_elem(const dd_partitioner *p) {
+    return p->ka.pid8 ? 1 : 4;
+}
 
 extern "C" const uint32_t *dd_partitioner_pids(const dd_partitioner *p) { return p->pid; }
 extern "C" const void *dd_partitioner_col_data(const dd_partitioner *p, int32_t c) {

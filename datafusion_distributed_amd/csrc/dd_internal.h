@@ -52,7 +52,9 @@ struct dd_kargs {
                            registers in K3 (no pid array: K1 skips its store, K3 its
                            load). Host gates: all-fixed no-validity batch, integer/bool
                            keys, wpb==16 (DD_RHASH=0 disables). */
-    int32_t nt;         /* pre path: non-temporal flush stores (experiment knob) */
+    int32_t nt;         /* pre path: non-temporal flush stores (default on) */
+    int32_t pid8;       /* pre path, P <= 256: pid array is u8 (saves 3 B/row of HBM
+                           write in K1 + read in K3); dd_partitioner_pid_elem reports */
     int32_t hl;         /* hidden-load scatter (default on for its shape; DD_K3_HL=0
                            reverts): preload loads in inline asm + hand-counted
                            s_waitcnt so flush stores never drain mid-loop. gmax 4,
@@ -78,7 +80,8 @@ hipError_t dd_launch_scan_deep(uint32_t *counts, int64_t nchunks, uint32_t npart
                                uint64_t *part_offsets, int fold_global, hipStream_t s);
 hipError_t dd_launch_hash_count_seg(const dd_kargs *a, int64_t nseg, int64_t seg_rows,
                                     uint32_t nparts, int nbits, uint32_t *pid_out,
-                                    uint32_t *counts, size_t lds_bytes, hipStream_t s);
+                                    uint32_t *counts, uint32_t *partials, int nranges,
+                                    size_t lds_bytes, hipStream_t s);
 hipError_t dd_launch_round_layout(const uint32_t *counts, const uint64_t *part_offsets,
                                   int64_t nrounds, int wpb, uint32_t nparts, uint32_t sP2,
                                   uint16_t *imgb, hipStream_t s);

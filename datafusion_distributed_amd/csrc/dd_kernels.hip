@@ -1300,8 +1300,16 @@ __global__ __launch_bounds__(16 * WAVE) void k_scatter_pre(
             actr[g] = row < rend;
             rowc[g] = (uint32_t)(actr[g] ? row : rend - 1); /* always issue (counts) */
         }
+        if (a.pid8) { /* uniform branch; exactly G pid loads issue either way */
 #pragma unroll
-        for (int g = 0; g < GMAX; g++) HL_LD32(pidr[g], pid_in + rowc[g]);
+            for (int g = 0; g < GMAX; g++)
+                asm volatile("global_load_ubyte %0, %1, off"
+                             : "=v"(pidr[g])
+                             : "v"((const uint8_t *)pid_in + rowc[g]));
+        } else {
+#pragma unroll
+            for (int g = 0; g < GMAX; g++) HL_LD32(pidr[g], pid_in + rowc[g]);
+        }
         hl_for<0, NC>([&](auto jc) {
             constexpr int J = jc.value;
             const void *base = a.cols[J].data;
@@ -1479,10 +1487,14 @@ __global__ __launch_bounds__(16 * WAVE) void k_scatter_pre(
 }
 
 /* K1 for the pre path: per-SEGMENT histograms (segment = one wave's SEG contiguous rows
- * of a globally-aligned round). 4 segments per 256-thread block; per-wave LDS hist. */
+ * of a globally-aligned round). 4 segments per 256-thread block; per-wave LDS hist.
+ * Fuses the first scan pass: each wave atomicAdds its histogram into its range row of
+ * `partials` (zeroed by the host first), so dd_launch_scan_deep skips k_scan_partial's
+ * full re-read of counts. a.pid8=1 stores pids as u8. */
 __global__ __launch_bounds__(BLOCK_THREADS) void k_hash_count_seg(
-    dd_kargs a, int64_t seg_rows, uint32_t nparts, int nbits, uint32_t *pid_out,
-    uint32_t *counts /* [nseg][P] */) {
+    dd_kargs a, int64_t nseg, int64_t seg_rows, uint32_t nparts, int nbits,
+    uint32_t *pid_out, uint32_t *counts /* [nseg][P] */,
+    uint32_t *partials /* [nranges][P], pre-zeroed */, int nranges) {
     extern __shared__ __attribute__((aligned(16))) char smem[];
     const int wid = threadIdx.x / WAVE;
     const int lane = threadIdx.x % WAVE;
@@ -1510,7 +1522,8 @@ __global__ __launch_bounds__(BLOCK_THREADS) void k_hash_count_seg(
                                     ? (uint32_t)(h & (uint64_t)(tot - 1))
                                     : (uint32_t)(h % (uint64_t)tot);
                 pidu[u] = fine >> a.pid_shift;
-                pid_out[row] = pidu[u];
+                if (a.pid8) ((uint8_t *)pid_out)[row] = (uint8_t)pidu[u];
+                else pid_out[row] = pidu[u];
             }
         }
 #pragma unroll
@@ -1523,8 +1536,15 @@ __global__ __launch_bounds__(BLOCK_THREADS) void k_hash_count_seg(
             }
         }
     }
-    for (uint32_t p = lane; p < nparts; p += WAVE)
-        counts[(size_t)seg * nparts + p] = hist[p];
+    /* range of this segment under k_scan_partial's fair-division mapping
+     * c0(r) = floor(nseg*r/nranges): r = ceil((seg+1)*nranges/nseg) - 1 */
+    const int64_t rw = ((seg + 1) * (int64_t)nranges + nseg - 1) / nseg - 1;
+    uint32_t *prow = partials + (size_t)rw * nparts;
+    for (uint32_t p = lane; p < nparts; p += WAVE) {
+        const uint32_t h = hist[p];
+        counts[(size_t)seg * nparts + p] = h;
+        if (h) atomicAdd(&prow[p], h);
+    }
 }
 
 /* K2d for the pre path: per-round image bases. counts must already be rewritten to
@@ -1833,8 +1853,7 @@ hipError_t dd_launch_scan_deep(uint32_t *counts, int64_t nchunks, uint32_t npart
     int threads = 256;
     int b1 = (int)(((int64_t)nr1 * nparts + threads - 1) / threads);
     int b2 = (int)(((int64_t)nr2 * nparts + threads - 1) / threads);
-    hipLaunchKernelGGL(k_scan_partial, dim3(b1), dim3(threads), 0, s, counts, nchunks,
-                       nparts, nr1, partials);
+    /* level-1 partials were fused into k_hash_count_seg (atomicAdd per wave) */
     hipLaunchKernelGGL(k_scan_partial, dim3(b2), dim3(threads), 0, s, partials, (int64_t)nr1,
                        nparts, nr2, partials2);
     hipLaunchKernelGGL(k_scan_combine, dim3(1), dim3(256), 0, s, partials2, nr2, nparts,
@@ -1849,7 +1868,8 @@ hipError_t dd_launch_scan_deep(uint32_t *counts, int64_t nchunks, uint32_t npart
 
 hipError_t dd_launch_hash_count_seg(const dd_kargs *a, int64_t nseg, int64_t seg_rows,
                                     uint32_t nparts, int nbits, uint32_t *pid_out,
-                                    uint32_t *counts, size_t lds_bytes, hipStream_t s) {
+                                    uint32_t *counts, uint32_t *partials, int nranges,
+                                    size_t lds_bytes, hipStream_t s) {
     dim3 grid((unsigned)(nseg / WAVES_PER_BLOCK));
     if (lds_bytes > 65536) {
         hipError_t e = hipFuncSetAttribute((const void *)k_hash_count_seg,
@@ -1857,8 +1877,8 @@ hipError_t dd_launch_hash_count_seg(const dd_kargs *a, int64_t nseg, int64_t seg
                                            (int)lds_bytes);
         if (e != hipSuccess) return e;
     }
-    hipLaunchKernelGGL(k_hash_count_seg, grid, dim3(BLOCK_THREADS), lds_bytes, s, *a,
-                       seg_rows, nparts, nbits, pid_out, counts);
+    hipLaunchKernelGGL(k_hash_count_seg, grid, dim3(BLOCK_THREADS), lds_bytes, s, *a, nseg,
+                       seg_rows, nparts, nbits, pid_out, counts, partials, nranges);
     return hipGetLastError();
 }
 

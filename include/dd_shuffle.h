@@ -119,6 +119,9 @@ void dd_partitioner_destroy(dd_partitioner *p);
  * not see it under default configuration; partition membership is always fully defined
  * by col_data + row_offsets either way. */
 const uint32_t *dd_partitioner_pids(const dd_partitioner *p);
+/* element size of the pid array: 4 (u32) normally, 1 (u8) when the precomputed-layout
+ * path stores byte pids (n_partitions <= 256) — cast the pids pointer accordingly */
+int32_t dd_partitioner_pid_elem(const dd_partitioner *p);
 const void *dd_partitioner_col_data(const dd_partitioner *p, int32_t col); /* partition-major */
 const uint8_t *dd_partitioner_col_validity(const dd_partitioner *p, int32_t col);
 const uint32_t *dd_partitioner_col_lengths(const dd_partitioner *p, int32_t col); /* utf8 */
