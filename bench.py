@@ -270,6 +270,9 @@ def main():
         # rhash spec path: K3 recomputes hashes from preloaded registers;
         # the 4 B/row pid read is not part of its algorithmic bytes
         k3_bytes -= 4 * rows
+    elif api.lib().dd_partitioner_pid_elem(part.h) == 1:
+        # u8 pid array (pre path, P <= 256): K3 reads 1 B/row of pid, not 4
+        k3_bytes -= 3 * rows
     comm = create_comm(rank, world) if world > 1 else None
     pipeline = world == 1 and args.pipeline and torch.cuda.is_available()
     if pipeline:

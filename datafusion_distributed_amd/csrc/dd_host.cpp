@@ -368,7 +368,11 @@ extern "C" dd_status dd_partitioner_create(const dd_batch_desc *batch, const int
                 int v = atoi(e);
                 if (v == 2 || v == 4) pgmax = v;
             }
-            const int wpb = 16;
+            int wpb = 16;
+            if (const char *e = getenv("DD_PRE_WPB")) { /* experiment knob (8: 2 blk/CU) */
+                int v = atoi(e);
+                if (v == 8 || v == 16) wpb = v;
+            }
             const int64_t R = (int64_t)pgmax * wpb * 64;
             const uint32_t sP2 = (n_partitions + 1) & ~1u;
             const size_t lds = (size_t)R * rowb + 4 * (size_t)R +
@@ -619,7 +623,7 @@ extern "C" dd_status dd_partitioner_run_phase2(dd_partitioner *p, void *stream) 
         const int64_t nblocks = (p->nrounds + p->rpb - 1) / p->rpb;
         HIP_TRY(dd_launch_scatter_pre(&p->ka, nblocks, p->nrounds, p->rpb, p->nparts,
                                       p->nbits, p->pid, p->counts, p->imgb, p->sP2,
-                                      p->gmax, p->lds_k3, s));
+                                      p->gmax, p->wpb, p->lds_k3, s));
     } else if (p->staged) {
         HIP_TRY(dd_launch_scatter_staged(&p->ka, p->nchunks, p->chunk_rows, p->nparts,
                                          p->nbits, p->pid, p->counts, p->part_offsets,

@@ -88,7 +88,7 @@ hipError_t dd_launch_round_layout(const uint32_t *counts, const uint64_t *part_o
 hipError_t dd_launch_scatter_pre(const dd_kargs *a, int64_t nblocks, int64_t nrounds,
                                  int rpb, uint32_t nparts, int nbits,
                                  const uint32_t *pid_in, const uint32_t *gbase,
-                                 const uint16_t *imgb, uint32_t sP2, int gmax,
+                                 const uint16_t *imgb, uint32_t sP2, int gmax, int wpb,
                                  size_t lds_bytes, hipStream_t s);
 hipError_t dd_launch_hash_count_tile(const dd_kargs *a, int64_t nblocks, int64_t tile_rows,
                                      uint32_t nparts, int nbits, uint32_t *pid_out,

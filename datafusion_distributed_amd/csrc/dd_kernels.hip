@@ -1224,12 +1224,12 @@ __global__ __launch_bounds__(16 * WAVE) void k_scatter_hlg(
  * the instantiated tier); roff rows are padded to even u16 (sP2) so dword loads stay
  * aligned. Parity: bit-exact vs the oracle (tests/test_gpu_fuzz.py::test_pre_ab). */
 
-template <int G, int NBG, int NBI, int... Es>
-__global__ __launch_bounds__(16 * WAVE) void k_scatter_pre(
+template <int G, int W, int NBG, int NBI, int... Es>
+__global__ __launch_bounds__(W * WAVE) void k_scatter_pre(
     dd_kargs a, int64_t nrounds, int rpb, uint32_t nparts, int nbits,
     const uint32_t *pid_in, const uint32_t *gbase /* [nseg][P] global slot bases */,
     const uint16_t *rofftab /* [nrounds][sP2] round image bases */, uint32_t sP2) {
-    constexpr int WPB = 16, GMAX = G;
+    constexpr int WPB = W, GMAX = G;
     constexpr int BT = WPB * WAVE;
     constexpr int R = GMAX * BT;
     constexpr int SEG = R / WPB;
@@ -1548,27 +1548,36 @@ __global__ __launch_bounds__(BLOCK_THREADS) void k_hash_count_seg(
 }
 
 /* K2d for the pre path: per-round image bases. counts must already be rewritten to
- * GLOBAL slot bases (k_scan_rewrite with the part_offsets fold). One block per round,
- * touching only the ROUND-BOUNDARY rows of counts:
+ * GLOBAL slot bases (k_scan_rewrite with the part_offsets fold). One WAVE per round
+ * (shfl scan, zero barriers — the one-block-per-round form with dd_block_excl_scan's 3
+ * barriers measured ~0.1 ms of pure launch/latency for 14.6k tiny blocks), touching only
+ * the ROUND-BOUNDARY rows of counts:
  *   roundcnt[p]  = gbase[seg0_{r+1}][p] - gbase[seg0_r][p]   (tail: part_offsets[p+1])
  *   roff[r][p]   = exclusive scan over p of roundcnt          (u16, <= R) */
-__global__ __launch_bounds__(256) void k_round_roff(
+__global__ __launch_bounds__(BLOCK_THREADS) void k_round_roff(
     const uint32_t *counts, const uint64_t *part_offsets, int64_t nrounds, int wpb,
     uint32_t nparts, uint32_t sP2, uint16_t *rofftab) {
-    __shared__ uint32_t cnt[DD_MAX_P];
-    __shared__ uint32_t roff[DD_MAX_P];
-    __shared__ uint32_t tmp[256];
-    const int tid = threadIdx.x;
-    const int64_t r = blockIdx.x;
+    const int wid = threadIdx.x / WAVE;
+    const int lane = threadIdx.x % WAVE;
+    const int64_t r = (int64_t)blockIdx.x * WAVES_PER_BLOCK + wid;
     if (r >= nrounds) return;
     const uint32_t *rb = counts + (size_t)r * wpb * nparts;
     const uint32_t *re = (r + 1 < nrounds) ? counts + (size_t)(r + 1) * wpb * nparts : nullptr;
-    for (uint32_t p = tid; p < nparts; p += 256)
-        cnt[p] = (re ? re[p] : (uint32_t)part_offsets[p + 1]) - rb[p];
-    __syncthreads();
-    dd_block_excl_scan<256>(cnt, roff, nparts, tmp);
     uint16_t *orow = rofftab + (size_t)r * sP2;
-    for (uint32_t p = tid; p < nparts; p += 256) orow[p] = (uint16_t)roff[p];
+    uint32_t carry = 0;
+    for (uint32_t p0 = 0; p0 < nparts; p0 += WAVE) {
+        const uint32_t p = p0 + lane;
+        uint32_t c = 0;
+        if (p < nparts) c = (re ? re[p] : (uint32_t)part_offsets[p + 1]) - rb[p];
+        uint32_t v = c;
+#pragma unroll
+        for (int d = 1; d < WAVE; d <<= 1) {
+            uint32_t u = (uint32_t)__shfl_up((int)v, d);
+            if (lane >= d) v += u;
+        }
+        if (p < nparts) orow[p] = (uint16_t)(carry + v - c); /* exclusive */
+        carry += (uint32_t)__shfl((int)v, WAVE - 1);
+    }
 }
 
 /* ================= K4: var-width bytes for the staged path =================
@@ -1885,7 +1894,8 @@ hipError_t dd_launch_hash_count_seg(const dd_kargs *a, int64_t nseg, int64_t seg
 hipError_t dd_launch_round_layout(const uint32_t *counts, const uint64_t *part_offsets,
                                   int64_t nrounds, int wpb, uint32_t nparts, uint32_t sP2,
                                   uint16_t *rofftab, hipStream_t s) {
-    hipLaunchKernelGGL(k_round_roff, dim3((unsigned)nrounds), dim3(256), 0, s, counts,
+    dim3 grid((unsigned)((nrounds + WAVES_PER_BLOCK - 1) / WAVES_PER_BLOCK));
+    hipLaunchKernelGGL(k_round_roff, grid, dim3(BLOCK_THREADS), 0, s, counts,
                        part_offsets, nrounds, wpb, nparts, sP2, rofftab);
     return hipGetLastError();
 }
@@ -1893,30 +1903,35 @@ hipError_t dd_launch_round_layout(const uint32_t *counts, const uint64_t *part_o
 hipError_t dd_launch_scatter_pre(const dd_kargs *a, int64_t nblocks, int64_t nrounds,
                                  int rpb, uint32_t nparts, int nbits,
                                  const uint32_t *pid_in, const uint32_t *gbase,
-                                 const uint16_t *rofftab, uint32_t sP2, int gmax,
+                                 const uint16_t *rofftab, uint32_t sP2, int gmax, int wpb,
                                  size_t lds_bytes, hipStream_t s) {
     dim3 grid((unsigned)nblocks);
     const int n = a->n_cols;
     auto el = [&](int c) { return (int)a->cols[c].elem; };
-#define DD_PRE(GM, NBG_, NBI_, ...)                                                          \
+#define DD_PREW(GM, W_, NBG_, NBI_, ...)                                                     \
     {                                                                                        \
         const int want[] = {__VA_ARGS__};                                                    \
         const int wn = (int)(sizeof(want) / sizeof(want[0]));                                \
-        bool m = (gmax == GM && n == wn && nparts <= (uint32_t)(NBG_ * WAVE));               \
+        bool m = (gmax == GM && wpb == W_ && n == wn &&                                      \
+                  nparts <= (uint32_t)(NBG_ * WAVE));                                        \
         for (int c = 0; c < wn && m; c++) m = el(c) == want[c];                              \
         if (m) {                                                                             \
             if (lds_bytes > 65536) {                                                         \
                 hipError_t e = hipFuncSetAttribute(                                          \
-                    (const void *)k_scatter_pre<GM, NBG_, NBI_, __VA_ARGS__>,                \
+                    (const void *)k_scatter_pre<GM, W_, NBG_, NBI_, __VA_ARGS__>,            \
                     hipFuncAttributeMaxDynamicSharedMemorySize, (int)lds_bytes);             \
                 if (e != hipSuccess) return e;                                               \
             }                                                                                \
-            hipLaunchKernelGGL((k_scatter_pre<GM, NBG_, NBI_, __VA_ARGS__>), grid,           \
-                               dim3(16 * WAVE), lds_bytes, s, *a, nrounds, rpb, nparts,      \
+            hipLaunchKernelGGL((k_scatter_pre<GM, W_, NBG_, NBI_, __VA_ARGS__>), grid,       \
+                               dim3(W_ * WAVE), lds_bytes, s, *a, nrounds, rpb, nparts,      \
                                nbits, pid_in, gbase, rofftab, sP2);                          \
             return hipGetLastError();                                                        \
         }                                                                                    \
     }
+#define DD_PRE(GM, NBG_, NBI_, ...) DD_PREW(GM, 16, NBG_, NBI_, __VA_ARGS__)
+    /* WPB=8 tier (2 blocks/CU) — headline shape experiment */
+    DD_PREW(4, 8, 2, 1, 8, 8, 8, 4)
+    DD_PREW(4, 8, 2, 1, 8, 8, 8, 4, 4)
     /* P <= 128 tier (NBG=2, NBI=1): all {4,8}^4 combos + the multikey and q1 shapes */
     DD_PRE(4, 2, 1, 4, 4, 4, 4)
     DD_PRE(4, 2, 1, 4, 4, 4, 8)
@@ -1946,6 +1961,7 @@ hipError_t dd_launch_scatter_pre(const dd_kargs *a, int64_t nblocks, int64_t nro
     /* 257..512 tier (NBG=8, NBI=4, G=2 for LDS): headline shape */
     DD_PRE(2, 8, 4, 8, 8, 8, 4)
 #undef DD_PRE
+#undef DD_PREW
     return hipErrorInvalidValue;
 }
 

@@ -26,12 +26,15 @@ CONFIGS = [
     ("pre-nt0", {"DD_PRE_NT": "0"}),
     ("pre-rpb2", {"DD_PRE_RPB": "2"}),
     ("pre-rpb4", {"DD_PRE_RPB": "4"}),
+    ("pre-wpb8", {"DD_PRE_WPB": "8"}),
+    ("pre-wpb8-rpb2", {"DD_PRE_WPB": "8", "DD_PRE_RPB": "2"}),
     ("hl", {"DD_K3_PRE": "0"}),
     ("plain", {"DD_K3_PRE": "0", "DD_K3_HL": "0"}),
     ("pre/again", {}),  # drift check
 ]
 
-KNOBS = ["DD_K3_PRE", "DD_K3_HL", "DD_PID8", "DD_PRE_NT", "DD_PRE_RPB", "DD_PRE_GMAX"]
+KNOBS = ["DD_K3_PRE", "DD_K3_HL", "DD_PID8", "DD_PRE_NT", "DD_PRE_RPB", "DD_PRE_GMAX",
+         "DD_PRE_WPB"]
 
 
 def main():
