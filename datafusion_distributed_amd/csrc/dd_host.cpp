@@ -388,8 +388,9 @@ extern "C" dd_status dd_partitioner_create(const dd_batch_desc *batch, const int
                 if (p->nrounds < 1) p->nrounds = 1;
                 const int64_t nseg = p->nrounds * wpb;
                 p->nseg_pad = (nseg + 3) & ~3LL;
-                p->rpb = 1; /* one round per block measured best (0.954 vs 1.079 ms
-                               at ceil(nrounds/2048)=8 on the bench shape) */
+                p->rpb = 2; /* sweep (tools/sweep_pre.py): rpb2 K3 0.896 ms vs rpb1
+                               0.953 / rpb4 0.924 / ceil(nrounds/2048)=8 1.079 on the
+                               bench shape — small rpb beats in-block pipelining depth */
                 /* non-temporal flush stores: +4.5% K3 measured (DD_PRE_NT=0 reverts) */
                 ka.nt = (getenv("DD_PRE_NT") && atoi(getenv("DD_PRE_NT")) == 0) ? 0 : 1;
                 /* u8 pid array when P fits: 3 B/row less HBM write (K1) + read (K3);
