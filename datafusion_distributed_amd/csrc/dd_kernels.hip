@@ -1507,11 +1507,13 @@ __global__ __launch_bounds__(BLOCK_THREADS) void k_hash_count_seg(
     const int64_t start = seg * seg_rows;
     const int64_t end = (start + seg_rows < a.n_rows) ? start + seg_rows : a.n_rows;
 
-    for (int64_t base = start; base < end; base += 2 * WAVE) {
-        uint32_t pidu[2];
-        bool actu[2];
+    /* 4 independent row-groups per iteration (SEG = 256 at G4: the whole segment in one
+     * iteration) — doubles the loads in flight vs the tile kernel's 2-group form */
+    for (int64_t base = start; base < end; base += 4 * WAVE) {
+        uint32_t pidu[4];
+        bool actu[4];
 #pragma unroll
-        for (int u = 0; u < 2; u++) {
+        for (int u = 0; u < 4; u++) {
             const int64_t row = base + u * WAVE + lane;
             actu[u] = row < end;
             pidu[u] = 0;
@@ -1527,7 +1529,7 @@ __global__ __launch_bounds__(BLOCK_THREADS) void k_hash_count_seg(
             }
         }
 #pragma unroll
-        for (int u = 0; u < 2; u++) {
+        for (int u = 0; u < 4; u++) {
             uint64_t act = __ballot(actu[u]);
             if (actu[u]) {
                 uint64_t eq = dd_eq_mask(pidu[u], act, nbits);
