@@ -1498,9 +1498,16 @@ __global__ __launch_bounds__(BLOCK_THREADS) void k_hash_count_seg(
     extern __shared__ __attribute__((aligned(16))) char smem[];
     const int wid = threadIdx.x / WAVE;
     const int lane = threadIdx.x % WAVE;
-    const int64_t seg = (int64_t)blockIdx.x * WAVES_PER_BLOCK + wid;
     uint32_t *hist = (uint32_t *)smem + (size_t)wid * nparts;
 
+    /* each wave owns a CONSECUTIVE run of segments (grid is capped: one block per
+     * segment-quad was ~58k tiny dispatches, a measurable share of K1) */
+    const int64_t wave_g = (int64_t)blockIdx.x * WAVES_PER_BLOCK + wid;
+    const int64_t nwaves = (int64_t)gridDim.x * WAVES_PER_BLOCK;
+    const int64_t per = (nseg + nwaves - 1) / nwaves;
+    const int64_t s0 = wave_g * per;
+    const int64_t s1 = (s0 + per < nseg) ? s0 + per : nseg;
+    for (int64_t seg = s0; seg < s1; seg++) {
     for (uint32_t p = lane; p < nparts; p += WAVE) hist[p] = 0;
     /* single wave: LDS program order; no barrier */
 
@@ -1547,6 +1554,7 @@ __global__ __launch_bounds__(BLOCK_THREADS) void k_hash_count_seg(
         counts[(size_t)seg * nparts + p] = h;
         if (h) atomicAdd(&prow[p], h);
     }
+    } /* segment loop */
 }
 
 /* K2d for the pre path: per-round image bases. counts must already be rewritten to
@@ -1881,7 +1889,10 @@ hipError_t dd_launch_hash_count_seg(const dd_kargs *a, int64_t nseg, int64_t seg
                                     uint32_t nparts, int nbits, uint32_t *pid_out,
                                     uint32_t *counts, uint32_t *partials, int nranges,
                                     size_t lds_bytes, hipStream_t s) {
-    dim3 grid((unsigned)(nseg / WAVES_PER_BLOCK));
+    int64_t nblk = nseg / WAVES_PER_BLOCK;
+    if (nblk > 2048) nblk = 2048;
+    if (nblk < 1) nblk = 1;
+    dim3 grid((unsigned)nblk);
     if (lds_bytes > 65536) {
         hipError_t e = hipFuncSetAttribute((const void *)k_hash_count_seg,
                                            hipFuncAttributeMaxDynamicSharedMemorySize,

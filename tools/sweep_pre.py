@@ -1,10 +1,10 @@
-"""One-box A/B sweep of the K3 scatter variants (pre / hl / knobs) on the bench shape.
+"""Interleaved one-box A/B of the K3 scatter variants on the bench shape.
 
-Runs every configuration back-to-back on the SAME box (box-to-box spread measured ~20%,
-far above the deltas under test), repeating the first config at the end to bound drift.
-Prints one line per config: mean kernel ms over reps (k1/k2/k3) + whole-step GB/s.
-Environment knobs are read at partitioner-create time, so each config rebuilds its
-partitioner. Usage: python tools/sweep_pre.py [rows] [reps]
+Sequential per-config blocks measured ±4-5% within-box drift (clock/thermal), swamping
+the ~3-6% deltas under test. This sweep creates every config's partitioner up-front
+(env knobs are read at create time; outputs stay resident — 288 GB HBM), then runs the
+configs ROUND-ROBIN and reports per-kernel MEDIANS, so drift hits all configs equally.
+Usage: python tools/sweep_pre.py [rows] [reps]
 """
 import json
 import os
@@ -17,20 +17,15 @@ import numpy as np
 from datafusion_distributed_amd import api
 
 N = int(sys.argv[1]) if len(sys.argv) > 1 else 59_986_052
-REPS = int(sys.argv[2]) if len(sys.argv) > 2 else 15
+REPS = int(sys.argv[2]) if len(sys.argv) > 2 else 21
 P = 128
 
 CONFIGS = [
-    ("pre", {}),
-    ("pre-pid32", {"DD_PID8": "0"}),
-    ("pre-nt0", {"DD_PRE_NT": "0"}),
+    ("pre-rpb1", {"DD_PRE_RPB": "1"}),
     ("pre-rpb2", {"DD_PRE_RPB": "2"}),
     ("pre-rpb4", {"DD_PRE_RPB": "4"}),
-    ("pre-wpb8", {"DD_PRE_WPB": "8"}),
-    ("pre-wpb8-rpb2", {"DD_PRE_WPB": "8", "DD_PRE_RPB": "2"}),
     ("hl", {"DD_K3_PRE": "0"}),
     ("plain", {"DD_K3_PRE": "0", "DD_K3_HL": "0"}),
-    ("pre/again", {}),  # drift check
 ]
 
 KNOBS = ["DD_K3_PRE", "DD_K3_HL", "DD_PID8", "DD_PRE_NT", "DD_PRE_RPB", "DD_PRE_GMAX",
@@ -48,26 +43,28 @@ def main():
     ]
     batch = api.DeviceBatch(cols)
     row_bytes = 28.0
-    results = {}
+    parts = []
     for name, env in CONFIGS:
         for k in KNOBS:
             os.environ.pop(k, None)
         os.environ.update(env)
-        part = api.Partitioner(batch, [0], P)
-        ks = np.zeros(3)
-        for r in range(REPS + 3):
+        parts.append((name, api.Partitioner(batch, [0], P)))
+    samples = {name: [] for name, _ in CONFIGS}
+    for r in range(REPS + 2):
+        for name, part in parts:
             part.run()
             part.sync()
-            if r >= 3:
-                ks += np.array(part.kernel_ms())
-        part.destroy()
-        ks /= REPS
-        step_ms = ks.sum()
-        gbps = N * row_bytes / step_ms / 1e6
-        results[name] = {"k1": round(ks[0], 4), "k2": round(ks[1], 4),
-                         "k3": round(ks[2], 4), "step_ms": round(step_ms, 4),
-                         "GBps": round(gbps, 1)}
+            if r >= 2:
+                samples[name].append(part.kernel_ms())
+    results = {}
+    for name, part in parts:
+        ks = np.median(np.array(samples[name]), axis=0)
+        step_ms = float(ks.sum())
+        results[name] = {"k1": round(float(ks[0]), 4), "k2": round(float(ks[1]), 4),
+                         "k3": round(float(ks[2]), 4), "step_ms": round(step_ms, 4),
+                         "GBps": round(N * row_bytes / step_ms / 1e6, 1)}
         print(name, json.dumps(results[name]), flush=True)
+        part.destroy()
     batch.free()
     print("SWEEP", json.dumps(results))
 
