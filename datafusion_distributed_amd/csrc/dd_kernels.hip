@@ -1889,8 +1889,10 @@ hipError_t dd_launch_hash_count_seg(const dd_kargs *a, int64_t nseg, int64_t seg
                                     uint32_t nparts, int nbits, uint32_t *pid_out,
                                     uint32_t *counts, uint32_t *partials, int nranges,
                                     size_t lds_bytes, hipStream_t s) {
+    /* full grid (one segment per wave): a capped grid with consecutive-segment runs
+     * per wave measured K1 0.26 -> 0.40 ms — the dense all-waves-in-one-row-window
+     * layout wins (DRAM page locality), dispatch overhead is not the cost */
     int64_t nblk = nseg / WAVES_PER_BLOCK;
-    if (nblk > 2048) nblk = 2048;
     if (nblk < 1) nblk = 1;
     dim3 grid((unsigned)nblk);
     if (lds_bytes > 65536) {
