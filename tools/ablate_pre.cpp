@@ -62,12 +62,21 @@ template <int CNT, typename T> __device__ __forceinline__ void tie_wait(T &v) {
     asm volatile("s_waitcnt vmcnt(%1)" : "+v"(v) : "n"(CNT) : "memory");
 }
 
-template <int ABLATE>
+/* DOUBLE: 0=none (full), 1=pid loads x2, 2=col loads x2, 3=base loads x2,
+ * 4=rank x2, 5=place x2, 6=flush stores x2. Marginal phase cost = t(DOUBLE=k) - t(full),
+ * with access patterns and results IDENTICAL to the full kernel (duplicated work is
+ * idempotent), so overlapped phases correctly show ~zero marginal cost. */
+template <int D>
 __global__ __launch_bounds__(BT) void k_abl(
     int64_t n_rows, int64_t nrounds, int rpb, uint32_t nparts, int nbits,
     const uint8_t *pid_in, const uint32_t *gbase, const uint16_t *rofftab, uint32_t sP2,
     const uint64_t *in0, const uint64_t *in1, const uint64_t *in2, const uint32_t *in3,
     uint64_t *o0, uint64_t *o1, uint64_t *o2, uint32_t *o3, uint32_t *sink) {
+    constexpr int NPID = GMAX * (D == 1 ? 2 : 1);
+    constexpr int NCOL = L * (D == 2 ? 2 : 1);
+    constexpr int NBAS = NB * (D == 3 ? 2 : 1);
+    constexpr int NSTO = L * (D == 6 ? 2 : 1);
+    static_assert(NSTO <= 63, "vmcnt");
     extern __shared__ __attribute__((aligned(16))) char smem[];
     char *ws = smem;
     char *const stage0 = ws;
@@ -95,10 +104,13 @@ __global__ __launch_bounds__(BT) void k_abl(
     if (r0 >= nrounds || n_rows == 0) return;
 
     uint32_t pidr[GMAX], rankr[GMAX];
+    uint32_t pidr2[GMAX]; /* D==1 duplicate */
     bool actr[GMAX];
     uint64_t c0v[GMAX], c1v[GMAX], c2v[GMAX];
     uint32_t c3v[GMAX];
-    uint32_t baser[NB];
+    uint64_t x0v[GMAX], x1v[GMAX], x2v[GMAX]; /* D==2 duplicates */
+    uint32_t x3v[GMAX];
+    uint32_t baser[NB], baser2[NB];
     const uint32_t ndw = (nparts + 1) / 2;
 
     auto preload = [&](int64_t r) {
@@ -113,55 +125,67 @@ __global__ __launch_bounds__(BT) void k_abl(
             actr[g] = row < rend;
             rowc[g] = (uint32_t)(actr[g] ? row : rend - 1);
         }
-        if (ABLATE == 6) {
 #pragma unroll
-            for (int g = 0; g < GMAX; g++) pidr[g] = rowc[g] & (nparts - 1);
-        } else {
+        for (int g = 0; g < GMAX; g++) LD8(pidr[g], pid_in + rowc[g]);
+        if (D == 1) {
 #pragma unroll
-            for (int g = 0; g < GMAX; g++) LD8(pidr[g], pid_in + rowc[g]);
+            for (int g = 0; g < GMAX; g++) LD8(pidr2[g], pid_in + rowc[g]);
         }
-        if (ABLATE != 2) {
 #pragma unroll
-            for (int g = 0; g < GMAX; g++) LD64(c0v[g], in0 + rowc[g]);
+        for (int g = 0; g < GMAX; g++) LD64(c0v[g], in0 + rowc[g]);
 #pragma unroll
-            for (int g = 0; g < GMAX; g++) LD64(c1v[g], in1 + rowc[g]);
+        for (int g = 0; g < GMAX; g++) LD64(c1v[g], in1 + rowc[g]);
 #pragma unroll
-            for (int g = 0; g < GMAX; g++) LD64(c2v[g], in2 + rowc[g]);
+        for (int g = 0; g < GMAX; g++) LD64(c2v[g], in2 + rowc[g]);
 #pragma unroll
-            for (int g = 0; g < GMAX; g++) LD32(c3v[g], in3 + rowc[g]);
-        } else {
+        for (int g = 0; g < GMAX; g++) LD32(c3v[g], in3 + rowc[g]);
+        if (D == 2) {
 #pragma unroll
-            for (int g = 0; g < GMAX; g++) {
-                c0v[g] = rowc[g];
-                c1v[g] = rowc[g] * 3;
-                c2v[g] = rowc[g] * 7;
-                c3v[g] = rowc[g];
-            }
+            for (int g = 0; g < GMAX; g++) LD64(x0v[g], in0 + rowc[g]);
+#pragma unroll
+            for (int g = 0; g < GMAX; g++) LD64(x1v[g], in1 + rowc[g]);
+#pragma unroll
+            for (int g = 0; g < GMAX; g++) LD64(x2v[g], in2 + rowc[g]);
+#pragma unroll
+            for (int g = 0; g < GMAX; g++) LD32(x3v[g], in3 + rowc[g]);
         }
-        if (ABLATE != 5) {
-            const uint32_t *grow = gbase + (size_t)seg * nparts;
+        const uint32_t *grow = gbase + (size_t)seg * nparts;
+        const uint32_t *irow = (const uint32_t *)(rofftab + (size_t)r * sP2);
+#pragma unroll
+        for (int k = 0; k < NBG; k++) {
+            uint32_t idx = (uint32_t)lane + k * WAVE;
+            if (idx >= nparts) idx = nparts - 1;
+            LD32(baser[k], grow + idx);
+        }
+#pragma unroll
+        for (int k = 0; k < NBI; k++) {
+            uint32_t idx = (uint32_t)lane + k * WAVE;
+            if (idx >= ndw) idx = ndw - 1;
+            LD32(baser[NBG + k], irow + idx);
+        }
+        if (D == 3) {
 #pragma unroll
             for (int k = 0; k < NBG; k++) {
                 uint32_t idx = (uint32_t)lane + k * WAVE;
                 if (idx >= nparts) idx = nparts - 1;
-                LD32(baser[k], grow + idx);
+                LD32(baser2[k], grow + idx);
             }
-            const uint32_t *irow = (const uint32_t *)(rofftab + (size_t)r * sP2);
 #pragma unroll
             for (int k = 0; k < NBI; k++) {
                 uint32_t idx = (uint32_t)lane + k * WAVE;
                 if (idx >= ndw) idx = ndw - 1;
-                LD32(baser[NBG + k], irow + idx);
+                LD32(baser2[NBG + k], irow + idx);
             }
-        } else {
-#pragma unroll
-            for (int k = 0; k < NB; k++) baser[k] = 0;
         }
     };
 
     auto wait_all = [&](auto cnt) {
 #pragma unroll
         for (int g = 0; g < GMAX; g++) tie_wait<cnt.value>(pidr[g]);
+        if (D == 1) {
+#pragma unroll
+            for (int g = 0; g < GMAX; g++) tie_wait<cnt.value>(pidr2[g]);
+        }
 #pragma unroll
         for (int g = 0; g < GMAX; g++) {
             tie_wait<cnt.value>(c0v[g]);
@@ -169,8 +193,21 @@ __global__ __launch_bounds__(BT) void k_abl(
             tie_wait<cnt.value>(c2v[g]);
             tie_wait<cnt.value>(c3v[g]);
         }
+        if (D == 2) {
+#pragma unroll
+            for (int g = 0; g < GMAX; g++) {
+                tie_wait<cnt.value>(x0v[g]);
+                tie_wait<cnt.value>(x1v[g]);
+                tie_wait<cnt.value>(x2v[g]);
+                tie_wait<cnt.value>(x3v[g]);
+            }
+        }
 #pragma unroll
         for (int k = 0; k < NB; k++) tie_wait<cnt.value>(baser[k]);
+        if (D == 3) {
+#pragma unroll
+            for (int k = 0; k < NB; k++) tie_wait<cnt.value>(baser2[k]);
+        }
     };
 
     auto write_base_rows = [&]() {
@@ -184,14 +221,16 @@ __global__ __launch_bounds__(BT) void k_abl(
             uint32_t idx = (uint32_t)lane + k * WAVE;
             if (idx < ndw) ib32[idx] = baser[NBG + k];
         }
+        if (D == 3) { /* idempotent duplicate */
+#pragma unroll
+            for (int k = 0; k < NBG; k++) {
+                uint32_t idx = (uint32_t)lane + k * WAVE;
+                if (idx < nparts) gb[idx] = baser2[k];
+            }
+        }
     };
 
-    auto rank = [&]() {
-        if (ABLATE == 3) {
-#pragma unroll
-            for (int g = 0; g < GMAX; g++) rankr[g] = 0; /* WRONG, timing only */
-            return;
-        }
+    auto rank1 = [&]() {
         for (uint32_t p = lane; p < nparts; p += WAVE) ms[p] = 0;
 #pragma unroll
         for (int g = 0; g < GMAX; g++) {
@@ -213,6 +252,30 @@ __global__ __launch_bounds__(BT) void k_abl(
             rankr[g] = rk;
         }
     };
+    auto rank = [&]() {
+        rank1();
+        if (D == 4) rank1(); /* identical recompute (LDS side effects; not elided) */
+    };
+
+    auto place = [&]() {
+#pragma unroll
+        for (int g = 0; g < GMAX; g++) {
+            if (!actr[g]) continue;
+            const uint32_t pid = pidr[g];
+            const uint32_t rk = rankr[g];
+            const uint32_t gd = gb[pid] + rk;
+            const uint32_t slot = (uint32_t)rf[pid] + (gd - gb0[pid]);
+            dstg[slot] = gd;
+            char *stage = stage0;
+            ((uint64_t *)stage)[slot] = c0v[g];
+            stage += (size_t)R * 8;
+            ((uint64_t *)stage)[slot] = c1v[g];
+            stage += (size_t)R * 8;
+            ((uint64_t *)stage)[slot] = c2v[g];
+            stage += (size_t)R * 8;
+            ((uint32_t *)stage)[slot] = c3v[g];
+        }
+    };
 
     preload(r0);
     wait_all(std::integral_constant<int, 0>{});
@@ -225,78 +288,19 @@ __global__ __launch_bounds__(BT) void k_abl(
         const int round_rows =
             (int)(((rstart + R < n_rows) ? rstart + R : n_rows) - rstart);
 
-        if (ABLATE == 4) { /* direct scatter, no LDS round trip */
-#pragma unroll
-            for (int g = 0; g < GMAX; g++) {
-                if (!actr[g]) continue;
-                const uint32_t pid = pidr[g];
-                uint64_t dst = (uint64_t)gb[pid] + rankr[g];
-                if (dst >= (uint64_t)n_rows) dst = 0;
-                __builtin_nontemporal_store(c0v[g], o0 + dst);
-                __builtin_nontemporal_store(c1v[g], o1 + dst);
-                __builtin_nontemporal_store(c2v[g], o2 + dst);
-                __builtin_nontemporal_store(c3v[g], o3 + dst);
-            }
-            __syncthreads();
-            const bool more4 = r + 1 < r1;
-            if (more4) {
-                preload(r + 1);
-                wait_all(std::integral_constant<int, 0>{});
-                write_base_rows();
-                rank();
-            }
-            __syncthreads();
-            continue;
-        }
-
-#pragma unroll
-        for (int g = 0; g < GMAX; g++) {
-            if (!actr[g]) continue;
-            const uint32_t pid = pidr[g];
-            const uint32_t rk = rankr[g];
-            const uint32_t gd = gb[pid] + rk;
-            uint32_t slot = (uint32_t)rf[pid] + (gd - gb0[pid]);
-            if (ABLATE == 3 || ABLATE == 5) slot = (uint32_t)(tid * GMAX + g); /* safety */
-            if (slot >= R) slot = 0;
-            dstg[slot] = gd;
-            char *stage = stage0;
-            ((uint64_t *)stage)[slot] = c0v[g];
-            stage += (size_t)R * 8;
-            ((uint64_t *)stage)[slot] = c1v[g];
-            stage += (size_t)R * 8;
-            ((uint64_t *)stage)[slot] = c2v[g];
-            stage += (size_t)R * 8;
-            ((uint32_t *)stage)[slot] = c3v[g];
-        }
+        place();
+        if (D == 5) place(); /* idempotent duplicate */
         __syncthreads();
 
         const bool more = r + 1 < r1;
         if (more) preload(r + 1);
 
-        if (ABLATE == 1) { /* read LDS into a sink; no global stores */
-            uint32_t acc = 0;
+        for (int rep = 0; rep < (D == 6 ? 2 : 1); rep++) {
 #pragma unroll
             for (int u = 0; u < GMAX; u++) {
                 const int i = tid + u * BT;
                 const int ic = (i < round_rows) ? i : (round_rows - 1);
-                acc ^= dstg[ic];
-                char *stage = stage0;
-                acc ^= (uint32_t)((uint64_t *)stage)[ic];
-                stage += (size_t)R * 8;
-                acc ^= (uint32_t)((uint64_t *)stage)[ic];
-                stage += (size_t)R * 8;
-                acc ^= (uint32_t)((uint64_t *)stage)[ic];
-                stage += (size_t)R * 8;
-                acc ^= ((uint32_t *)stage)[ic];
-            }
-            if (acc == 0xdeadbeefu) sink[0] = acc;
-        } else {
-#pragma unroll
-            for (int u = 0; u < GMAX; u++) {
-                const int i = tid + u * BT;
-                const int ic = (i < round_rows) ? i : (round_rows - 1);
-                uint64_t dst = dstg[ic];
-                if (dst >= (uint64_t)n_rows) dst = 0; /* safety for 3/5 */
+                const uint64_t dst = dstg[ic];
                 char *stage = stage0;
                 __builtin_nontemporal_store(((const uint64_t *)stage)[ic], o0 + dst);
                 stage += (size_t)R * 8;
@@ -309,12 +313,13 @@ __global__ __launch_bounds__(BT) void k_abl(
         }
 
         if (more) {
-            wait_all(std::integral_constant<int, L>{});
+            wait_all(std::integral_constant<int, NSTO>{});
             write_base_rows();
             rank();
         }
         __syncthreads();
     }
+    if (sink && tid == 0x7fffffff) sink[0] = rankr[0] + (uint32_t)c0v[0] + baser[0];
 }
 
 int main() {
@@ -413,10 +418,10 @@ int main() {
         printf("%-28s %.4f ms\n", name, best);
         HC(hipEventDestroy(e0));
         HC(hipEventDestroy(e1));
+        return best;
     };
 
-    run(std::integral_constant<int, 0>{}, "full");
-    /* verify ABLATE=0 on a sample */
+    float full = run(std::integral_constant<int, 0>{}, "full");
     {
         std::vector<uint64_t> out0(n);
         HC(hipMemcpy(out0.data(), d_o0, n * 8, hipMemcpyDeviceToHost));
@@ -432,11 +437,15 @@ int main() {
         }
         printf(bad ? "VERIFY FAILED\n" : "verify ok (full kernel, all rows)\n");
     }
-    run(std::integral_constant<int, 1>{}, "no_flush_stores");
-    run(std::integral_constant<int, 2>{}, "no_column_loads");
-    run(std::integral_constant<int, 3>{}, "no_rank");
-    run(std::integral_constant<int, 4>{}, "no_lds_staging_direct");
-    run(std::integral_constant<int, 5>{}, "no_base_rows");
-    run(std::integral_constant<int, 6>{}, "no_pid_load");
+    const char *names[] = {"", "2x_pid_loads", "2x_col_loads", "2x_base_loads", "2x_rank",
+                           "2x_place", "2x_flush_stores"};
+    float t1 = run(std::integral_constant<int, 1>{}, names[1]);
+    float t2 = run(std::integral_constant<int, 2>{}, names[2]);
+    float t3 = run(std::integral_constant<int, 3>{}, names[3]);
+    float t4 = run(std::integral_constant<int, 4>{}, names[4]);
+    float t5 = run(std::integral_constant<int, 5>{}, names[5]);
+    float t6 = run(std::integral_constant<int, 6>{}, names[6]);
+    printf("marginal ms: pid=%.3f cols=%.3f bases=%.3f rank=%.3f place=%.3f flush=%.3f\n",
+           t1 - full, t2 - full, t3 - full, t4 - full, t5 - full, t6 - full);
     return 0;
 }
