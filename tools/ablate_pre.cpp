@@ -71,10 +71,8 @@ __global__ __launch_bounds__(BT) void k_abl(
     int64_t n_rows, int64_t nrounds, int rpb, uint32_t nparts, int nbits,
     const uint8_t *pid_in, const uint32_t *gbase, const uint16_t *rofftab, uint32_t sP2,
     const uint64_t *in0, const uint64_t *in1, const uint64_t *in2, const uint32_t *in3,
-    uint64_t *o0, uint64_t *o1, uint64_t *o2, uint32_t *o3, uint32_t *sink) {
-    constexpr int NPID = GMAX * (D == 1 ? 2 : 1);
-    constexpr int NCOL = L * (D == 2 ? 2 : 1);
-    constexpr int NBAS = NB * (D == 3 ? 2 : 1);
+    uint64_t *o0, uint64_t *o1, uint64_t *o2, uint32_t *o3, uint64_t *ob0, uint64_t *ob1,
+    uint64_t *ob2, uint32_t *ob3, uint32_t *sink) {
     constexpr int NSTO = L * (D == 6 ? 2 : 1);
     static_assert(NSTO <= 63, "vmcnt");
     extern __shared__ __attribute__((aligned(16))) char smem[];
@@ -296,19 +294,32 @@ __global__ __launch_bounds__(BT) void k_abl(
         if (more) preload(r + 1);
 
         for (int rep = 0; rep < (D == 6 ? 2 : 1); rep++) {
+            /* rep 1 (D==6) writes a SECOND buffer set: same pattern, no WAW hazard */
+            uint64_t *q0 = rep ? ob0 : o0, *q1 = rep ? ob1 : o1, *q2 = rep ? ob2 : o2;
+            uint32_t *q3 = rep ? ob3 : o3;
 #pragma unroll
             for (int u = 0; u < GMAX; u++) {
                 const int i = tid + u * BT;
                 const int ic = (i < round_rows) ? i : (round_rows - 1);
                 const uint64_t dst = dstg[ic];
                 char *stage = stage0;
-                __builtin_nontemporal_store(((const uint64_t *)stage)[ic], o0 + dst);
-                stage += (size_t)R * 8;
-                __builtin_nontemporal_store(((const uint64_t *)stage)[ic], o1 + dst);
-                stage += (size_t)R * 8;
-                __builtin_nontemporal_store(((const uint64_t *)stage)[ic], o2 + dst);
-                stage += (size_t)R * 8;
-                __builtin_nontemporal_store(((const uint32_t *)stage)[ic], o3 + dst);
+                if (D == 7) { /* normal (cached) stores */
+                    ((uint64_t *)q0)[dst] = ((const uint64_t *)stage)[ic];
+                    stage += (size_t)R * 8;
+                    ((uint64_t *)q1)[dst] = ((const uint64_t *)stage)[ic];
+                    stage += (size_t)R * 8;
+                    ((uint64_t *)q2)[dst] = ((const uint64_t *)stage)[ic];
+                    stage += (size_t)R * 8;
+                    ((uint32_t *)q3)[dst] = ((const uint32_t *)stage)[ic];
+                } else {
+                    __builtin_nontemporal_store(((const uint64_t *)stage)[ic], q0 + dst);
+                    stage += (size_t)R * 8;
+                    __builtin_nontemporal_store(((const uint64_t *)stage)[ic], q1 + dst);
+                    stage += (size_t)R * 8;
+                    __builtin_nontemporal_store(((const uint64_t *)stage)[ic], q2 + dst);
+                    stage += (size_t)R * 8;
+                    __builtin_nontemporal_store(((const uint32_t *)stage)[ic], q3 + dst);
+                }
             }
         }
 
@@ -365,9 +376,9 @@ int main() {
     }
 
     uint8_t *d_pid;
-    uint32_t *d_gbase, *d_in3, *d_o3, *d_sink;
+    uint32_t *d_gbase, *d_in3, *d_o3, *d_ob3, *d_sink;
     uint16_t *d_roff;
-    uint64_t *d_in0, *d_in1, *d_in2, *d_o0, *d_o1, *d_o2;
+    uint64_t *d_in0, *d_in1, *d_in2, *d_o0, *d_o1, *d_o2, *d_ob0, *d_ob1, *d_ob2;
     HC(hipMalloc(&d_pid, n));
     HC(hipMalloc(&d_gbase, gbase.size() * 4));
     HC(hipMalloc(&d_roff, roff.size() * 2));
@@ -379,6 +390,10 @@ int main() {
     HC(hipMalloc(&d_o1, n * 8));
     HC(hipMalloc(&d_o2, n * 8));
     HC(hipMalloc(&d_o3, n * 4));
+    HC(hipMalloc(&d_ob0, n * 8));
+    HC(hipMalloc(&d_ob1, n * 8));
+    HC(hipMalloc(&d_ob2, n * 8));
+    HC(hipMalloc(&d_ob3, n * 4));
     HC(hipMalloc(&d_sink, 4));
     HC(hipMemcpy(d_pid, pid.data(), n, hipMemcpyHostToDevice));
     HC(hipMemcpy(d_gbase, gbase.data(), gbase.size() * 4, hipMemcpyHostToDevice));
@@ -408,7 +423,8 @@ int main() {
             HC(hipEventRecord(e0, 0));
             hipLaunchKernelGGL((k_abl<A>), dim3((unsigned)nblocks), dim3(BT), lds, 0, n,
                                nrounds, rpb, P, nbits, d_pid, d_gbase, d_roff, sP2, d_in0,
-                               d_in1, d_in2, d_in3, d_o0, d_o1, d_o2, d_o3, d_sink);
+                               d_in1, d_in2, d_in3, d_o0, d_o1, d_o2, d_o3, d_ob0,
+                               d_ob1, d_ob2, d_ob3, d_sink);
             HC(hipEventRecord(e1, 0));
             HC(hipEventSynchronize(e1));
             float ms;
@@ -445,7 +461,9 @@ int main() {
     float t4 = run(std::integral_constant<int, 4>{}, names[4]);
     float t5 = run(std::integral_constant<int, 5>{}, names[5]);
     float t6 = run(std::integral_constant<int, 6>{}, names[6]);
-    printf("marginal ms: pid=%.3f cols=%.3f bases=%.3f rank=%.3f place=%.3f flush=%.3f\n",
-           t1 - full, t2 - full, t3 - full, t4 - full, t5 - full, t6 - full);
+    float t7 = run(std::integral_constant<int, 7>{}, "normal_stores(non-NT)");
+    printf("marginal ms: pid=%.3f cols=%.3f bases=%.3f rank=%.3f place=%.3f flush=%.3f "
+           "nt_delta=%.3f\n",
+           t1 - full, t2 - full, t3 - full, t4 - full, t5 - full, t6 - full, t7 - full);
     return 0;
 }
