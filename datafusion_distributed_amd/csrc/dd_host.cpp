@@ -391,8 +391,6 @@ extern "C" dd_status dd_partitioner_create(const dd_batch_desc *batch, const int
                 p->rpb = 2; /* sweep (tools/sweep_pre.py): rpb2 K3 0.896 ms vs rpb1
                                0.953 / rpb4 0.924 / ceil(nrounds/2048)=8 1.079 on the
                                bench shape — small rpb beats in-block pipelining depth */
-                /* non-temporal flush stores: +4.5% K3 measured (DD_PRE_NT=0 reverts) */
-                ka.nt = (getenv("DD_PRE_NT") && atoi(getenv("DD_PRE_NT")) == 0) ? 0 : 1;
                 /* u8 pid array when P fits: 3 B/row less HBM write (K1) + read (K3);
                    DD_PID8=0 reverts */
                 if (n_partitions <= 256 &&

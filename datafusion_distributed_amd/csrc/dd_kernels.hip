@@ -1442,9 +1442,13 @@ __global__ __launch_bounds__(W * WAVE) void k_scatter_pre(
         const bool more = r + 1 < r1;
         if (more) preload(r + 1);
 
-        /* flush padded to GMAX iterations (clamped rewrite is idempotent); a.nt=1
-         * uses non-temporal stores (partition-major output is write-once, never
-         * re-read on this GPU — bypassing L2 leaves it to the read streams) */
+        /* flush padded to GMAX iterations (clamped rewrite is idempotent). Stores are
+         * PLAIN (cached): true non-temporal stores (global_store ... nt) measured 30%
+         * SLOWER in the standalone replica (tools/ablate_pre.cpp: 1.51 vs 1.16 ms) —
+         * partition runs average ~R/P bytes and their 128 B line edges only coalesce
+         * through L2, which nt bypasses. (The earlier a.nt knob never actually emitted
+         * nt bits: hipcc merged the branch arms and dropped the hint — the measured
+         * "nt wins" were box noise. Knob removed.) */
 #pragma unroll
         for (int u = 0; u < GMAX; u++) {
             const int i = tid + u * BT;
@@ -1454,23 +1458,14 @@ __global__ __launch_bounds__(W * WAVE) void k_scatter_pre(
             hl_for<0, NC>([&](auto jc) {
                 constexpr int J = jc.value;
                 void *out = a.cols[J].out_data;
-                if constexpr (EL[J] == 8) {
-                    uint64_t v = ((const uint64_t *)stage)[ic];
-                    if (a.nt) __builtin_nontemporal_store(v, (uint64_t *)out + dst);
-                    else ((uint64_t *)out)[dst] = v;
-                } else if constexpr (EL[J] == 4) {
-                    uint32_t v = ((const uint32_t *)stage)[ic];
-                    if (a.nt) __builtin_nontemporal_store(v, (uint32_t *)out + dst);
-                    else ((uint32_t *)out)[dst] = v;
-                } else if constexpr (EL[J] == 2) {
-                    uint16_t v = ((const uint16_t *)stage)[ic];
-                    if (a.nt) __builtin_nontemporal_store(v, (uint16_t *)out + dst);
-                    else ((uint16_t *)out)[dst] = v;
-                } else {
-                    uint8_t v = ((const uint8_t *)stage)[ic];
-                    if (a.nt) __builtin_nontemporal_store(v, (uint8_t *)out + dst);
-                    else ((uint8_t *)out)[dst] = v;
-                }
+                if constexpr (EL[J] == 8)
+                    ((uint64_t *)out)[dst] = ((const uint64_t *)stage)[ic];
+                else if constexpr (EL[J] == 4)
+                    ((uint32_t *)out)[dst] = ((const uint32_t *)stage)[ic];
+                else if constexpr (EL[J] == 2)
+                    ((uint16_t *)out)[dst] = ((const uint16_t *)stage)[ic];
+                else
+                    ((uint8_t *)out)[dst] = ((const uint8_t *)stage)[ic];
                 stage += (size_t)R * EL[J];
             });
         }

@@ -303,15 +303,7 @@ __global__ __launch_bounds__(BT) void k_abl(
                 const int ic = (i < round_rows) ? i : (round_rows - 1);
                 const uint64_t dst = dstg[ic];
                 char *stage = stage0;
-                if (D == 7) { /* normal (cached) stores */
-                    ((uint64_t *)q0)[dst] = ((const uint64_t *)stage)[ic];
-                    stage += (size_t)R * 8;
-                    ((uint64_t *)q1)[dst] = ((const uint64_t *)stage)[ic];
-                    stage += (size_t)R * 8;
-                    ((uint64_t *)q2)[dst] = ((const uint64_t *)stage)[ic];
-                    stage += (size_t)R * 8;
-                    ((uint32_t *)q3)[dst] = ((const uint32_t *)stage)[ic];
-                } else {
+                if (D == 7) { /* true-NT stores (negative result: ~30% slower) */
                     __builtin_nontemporal_store(((const uint64_t *)stage)[ic], q0 + dst);
                     stage += (size_t)R * 8;
                     __builtin_nontemporal_store(((const uint64_t *)stage)[ic], q1 + dst);
@@ -319,6 +311,14 @@ __global__ __launch_bounds__(BT) void k_abl(
                     __builtin_nontemporal_store(((const uint64_t *)stage)[ic], q2 + dst);
                     stage += (size_t)R * 8;
                     __builtin_nontemporal_store(((const uint32_t *)stage)[ic], q3 + dst);
+                } else { /* plain (cached) stores — the product default */
+                    ((uint64_t *)q0)[dst] = ((const uint64_t *)stage)[ic];
+                    stage += (size_t)R * 8;
+                    ((uint64_t *)q1)[dst] = ((const uint64_t *)stage)[ic];
+                    stage += (size_t)R * 8;
+                    ((uint64_t *)q2)[dst] = ((const uint64_t *)stage)[ic];
+                    stage += (size_t)R * 8;
+                    ((uint32_t *)q3)[dst] = ((const uint32_t *)stage)[ic];
                 }
             }
         }
@@ -461,7 +461,7 @@ int main() {
     float t4 = run(std::integral_constant<int, 4>{}, names[4]);
     float t5 = run(std::integral_constant<int, 5>{}, names[5]);
     float t6 = run(std::integral_constant<int, 6>{}, names[6]);
-    float t7 = run(std::integral_constant<int, 7>{}, "normal_stores(non-NT)");
+    float t7 = run(std::integral_constant<int, 7>{}, "nt_stores(negative)");
     printf("marginal ms: pid=%.3f cols=%.3f bases=%.3f rank=%.3f place=%.3f flush=%.3f "
            "nt_delta=%.3f\n",
            t1 - full, t2 - full, t3 - full, t4 - full, t5 - full, t6 - full, t7 - full);
