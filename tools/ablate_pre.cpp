@@ -334,6 +334,7 @@ __global__ __launch_bounds__(BT) void k_abl(
 }
 
 int main() {
+    setvbuf(stdout, nullptr, _IONBF, 0);
     const int64_t n = 59986052;
     const uint32_t P = 128;
     const uint32_t sP2 = P;
