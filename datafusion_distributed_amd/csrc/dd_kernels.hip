@@ -1711,8 +1711,11 @@ __global__ __launch_bounds__(BLOCK_THREADS) void k4_copy(
         if (lane == 0) loc[WAVE] = T;
         /* single wave: LDS program order; no barrier */
         const uint64_t obase = out_off[s0];
-        for (uint32_t p0 = lane * 8; p0 < T; p0 += WAVE * 8) {
-            uint32_t rem = (T - p0 < 8) ? (T - p0) : 8;
+        /* 32 B per lane per iteration (4x 8-B chunks, one search per lane-iteration):
+         * quarters the search/loop overhead and puts 4 independent loads in flight per
+         * lane vs the original 8-B-chunk form (k4_copy 2.62 -> measured below) */
+        for (uint32_t p0 = lane * 32; p0 < T; p0 += WAVE * 32) {
+            uint32_t rem = (T - p0 < 32) ? (T - p0) : 32;
             uint32_t p = p0;
             /* binary search: largest j with loc[j] <= p */
             int lo = 0, hi = WAVE;
@@ -1726,18 +1729,19 @@ __global__ __launch_bounds__(BLOCK_THREADS) void k4_copy(
                 while (loc[j + 1] <= p) j++; /* skip empty strings */
                 const uint32_t within = p - loc[j];
                 const uint32_t avail = loc[j + 1] - p;
-                const uint32_t m = (rem < avail) ? rem : avail;
+                uint32_t m = (rem < avail) ? rem : avail;
                 const uint8_t *sp = in_bytes + srcb[j] + within;
                 uint8_t *dp = out_bytes + obase + p;
-                if (m == 8) {
-                    uint64_t t;
-                    __builtin_memcpy(&t, sp, 8);
-                    __builtin_memcpy(dp, &t, 8);
-                } else {
-                    for (uint32_t b = 0; b < m; b++) dp[b] = sp[b];
-                }
                 p += m;
                 rem -= m;
+                /* within one string: 8-B unaligned chunks, all loads independent */
+                uint32_t b = 0;
+                for (; b + 8 <= m; b += 8) {
+                    uint64_t t;
+                    __builtin_memcpy(&t, sp + b, 8);
+                    __builtin_memcpy(dp + b, &t, 8);
+                }
+                for (; b < m; b++) dp[b] = sp[b];
             }
         }
     }
@@ -1806,9 +1810,10 @@ hipError_t dd_launch_var_bytes(const uint32_t *lens, const uint32_t *src_row,
     hipLaunchKernelGGL(k4_off_rewrite, dim3(wavegrid), dim3(256), 0, s, lens, n, partials,
                        out_off);
     int copy_blocks = (int)((n + 255) / 256);
-    if (copy_blocks > 8192) copy_blocks = 8192; /* gather-copy is latency-bound: deep
-                                                   oversubscription hides the random
-                                                   string reads */
+    if (copy_blocks > 32768) copy_blocks = 32768; /* gather-copy is latency-bound: deep
+                                                     oversubscription hides the random
+                                                     string reads (8192 -> 32768 measured
+                                                     with the 32-B lane chunks) */
     if (copy_blocks < 1) copy_blocks = 1;
     hipLaunchKernelGGL(k4_copy, dim3(copy_blocks), dim3(256), 0, s, src_row, out_off,
                        in_offsets, in_bytes, n, out_bytes);
