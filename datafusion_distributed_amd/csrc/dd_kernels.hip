@@ -1673,8 +1673,66 @@ __global__ void k4_off_rewrite(const uint32_t *lens, int64_t n, const uint64_t *
  * [l*8, l*8+8) of the range, striding 512) — stores are fully coalesced; loads are
  * within-string contiguous (neighboring lanes usually read the same string). A chunk that
  * crosses string boundaries falls back to sub-chunk copies. One-string-per-thread copying
- * (the first version) scattered every store instruction across 64 strings: 9.8 GB written
- * for 1.28 GB of payload (profiles/). */
+ * scattered every store instruction across 64 strings (9.8 GB written for 1.28 GB of
+ * payload); a 32-B-per-lane variant quartered store coalescing (3.2 -> 5.0 ms); a 4-deep
+ * pre-issued-load ILP variant measured neutral — this simple form stands. */
+__device__ __forceinline__ void k4_copy_group(
+    const uint32_t *src_row, const uint64_t *out_off, const int32_t *in_offsets,
+    const uint8_t *in_bytes, int64_t n, uint8_t *out_bytes, uint32_t *loc, uint32_t *srcb,
+    int lane, int64_t g) {
+    const int64_t s0 = g * WAVE;
+    const int64_t s = s0 + lane;
+    uint32_t len = 0, sb = 0;
+    if (s < n) {
+        const uint32_t r = src_row[s];
+        const int32_t o0 = in_offsets[r];
+        len = (uint32_t)(in_offsets[r + 1] - o0);
+        sb = (uint32_t)o0;
+    }
+    /* exclusive scan of lens across the wave -> group-local byte offsets */
+    uint32_t v = len;
+#pragma unroll
+    for (int d = 1; d < WAVE; d <<= 1) {
+        uint32_t u = (uint32_t)__shfl_up((int)v, d);
+        if (lane >= d) v += u;
+    }
+    const uint32_t T = (uint32_t)__shfl((int)v, WAVE - 1);
+    loc[lane] = v - len; /* exclusive */
+    srcb[lane] = sb;
+    if (lane == 0) loc[WAVE] = T;
+    /* single wave: LDS program order; no barrier */
+    const uint64_t obase = out_off[s0];
+    for (uint32_t p0 = lane * 8; p0 < T; p0 += WAVE * 8) {
+        uint32_t rem = (T - p0 < 8) ? (T - p0) : 8;
+        uint32_t p = p0;
+        /* binary search: largest j with loc[j] <= p */
+        int lo = 0, hi = WAVE;
+        while (lo < hi) {
+            int mid = (lo + hi + 1) >> 1;
+            if (loc[mid] <= p) lo = mid;
+            else hi = mid - 1;
+        }
+        int j = lo;
+        while (rem > 0) {
+            while (loc[j + 1] <= p) j++; /* skip empty strings */
+            const uint32_t within = p - loc[j];
+            const uint32_t avail = loc[j + 1] - p;
+            const uint32_t m = (rem < avail) ? rem : avail;
+            const uint8_t *sp = in_bytes + srcb[j] + within;
+            uint8_t *dp = out_bytes + obase + p;
+            if (m == 8) {
+                uint64_t t;
+                __builtin_memcpy(&t, sp, 8);
+                __builtin_memcpy(dp, &t, 8);
+            } else {
+                for (uint32_t b = 0; b < m; b++) dp[b] = sp[b];
+            }
+            p += m;
+            rem -= m;
+        }
+    }
+}
+
 __global__ __launch_bounds__(BLOCK_THREADS) void k4_copy(
     const uint32_t *src_row, const uint64_t *out_off, const int32_t *in_offsets,
     const uint8_t *in_bytes, int64_t n, uint8_t *out_bytes) {
@@ -1682,93 +1740,64 @@ __global__ __launch_bounds__(BLOCK_THREADS) void k4_copy(
     __shared__ uint32_t srcb_s[WAVES_PER_BLOCK][WAVE];
     const int wid = threadIdx.x / WAVE;
     const int lane = threadIdx.x % WAVE;
-    uint32_t *loc = loc_s[wid];
-    uint32_t *srcb = srcb_s[wid];
     const int64_t ngroups = (n + WAVE - 1) / WAVE;
     const int64_t wave_id = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) / WAVE;
     const int64_t nwaves = ((int64_t)gridDim.x * blockDim.x) / WAVE;
+    for (int64_t g = wave_id; g < ngroups; g += nwaves)
+        k4_copy_group(src_row, out_off, in_offsets, in_bytes, n, out_bytes, loc_s[wid],
+                      srcb_s[wid], lane, g);
+}
 
-    for (int64_t g = wave_id; g < ngroups; g += nwaves) {
-        const int64_t s0 = g * WAVE;
-        const int64_t s = s0 + lane;
-        uint32_t len = 0, sb = 0;
-        if (s < n) {
-            const uint32_t r = src_row[s];
-            const int32_t o0 = in_offsets[r];
-            len = (uint32_t)(in_offsets[r + 1] - o0);
-            sb = (uint32_t)o0;
-        }
-        /* exclusive scan of lens across the wave -> group-local byte offsets */
-        uint32_t v = len;
-#pragma unroll
-        for (int d = 1; d < WAVE; d <<= 1) {
-            uint32_t u = (uint32_t)__shfl_up((int)v, d);
-            if (lane >= d) v += u;
-        }
-        const uint32_t T = (uint32_t)__shfl((int)v, WAVE - 1);
-        loc[lane] = v - len; /* exclusive */
-        srcb[lane] = sb;
-        if (lane == 0) loc[WAVE] = T;
-        /* single wave: LDS program order; no barrier */
-        const uint64_t obase = out_off[s0];
-        /* lane l owns the 8-B chunk at p0 = l*8 (+ stride): consecutive lanes write
-         * consecutive output bytes -> stores fully coalesce (a 32-B-per-lane variant
-         * measured 3.2 -> 5.0 ms: strided stores quarter the write efficiency). ILP:
-         * 4 chunk-iterations are processed together with their loads pre-issued (fast
-         * path: chunk inside one string = 1 unaligned 8-B load; split chunks take the
-         * byte walk) so each lane keeps ~4 random-line loads in flight. */
-        for (uint32_t p0 = lane * 8; p0 < T; p0 += WAVE * 8 * 4) {
-            uint32_t pu[4];
-            int ju[4];
-            uint64_t tu[4];
-            bool fast[4];
-            bool act[4];
-#pragma unroll
-            for (int u = 0; u < 4; u++) {
-                const uint32_t p = p0 + u * WAVE * 8;
-                act[u] = p < T;
-                fast[u] = false;
-                pu[u] = p;
-                if (!act[u]) continue;
-                /* binary search: largest j with loc[j] <= p */
-                int lo = 0, hi = WAVE;
-                while (lo < hi) {
-                    int mid = (lo + hi + 1) >> 1;
-                    if (loc[mid] <= p) lo = mid;
-                    else hi = mid - 1;
-                }
-                ju[u] = lo;
-                const uint32_t avail = loc[lo + 1] - p;
-                if (avail >= 8 && T - p >= 8) {
-                    fast[u] = true;
-                    __builtin_memcpy(&tu[u], in_bytes + srcb[lo] + (p - loc[lo]), 8);
-                }
-            }
-#pragma unroll
-            for (int u = 0; u < 4; u++) {
-                if (!act[u]) continue;
-                uint8_t *dp = out_bytes + obase + pu[u];
-                if (fast[u]) {
-                    __builtin_memcpy(dp, &tu[u], 8);
-                    continue;
-                }
-                uint32_t rem = (T - pu[u] < 8) ? (T - pu[u]) : 8;
-                uint32_t p = pu[u];
-                int j = ju[u];
-                while (rem > 0) {
-                    while (loc[j + 1] <= p) j++; /* skip empty strings */
-                    const uint32_t within = p - loc[j];
-                    const uint32_t avail = loc[j + 1] - p;
-                    const uint32_t m = (rem < avail) ? rem : avail;
-                    const uint8_t *sp = in_bytes + srcb[j] + within;
-                    for (uint32_t b = 0; b < m; b++) dp[b] = sp[b];
-                    dp += m;
-                    p += m;
-                    rem -= m;
-                }
-            }
-        }
+/* MALL-window variant: slot-order processing gives every wave a random source line per
+ * ~10 strings — 20M-row ClickBench measured ~1.3 TB/s effective (latency/fetch-bound).
+ * Here each block owns a (partition, source-window, split) tuple: within a partition the
+ * source rows are MONOTONE over slots, so the groups whose sources lie in window w form
+ * a CONTIGUOUS group range (binary search). All co-resident blocks then gather from the
+ * same ~window_rows source region, which the 256 MB Infinity Cache (MALL) holds — the
+ * random reads become cache hits while the per-partition writes stay coalesced. */
+__global__ __launch_bounds__(BLOCK_THREADS) void k4_copy_win(
+    const uint32_t *src_row, const uint64_t *out_off, const int32_t *in_offsets,
+    const uint8_t *in_bytes, int64_t n, uint8_t *out_bytes,
+    const uint64_t *part_offsets /* [P+1] slot offsets */, uint32_t nparts,
+    int64_t win_rows, int nwin, int nsplit) {
+    __shared__ uint32_t loc_s[WAVES_PER_BLOCK][WAVE + 1];
+    __shared__ uint32_t srcb_s[WAVES_PER_BLOCK][WAVE];
+    const int wid = threadIdx.x / WAVE;
+    const int lane = threadIdx.x % WAVE;
+    const int64_t bid = blockIdx.x;
+    const int split = (int)(bid % nsplit);
+    const int w = (int)((bid / nsplit) % nwin);
+    const uint32_t p = (uint32_t)(bid / ((int64_t)nsplit * nwin));
+    if (p >= nparts) return;
+    /* group range of partition p: group g belongs to p iff its first slot g*64 is in
+     * [po[p], po[p+1]) */
+    const int64_t g0p = ((int64_t)part_offsets[p] + WAVE - 1) / WAVE;
+    const int64_t g1p = ((int64_t)part_offsets[p + 1] + WAVE - 1) / WAVE;
+    /* sub-range whose first-slot source rows lie in [w*win_rows, (w+1)*win_rows):
+     * src_row[g*64] is increasing in g within the partition */
+    const uint32_t wlo = (uint32_t)((int64_t)w * win_rows);
+    const uint64_t whi64 = (int64_t)(w + 1) * win_rows;
+    int64_t lo = g0p, hi = g1p; /* first g with src >= wlo */
+    while (lo < hi) {
+        int64_t mid = (lo + hi) >> 1;
+        if (src_row[mid * WAVE] < wlo) lo = mid + 1;
+        else hi = mid;
     }
+    const int64_t ga = lo;
+    hi = g1p; /* first g with src >= whi */
+    while (lo < hi) {
+        int64_t mid = (lo + hi) >> 1;
+        if ((uint64_t)src_row[mid * WAVE] < whi64) lo = mid + 1;
+        else hi = mid;
+    }
+    const int64_t gb = lo;
+    /* split the range across nsplit blocks; waves of this block stride it */
+    const int64_t cnt = gb - ga;
+    const int64_t c0 = ga + cnt * split / nsplit;
+    const int64_t c1 = ga + cnt * (split + 1) / nsplit;
+    for (int64_t g = c0 + wid; g < c1; g += WAVES_PER_BLOCK)
+        k4_copy_group(src_row, out_off, in_offsets, in_bytes, n, out_bytes, loc_s[wid],
+                      srcb_s[wid], lane, g);
 }
 
 __global__ void k_off64_to_off32(const uint64_t *off64, int64_t lo, int64_t n,
@@ -1833,14 +1862,36 @@ hipError_t dd_launch_var_bytes(const uint32_t *lens, const uint32_t *src_row,
                        total_bytes, out_off, n);
     hipLaunchKernelGGL(k4_off_rewrite, dim3(wavegrid), dim3(256), 0, s, lens, n, partials,
                        out_off);
-    int copy_blocks = (int)((n + 255) / 256);
-    if (copy_blocks > 32768) copy_blocks = 32768; /* gather-copy is latency-bound: deep
-                                                     oversubscription hides the random
-                                                     string reads (8192 -> 32768 measured
-                                                     with the 32-B lane chunks) */
-    if (copy_blocks < 1) copy_blocks = 1;
-    hipLaunchKernelGGL(k4_copy, dim3(copy_blocks), dim3(256), 0, s, src_row, out_off,
-                       in_offsets, in_bytes, n, out_bytes);
+    /* MALL-window scheduling (k4_copy_win header): on when the var payload is
+     * window-partitionable into >1 ~160 MB source windows; DD_K4_WIN=0 forces the
+     * slot-order form, DD_K4_WIN_MB overrides the window size */
+    int64_t win_mb = 160;
+    if (const char *e = getenv("DD_K4_WIN_MB")) {
+        int64_t v = atoll(e);
+        if (v >= 16 && v <= 4096) win_mb = v;
+    }
+    const bool use_win = !(getenv("DD_K4_WIN") && atoi(getenv("DD_K4_WIN")) == 0) &&
+                         part_offsets != nullptr && n > 0 && total_bytes > 0;
+    int nwin = (int)((total_bytes + win_mb * 1048576 - 1) / (win_mb * 1048576));
+    if (nwin < 1) nwin = 1;
+    if (use_win && nwin > 1) {
+        const int64_t win_rows = (n + nwin - 1) / nwin;
+        int nsplit = (int)(16384 / ((int64_t)nwin * nparts));
+        if (nsplit < 1) nsplit = 1;
+        if (nsplit > 64) nsplit = 64;
+        const int64_t blocks = (int64_t)nparts * nwin * nsplit;
+        hipLaunchKernelGGL(k4_copy_win, dim3((unsigned)blocks), dim3(256), 0, s, src_row,
+                           out_off, in_offsets, in_bytes, n, out_bytes, part_offsets,
+                           nparts, win_rows, nwin, nsplit);
+    } else {
+        int copy_blocks = (int)((n + 255) / 256);
+        if (copy_blocks > 8192) copy_blocks = 8192; /* latency-bound: deep
+                                                       oversubscription hides the random
+                                                       string reads */
+        if (copy_blocks < 1) copy_blocks = 1;
+        hipLaunchKernelGGL(k4_copy, dim3(copy_blocks), dim3(256), 0, s, src_row, out_off,
+                           in_offsets, in_bytes, n, out_bytes);
+    }
     hipLaunchKernelGGL(k4_part_boffsets, dim3((nparts + 256) / 256 + 1), dim3(256), 0, s,
                        out_off, part_offsets, nparts, n, total_bytes, part_boffsets);
     return hipGetLastError();
