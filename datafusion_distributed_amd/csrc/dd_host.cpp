@@ -142,6 +142,8 @@ struct dd_partitioner {
     uint16_t *imgb = nullptr;            /* pre: [nrounds][sP2] round image bases (roff) */
     uint32_t *partials2 = nullptr;       /* pre: second-level scan partials [64][P] */
     uint32_t *src_row = nullptr;         /* staged-var: permutation out[slot] = input row */
+    uint32_t *k4w_meta = nullptr;        /* staged-var: window hist/base (k4_copy_ord) */
+    uint32_t *k4w_order = nullptr;       /* staged-var: window-bucketed group order */
     uint64_t *out_off[DD_KMAX_VAR] = {}; /* staged-var: Arrow byte offsets [n+1] per var */
     uint64_t *k4_partials = nullptr;     /* staged-var scan scratch */
     uint64_t *dict_hashes[DD_KMAX_COLS] = {};
@@ -163,6 +165,8 @@ struct dd_partitioner {
         (void)hipFree(bpartials);
         (void)hipFree(part_boffsets);
         (void)hipFree(src_row);
+        (void)hipFree(k4w_meta);
+        (void)hipFree(k4w_order);
         for (auto &o : out_off) hipFree(o);
         (void)hipFree(k4_partials);
         for (int i = 0; i < DD_KMAX_COLS; i++) {
@@ -486,7 +490,9 @@ extern "C" dd_status dd_partitioner_create(const dd_batch_desc *batch, const int
         ok = halloc((void **)&p->part_boffsets, (size_t)nvar * (P + 1) * 8);
         if (p->staged) {
             ok = ok && halloc((void **)&p->src_row, (size_t)n * 4) &&
-                 halloc((void **)&p->k4_partials, (size_t)8192 * 8);
+                 halloc((void **)&p->k4_partials, (size_t)8192 * 8) &&
+                 halloc((void **)&p->k4w_meta, (64 + 65) * 4) &&
+                 halloc((void **)&p->k4w_order, ((size_t)(n + 63) / 64) * 4);
             for (int v = 0; v < nvar && ok; v++)
                 ok = halloc((void **)&p->out_off[v], (size_t)(n + 1) * 8);
         } else {
@@ -634,7 +640,8 @@ extern "C" dd_status dd_partitioner_run_phase2(dd_partitioner *p, void *stream) 
                 p->out_lengths[ci], p->src_row, cd.offsets, (const uint8_t *)cd.data,
                 p->ka.n_rows, cd.data_len, p->k4_partials, p->out_off[v],
                 (uint8_t *)p->out_data[ci], p->part_offsets, p->nparts,
-                p->part_boffsets + (size_t)v * (p->nparts + 1), s));
+                p->part_boffsets + (size_t)v * (p->nparts + 1), p->k4w_meta,
+                p->k4w_order, s));
         }
     } else {
         HIP_TRY(dd_launch_scatter(&p->ka, p->nchunks, p->chunk_rows, p->nparts, p->nbits,

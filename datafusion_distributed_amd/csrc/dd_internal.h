@@ -109,7 +109,8 @@ hipError_t dd_launch_var_bytes(const uint32_t *lens, const uint32_t *src_row,
                                int64_t n, int64_t total_bytes, uint64_t *partials,
                                uint64_t *out_off, uint8_t *out_bytes,
                                const uint64_t *part_offsets, uint32_t nparts,
-                               uint64_t *part_boffsets, hipStream_t s);
+                               uint64_t *part_boffsets, uint32_t *k4w_meta,
+                               uint32_t *k4w_order, hipStream_t s);
 hipError_t dd_launch_partial_reduce(const dd_kargs *a, int64_t nblocks, int64_t chunk_rows,
                                     int n_aggs, const int32_t *agg_cols,
                                     const int32_t *agg_ops, uint64_t *out_keys,
