@@ -1748,60 +1748,6 @@ __global__ __launch_bounds__(BLOCK_THREADS) void k4_copy(
                       srcb_s[wid], lane, g);
 }
 
-/* MALL-window variant: slot-order processing gives every wave a random source line per
- * ~10 strings — 20M-row ClickBench measured ~1.3 TB/s effective (latency/fetch-bound).
- * Windowing by (partition, window) BLOCKS first (binary-searched contiguous group
- * ranges) measured 5.1 vs 3.3 ms: Zipf-skewed partitions make block workloads differ by
- * >10x. Instead the groups are BUCKETED BY SOURCE WINDOW into a flat order array
- * (k4w_count / k4w_scatter below) and grid-strided — perfect balance, and all
- * co-resident waves still gather from one ~window of source bytes, which the 256 MB
- * Infinity Cache (MALL) holds, so the random string reads become cache hits while the
- * writes stay coalesced. */
-__global__ void k4w_count(const uint32_t *src_row, int64_t ngroups, int64_t n,
-                          int64_t win_rows, uint32_t *hist /* [nwin] zeroed */) {
-    const int64_t g = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
-    if (g >= ngroups) return;
-    const int64_t s = g * WAVE < n ? g * WAVE : n - 1;
-    atomicAdd(&hist[src_row[s] / win_rows], 1u);
-}
-
-__global__ void k4w_scan(uint32_t *hist, int nwin, uint32_t *base /* [nwin+1] */) {
-    if (threadIdx.x == 0) { /* nwin <= ~64: serial scan is trivial */
-        uint32_t run = 0;
-        for (int w = 0; w < nwin; w++) {
-            base[w] = run;
-            run += hist[w];
-            hist[w] = 0; /* reuse as the scatter cursor */
-        }
-        base[nwin] = run;
-    }
-}
-
-__global__ void k4w_scatter(const uint32_t *src_row, int64_t ngroups, int64_t n,
-                            int64_t win_rows, const uint32_t *base, uint32_t *cursor,
-                            uint32_t *order) {
-    const int64_t g = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
-    if (g >= ngroups) return;
-    const int64_t s = g * WAVE < n ? g * WAVE : n - 1;
-    const int64_t w = src_row[s] / win_rows;
-    order[base[w] + atomicAdd(&cursor[w], 1u)] = (uint32_t)g;
-}
-
-__global__ __launch_bounds__(BLOCK_THREADS) void k4_copy_ord(
-    const uint32_t *src_row, const uint64_t *out_off, const int32_t *in_offsets,
-    const uint8_t *in_bytes, int64_t n, uint8_t *out_bytes, const uint32_t *order,
-    int64_t ngroups) {
-    __shared__ uint32_t loc_s[WAVES_PER_BLOCK][WAVE + 1];
-    __shared__ uint32_t srcb_s[WAVES_PER_BLOCK][WAVE];
-    const int wid = threadIdx.x / WAVE;
-    const int lane = threadIdx.x % WAVE;
-    const int64_t wave_id = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) / WAVE;
-    const int64_t nwaves = ((int64_t)gridDim.x * blockDim.x) / WAVE;
-    for (int64_t i = wave_id; i < ngroups; i += nwaves)
-        k4_copy_group(src_row, out_off, in_offsets, in_bytes, n, out_bytes, loc_s[wid],
-                      srcb_s[wid], lane, (int64_t)order[i]);
-}
-
 __global__ void k_off64_to_off32(const uint64_t *off64, int64_t lo, int64_t n,
                                  int32_t *out32) {
     const int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
@@ -1865,47 +1811,27 @@ hipError_t dd_launch_var_bytes(const uint32_t *lens, const uint32_t *src_row,
                        total_bytes, out_off, n);
     hipLaunchKernelGGL(k4_off_rewrite, dim3(wavegrid), dim3(256), 0, s, lens, n, partials,
                        out_off);
-    /* MALL-window scheduling (k4_copy_ord header): on when the var payload spans >1
-     * ~160 MB source window; DD_K4_WIN=0 forces the slot-order form, DD_K4_WIN_MB
-     * overrides the window size. Workspace: k4w (hist/base/cursor + order array). */
-    int64_t win_mb = 160;
-    if (const char *e = getenv("DD_K4_WIN_MB")) {
-        int64_t v = atoll(e);
-        if (v >= 16 && v <= 4096) win_mb = v;
+    /* Slot-order gather (k4_copy). MEASURED NEGATIVES (kept out of the tree, see
+     * DESIGN.md §13): (a) 32 B-per-lane chunks — strided stores quarter write
+     * coalescing (3.3 -> 5.0 ms); (b) 4-deep pre-issued-load ILP — neutral; (c) MALL
+     * source-windowing, both as per-(partition,window) blocks (Zipf imbalance, 5.1 ms)
+     * and as a balanced window-bucketed group order (randomized group-level write
+     * order, 10 ms): the write side loses more than the read side gains.
+     * DD_K4_BLOCKS overrides the grid cap (latency-hiding depth experiment). */
+    int copy_blocks = (int)((n + 255) / 256);
+    int cap = 8192;
+    if (const char *e = getenv("DD_K4_BLOCKS")) {
+        int v = atoi(e);
+        if (v >= 256 && v <= 65536) cap = v;
     }
-    const bool use_win = !(getenv("DD_K4_WIN") && atoi(getenv("DD_K4_WIN")) == 0) &&
-                         k4w_order != nullptr && n > 0 && total_bytes > 0;
-    int nwin = (int)((total_bytes + win_mb * 1048576 - 1) / (win_mb * 1048576));
-    if (nwin < 1) nwin = 1;
-    if (nwin > 64) nwin = 64;
-    const int64_t ngroups = (n + WAVE - 1) / WAVE;
-    if (use_win && nwin > 1) {
-        const int64_t win_rows = (n + nwin - 1) / nwin;
-        uint32_t *hist = k4w_meta;              /* [64] */
-        uint32_t *base = k4w_meta + 64;         /* [65] */
-        hipError_t em = hipMemsetAsync(hist, 0, 64 * 4, s);
-        if (em != hipSuccess) return em;
-        int gb = (int)((ngroups + 255) / 256);
-        hipLaunchKernelGGL(k4w_count, dim3(gb), dim3(256), 0, s, src_row, ngroups, n,
-                           win_rows, hist);
-        hipLaunchKernelGGL(k4w_scan, dim3(1), dim3(64), 0, s, hist, nwin, base);
-        hipLaunchKernelGGL(k4w_scatter, dim3(gb), dim3(256), 0, s, src_row, ngroups, n,
-                           win_rows, base, hist, k4w_order);
-        int copy_blocks = (int)((ngroups * WAVE + 255) / 256);
-        if (copy_blocks > 8192) copy_blocks = 8192;
-        if (copy_blocks < 1) copy_blocks = 1;
-        hipLaunchKernelGGL(k4_copy_ord, dim3(copy_blocks), dim3(256), 0, s, src_row,
-                           out_off, in_offsets, in_bytes, n, out_bytes, k4w_order,
-                           ngroups);
-    } else {
-        int copy_blocks = (int)((n + 255) / 256);
-        if (copy_blocks > 8192) copy_blocks = 8192; /* latency-bound: deep
-                                                       oversubscription hides the random
-                                                       string reads */
-        if (copy_blocks < 1) copy_blocks = 1;
-        hipLaunchKernelGGL(k4_copy, dim3(copy_blocks), dim3(256), 0, s, src_row, out_off,
-                           in_offsets, in_bytes, n, out_bytes);
-    }
+    if (copy_blocks > cap) copy_blocks = cap; /* latency-bound: deep oversubscription
+                                                 hides the random string reads */
+    if (copy_blocks < 1) copy_blocks = 1;
+    hipLaunchKernelGGL(k4_copy, dim3(copy_blocks), dim3(256), 0, s, src_row, out_off,
+                       in_offsets, in_bytes, n, out_bytes);
+    (void)k4w_meta;
+    (void)k4w_order;
+    (void)total_bytes;
     hipLaunchKernelGGL(k4_part_boffsets, dim3((nparts + 256) / 256 + 1), dim3(256), 0, s,
                        out_off, part_offsets, nparts, n, total_bytes, part_boffsets);
     return hipGetLastError();
