@@ -33,6 +33,11 @@ static dd_status set_err(dd_status s, const std::string &msg) {
 }
 
 extern "C" const char *dd_last_error(void) { return g_last_error.c_str(); }
+/* shared with dd_proto.cpp (declared in dd_internal.h) */
+extern "C" dd_status dd_set_error(dd_status s, const char *msg) {
+    g_last_error = msg;
+    return s;
+}
 extern "C" const char *dd_version(void) { return "dd_shuffle 0.1 (gfx950)"; }
 
 #define HIP_TRY(expr)                                                                        \

@@ -65,7 +65,10 @@ struct dd_kargs {
     dd_kcol cols[DD_KMAX_COLS];
 };
 
+#include "dd_shuffle.h" /* dd_status for dd_set_error */
+
 extern "C" {
+dd_status dd_set_error(dd_status s, const char *msg); /* thread-local dd_last_error */
 hipError_t dd_launch_dict_hashes(const uint8_t *bytes, const int32_t *offsets, int64_t n,
                                  uint64_t *out, hipStream_t s);
 hipError_t dd_launch_hash_count(const dd_kargs *a, int64_t nchunks, int64_t chunk_rows,

@@ -35,16 +35,7 @@
 #define R_CAP 1024      /* table slots per block */
 #define R_PROBE 64      /* bounded probe; then spill */
 
-/* agg op codes (mirrored in include/dd_shuffle.h) */
-enum {
-    DD_AGG_SUM_F64 = 0,
-    DD_AGG_COUNT = 1,
-    DD_AGG_SUM_I64 = 2,
-    DD_AGG_MIN_F64 = 3,
-    DD_AGG_MAX_F64 = 4,
-    DD_AGG_MIN_I64 = 5,
-    DD_AGG_MAX_I64 = 6,
-};
+/* agg op codes come from include/dd_shuffle.h (via dd_internal.h) */
 
 /* order-preserving f64 <-> u64 map (IEEE total order; NaN sorts above +inf, matching
  * Arrow's max semantics): x >= 0 -> bits | sign, x < 0 -> ~bits */

@@ -236,6 +236,42 @@ dd_status dd_execute_task(const dd_task_key *key, uint32_t part_lo, uint32_t par
                           void *stream, dd_partitioner **out);
 dd_status dd_drop_task(const dd_task_key *key); /* task cleanup (stateful_data_cleanup) */
 
+/* ---------------- protobuf plan/stage wire payload (dd_proto.cpp) ----------------
+ * The reference's own wire shapes at this boundary (field numbers cited in
+ * dd_proto.cpp): SetPlanRequest / ExecuteTaskRequest / TaskKey
+ * (src/protocol/grpc/worker.proto:84-110,134-155,171-179) with the producer_head's
+ * RepartitionExecHead carrying a datafusion-proto `Partitioning` message
+ * (worker.proto:167-170; datafusion-proto 55.0.0, Cargo.lock pin). A Rust shim passes
+ * the prost-encoded request bytes straight through — no re-encoding. */
+
+enum { DD_HEAD_NONE = 0, DD_HEAD_BROADCAST = 1, DD_HEAD_REPARTITION = 2 };
+
+/* decode a datafusion-proto Partitioning (hash exprs must be columns) */
+dd_status dd_decode_partitioning(const uint8_t *buf, int64_t len, int32_t *key_cols,
+                                 int32_t max_keys, int32_t *n_keys,
+                                 uint32_t *n_partitions);
+/* decode an ExecuteTaskRequest: task key, partition range, producer head (repartition
+ * heads also fill key_cols/n_keys/n_partitions; broadcast fills n_partitions) */
+dd_status dd_decode_execute_task(const uint8_t *buf, int64_t len, dd_task_key *key,
+                                 uint64_t *part_start, uint64_t *part_end,
+                                 int32_t *head_kind, int32_t *key_cols, int32_t max_keys,
+                                 int32_t *n_keys, uint32_t *n_partitions);
+/* decode a SetPlanRequest; *plan_proto is a borrowed view into buf */
+dd_status dd_decode_set_plan(const uint8_t *buf, int64_t len, dd_task_key *key,
+                             uint64_t *task_count, const uint8_t **plan_proto,
+                             int64_t *plan_len);
+/* register a plan under the TaskKey decoded from a prost-encoded SetPlanRequest; the
+ * device batch is what the subplan below the network boundary produces (the shim
+ * materializes it — the plan_proto subplan itself stays host-side per the tier) */
+dd_status dd_set_plan_proto(const uint8_t *set_plan_pb, int64_t len,
+                            const dd_batch_desc *batch);
+/* execute from a prost-encoded ExecuteTaskRequest: decodes the RepartitionExecHead,
+ * lazily inserts the head (TaskData::plan / ProducerHead::insert semantics) and runs
+ * the partition kernels; same returned-pointer lifetime as dd_execute_task */
+dd_status dd_execute_task_proto(const uint8_t *execute_task_pb, int64_t len, void *stream,
+                                dd_partitioner **out);
+dd_status dd_drop_task_proto(const uint8_t *task_key_pb, int64_t len);
+
 /* ---------------- partial aggregation (below the shuffle) ----------------
  * Mirrors the partial-reduce pass (src/distributed_planner/
  * partial_reduce_below_network_shuffles.rs; `distributed.partial_reduce`,
