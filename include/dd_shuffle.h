@@ -272,6 +272,52 @@ dd_status dd_execute_task_proto(const uint8_t *execute_task_pb, int64_t len, voi
                                 dd_partitioner **out);
 dd_status dd_drop_task_proto(const uint8_t *task_key_pb, int64_t len);
 
+/* ---------------- Arrow IPC + lz4 wire format (dd_wire.cpp) ----------------
+ * The on-wire batch encoding for cross-node (non-xGMI) hops, reimplementing the
+ * reference's Flight encode/decode (src/protocol/grpc/worker_service.rs:363-433 /
+ * worker_client.rs:302): Apache Arrow IPC STREAMING format, MetadataVersion V5, with
+ * per-buffer lz4-frame BodyCompression (the reference's default,
+ * distributed_config.rs:36-38). Zero-column batches (row count, no fields) round-trip —
+ * the edge case the reference wire-tests pin (tests/empty_columns_between_workers.rs).
+ * Validity and bool data are unpacked u8 at this ABI (packed to Arrow bitmaps on the
+ * wire). Host memory only; this is the serialization layer, not the device path. */
+
+typedef struct dd_ipc_field {
+    int32_t dtype;    /* dd_dtype (DICT32 unsupported on the wire: indices + values
+                         cross via arrow_boundary materialization) */
+    const char *name; /* optional */
+    int32_t nullable;
+} dd_ipc_field;
+
+typedef struct dd_ipc_array {
+    const void *data;        /* fixed: values (bool: unpacked u8); utf8: byte buffer */
+    int64_t data_len;        /* utf8: byte length; reader fills for fixed too */
+    const uint8_t *validity; /* unpacked u8, NULL = all valid */
+    int64_t null_count;
+    const int32_t *offsets;  /* utf8: [n_rows+1] */
+} dd_ipc_array;
+
+typedef struct dd_ipc_writer dd_ipc_writer;
+typedef struct dd_ipc_reader dd_ipc_reader;
+
+dd_status dd_ipc_writer_create(const dd_ipc_field *fields, int32_t n_fields,
+                               int32_t use_lz4, dd_ipc_writer **out);
+dd_status dd_ipc_writer_batch(dd_ipc_writer *w, int64_t n_rows, const dd_ipc_array *cols);
+/* appends the EOS marker (idempotent) and exposes the stream; owned by the writer */
+dd_status dd_ipc_writer_finish(dd_ipc_writer *w, const uint8_t **data, int64_t *len);
+void dd_ipc_writer_destroy(dd_ipc_writer *w);
+
+dd_status dd_ipc_reader_create(const uint8_t *data, int64_t len, dd_ipc_reader **out);
+int32_t dd_ipc_reader_n_fields(const dd_ipc_reader *r);
+int32_t dd_ipc_reader_n_batches(const dd_ipc_reader *r);
+int32_t dd_ipc_reader_field_dtype(const dd_ipc_reader *r, int32_t i);
+const char *dd_ipc_reader_field_name(const dd_ipc_reader *r, int32_t i);
+int64_t dd_ipc_reader_batch_rows(const dd_ipc_reader *r, int32_t b);
+/* pointers into reader-owned memory (decompressed; validity/bool unpacked to u8) */
+dd_status dd_ipc_reader_batch_col(const dd_ipc_reader *r, int32_t b, int32_t c,
+                                  dd_ipc_array *out);
+void dd_ipc_reader_destroy(dd_ipc_reader *r);
+
 /* ---------------- partial aggregation (below the shuffle) ----------------
  * Mirrors the partial-reduce pass (src/distributed_planner/
  * partial_reduce_below_network_shuffles.rs; `distributed.partial_reduce`,
