@@ -9,6 +9,7 @@ reference's correctness suites pin (tests/tpch_correctness_test.rs:139-158)."""
 
 import numpy as np
 import pytest
+import pyarrow.compute as pc
 
 from datafusion_distributed_amd import api
 
@@ -43,15 +44,15 @@ def direct_answer(pa, cust, orders, li):
     tc = pa.table(cust)
     to = pa.table(orders)
     tl = pa.table(li)
-    tc = tc.filter(pa.compute.equal(tc["c_mktsegment"], 0))
-    to = to.filter(pa.compute.less(to["o_orderdate"], CUTOFF))
-    tl = tl.filter(pa.compute.greater(tl["l_shipdate"], CUTOFF))
+    tc = tc.filter(pc.equal(tc["c_mktsegment"], 0))
+    to = to.filter(pc.less(to["o_orderdate"], CUTOFF))
+    tl = tl.filter(pc.greater(tl["l_shipdate"], CUTOFF))
     oj = to.join(tc.select(["c_custkey"]), keys="o_custkey", right_keys="c_custkey",
                  join_type="inner")
     lj = tl.join(oj.select(["o_orderkey", "o_orderdate", "o_shippriority"]),
                  keys="l_orderkey", right_keys="o_orderkey", join_type="inner")
-    rev = pa.compute.multiply(lj["l_extendedprice"],
-                              pa.compute.subtract(pa.scalar(1.0), lj["l_discount"]))
+    rev = pc.multiply(lj["l_extendedprice"],
+                              pc.subtract(pa.scalar(1.0), lj["l_discount"]))
     lj = lj.append_column("revenue", rev)
     agg = lj.group_by(["l_orderkey", "o_orderdate", "o_shippriority"]).aggregate(
         [("revenue", "sum")])
@@ -82,7 +83,7 @@ def test_q3_join_pipeline_matches_direct_plan():
     # side (the join above BroadcastExec, host-side), then GPU-shuffled on o_orderkey.
     omask = orders["o_orderdate"] < CUTOFF
     of = {k: v[omask] for k, v in orders.items()}
-    isin = pa.compute.is_in(pa.array(of["o_custkey"]), value_set=pa.array(build_set))
+    isin = pc.is_in(pa.array(of["o_custkey"]), value_set=pa.array(build_set))
     omask2 = np.asarray(isin)
     oj = {k: v[omask2] for k, v in of.items()}
     obatch = api.DeviceBatch([
@@ -132,8 +133,8 @@ def test_q3_join_pipeline_matches_direct_plan():
         if j.num_rows == 0:
             continue
         total_joined += j.num_rows
-        rev = pa.compute.multiply(j["l_extendedprice"],
-                                  pa.compute.subtract(pa.scalar(1.0), j["l_discount"]))
+        rev = pc.multiply(j["l_extendedprice"],
+                                  pc.subtract(pa.scalar(1.0), j["l_discount"]))
         j = j.append_column("revenue", rev)
         agg = j.group_by(["l_orderkey", "o_orderdate", "o_shippriority"]).aggregate(
             [("revenue", "sum")])
