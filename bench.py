@@ -376,6 +376,7 @@ def main():
         ex = comm.exchange(part)
         ms, egress = ex.stats()
         config["xgmi_egress_GBps"] = round(egress / (ms / 1e3) / 1e9, 2) if ms > 0 else None
+        config["data_plane"] = getattr(comm, "data_plane", "rccl/xgmi")
         ex.destroy()
 
     line = {
