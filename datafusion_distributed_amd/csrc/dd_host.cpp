@@ -149,6 +149,14 @@ struct dd_partitioner {
     uint32_t *src_row = nullptr;         /* staged-var: permutation out[slot] = input row */
     uint32_t *k4w_meta = nullptr;        /* staged-var: window hist/base (k4_copy_ord) */
     uint32_t *k4w_order = nullptr;       /* staged-var: window-bucketed group order */
+    int k5 = 0;                          /* LDS-staged var-byte scatter (small strings) */
+    int64_t k5_nrounds = 0, k5_nseg = 0;
+    uint32_t k5_maxlen = 0;
+    size_t lds_k5 = 0, lds_k5c = 0;
+    uint32_t *k5_bcounts = nullptr;      /* [nseg5][P] */
+    uint32_t *k5_partials = nullptr;     /* [2048][P] */
+    uint32_t *k5_partials2 = nullptr;    /* [64][P] */
+    uint32_t *k5_roffB = nullptr;        /* [nrounds5][P+1] */
     uint64_t *out_off[DD_KMAX_VAR] = {}; /* staged-var: Arrow byte offsets [n+1] per var */
     uint64_t *k4_partials = nullptr;     /* staged-var scan scratch */
     uint64_t *dict_hashes[DD_KMAX_COLS] = {};
@@ -172,6 +180,10 @@ struct dd_partitioner {
         (void)hipFree(src_row);
         (void)hipFree(k4w_meta);
         (void)hipFree(k4w_order);
+        (void)hipFree(k5_bcounts);
+        (void)hipFree(k5_partials);
+        (void)hipFree(k5_partials2);
+        (void)hipFree(k5_roffB);
         for (auto &o : out_off) hipFree(o);
         (void)hipFree(k4_partials);
         for (int i = 0; i < DD_KMAX_COLS; i++) {
@@ -500,6 +512,42 @@ extern "C" dd_status dd_partitioner_create(const dd_batch_desc *batch, const int
                  halloc((void **)&p->k4w_order, ((size_t)(n + 63) / 64) * 4);
             for (int v = 0; v < nvar && ok; v++)
                 ok = halloc((void **)&p->out_off[v], (size_t)(n + 1) * 8);
+            /* K5 (LDS-staged var-byte scatter, dd_kernels.hip K5 header): one var
+             * column with small strings. Needs the real max string length -> one tiny
+             * create-time kernel + 4 B d2h (amortized across runs). DD_K5=0 reverts
+             * to the K4 gather. */
+            if (ok && nvar == 1 && n > 0 &&
+                !(getenv("DD_K5") && atoi(getenv("DD_K5")) == 0)) {
+                const dd_col_desc &vc = batch->cols[ka.var_idx[0]];
+                uint32_t *mdev = nullptr;
+                if (hipMalloc((void **)&mdev, 4) == hipSuccess) {
+                    uint32_t maxlen = 0;
+                    bool mok = hipMemset(mdev, 0, 4) == hipSuccess &&
+                               dd_launch_k5_maxlen(vc.offsets, n, mdev, nullptr) ==
+                                   hipSuccess &&
+                               hipMemcpy(&maxlen, mdev, 4, hipMemcpyDeviceToHost) ==
+                                   hipSuccess;
+                    (void)hipFree(mdev);
+                    const int64_t R5 = 16 * 64;
+                    const size_t img = (size_t)R5 * maxlen + 8;
+                    const size_t lds5 =
+                        (size_t)16 * P * 4 + (size_t)(P + 1) * 4 + img;
+                    if (mok && maxlen > 0 && maxlen <= 128 && lds5 <= 163840) {
+                        p->k5_nrounds = (n + R5 - 1) / R5;
+                        p->k5_nseg = p->k5_nrounds * 16;
+                        p->k5_maxlen = maxlen;
+                        p->lds_k5 = lds5;
+                        p->lds_k5c = (size_t)WAVES_PER_BLOCK_H * P * 4;
+                        ok = halloc((void **)&p->k5_bcounts,
+                                    (size_t)p->k5_nseg * P * 4) &&
+                             halloc((void **)&p->k5_partials, (size_t)2048 * P * 4) &&
+                             halloc((void **)&p->k5_partials2, (size_t)64 * P * 4) &&
+                             halloc((void **)&p->k5_roffB,
+                                    (size_t)p->k5_nrounds * (P + 1) * 4);
+                        if (ok) p->k5 = 1;
+                    }
+                }
+            }
         } else {
             ok = ok && halloc((void **)&p->bcounts, (size_t)nvar * nchunks * P * 4) &&
                  halloc((void **)&p->bpartials, (size_t)nvar * DD_SCAN_RANGES * P * 4);
@@ -613,6 +661,18 @@ extern "C" dd_status dd_partitioner_run_phase1(dd_partitioner *p, void *stream) 
         HIP_TRY(dd_launch_round_layout(p->counts, p->part_offsets, p->nrounds, p->wpb,
                                        p->nparts, p->sP2, p->imgb, s));
     }
+    if (p->k5) { /* byte-base precompute for the K5 var scatter */
+        const int ci = p->ka.var_idx[0];
+        HIP_TRY(hipMemsetAsync(p->k5_partials, 0, (size_t)2048 * p->nparts * 4, s));
+        HIP_TRY(dd_launch_k5_count(p->ka.n_rows, p->nparts, p->pid,
+                                   p->batch.cols[ci].offsets, p->k5_bcounts,
+                                   p->k5_partials, 2048, p->k5_nseg, p->lds_k5c, s));
+        HIP_TRY(dd_launch_scan_deep(p->k5_bcounts, p->k5_nseg, p->nparts, 2048, 64,
+                                    p->k5_partials, p->k5_partials2, p->part_boffsets,
+                                    1, s));
+        HIP_TRY(dd_launch_k5_roff(p->k5_bcounts, p->part_boffsets, p->k5_nrounds, 16,
+                                  p->nparts, p->k5_roffB, s));
+    }
     if (!p->staged) {
         for (int v = 0; v < p->ka.n_var; v++) {
             HIP_TRY(dd_launch_scan(p->bcounts + (size_t)v * p->nchunks * p->nparts,
@@ -646,7 +706,13 @@ extern "C" dd_status dd_partitioner_run_phase2(dd_partitioner *p, void *stream) 
                 p->ka.n_rows, cd.data_len, p->k4_partials, p->out_off[v],
                 (uint8_t *)p->out_data[ci], p->part_offsets, p->nparts,
                 p->part_boffsets + (size_t)v * (p->nparts + 1), p->k4w_meta,
-                p->k4w_order, s));
+                p->k4w_order, p->k5 ? 1 : 0, s));
+            if (p->k5) {
+                HIP_TRY(dd_launch_k5_scatter(
+                    p->ka.n_rows, p->nparts, p->nbits, p->pid, cd.offsets,
+                    (const uint8_t *)cd.data, p->k5_bcounts, p->k5_roffB, p->k5_nrounds,
+                    (uint8_t *)p->out_data[ci], p->lds_k5, s));
+            }
         }
     } else {
         HIP_TRY(dd_launch_scatter(&p->ka, p->nchunks, p->chunk_rows, p->nparts, p->nbits,

@@ -71,6 +71,20 @@ extern "C" {
 dd_status dd_set_error(dd_status s, const char *msg); /* thread-local dd_last_error */
 hipError_t dd_launch_dict_hashes(const uint8_t *bytes, const int32_t *offsets, int64_t n,
                                  uint64_t *out, hipStream_t s);
+hipError_t dd_launch_k5_maxlen(const int32_t *offsets, int64_t n, uint32_t *out_max,
+                               hipStream_t s);
+hipError_t dd_launch_k5_count(int64_t n, uint32_t nparts, const uint32_t *pid,
+                              const int32_t *offsets, uint32_t *bcounts,
+                              uint32_t *partials, int nranges, int64_t nseg5,
+                              size_t lds_bytes, hipStream_t s);
+hipError_t dd_launch_k5_roff(const uint32_t *bcounts, const uint64_t *part_boffsets,
+                             int64_t nrounds, int wpb, uint32_t nparts, uint32_t *roffB,
+                             hipStream_t s);
+hipError_t dd_launch_k5_scatter(int64_t n, uint32_t nparts, int nbits, const uint32_t *pid,
+                                const int32_t *offsets, const uint8_t *in_bytes,
+                                const uint32_t *gbaseB, const uint32_t *roffB,
+                                int64_t nrounds, uint8_t *out_bytes, size_t lds_bytes,
+                                hipStream_t s);
 hipError_t dd_launch_hash_count(const dd_kargs *a, int64_t nchunks, int64_t chunk_rows,
                                 uint32_t nparts, int nbits, uint32_t *pid_out,
                                 uint32_t *counts, uint32_t *bcounts, size_t lds_bytes,
@@ -113,7 +127,7 @@ hipError_t dd_launch_var_bytes(const uint32_t *lens, const uint32_t *src_row,
                                uint64_t *out_off, uint8_t *out_bytes,
                                const uint64_t *part_offsets, uint32_t nparts,
                                uint64_t *part_boffsets, uint32_t *k4w_meta,
-                               uint32_t *k4w_order, hipStream_t s);
+                               uint32_t *k4w_order, int skip_copy, hipStream_t s);
 hipError_t dd_launch_partial_reduce(const dd_kargs *a, int64_t nblocks, int64_t chunk_rows,
                                     int n_aggs, const int32_t *agg_cols,
                                     const int32_t *agg_ops, uint64_t *out_keys,

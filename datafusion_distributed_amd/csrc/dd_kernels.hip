@@ -1585,6 +1585,218 @@ __global__ __launch_bounds__(BLOCK_THREADS) void k_round_roff(
     }
 }
 
+/* ================= K5: LDS-staged var-byte scatter (round 2) =================
+ * Replaces K4's output-slot GATHER (k4_copy) for var batches whose strings are small:
+ * the gather reads a random ~64 B string per slot and fetches whole 128 B lines —
+ * measured 2.79 GB HBM reads for 1.51 GB algorithmic on ClickBench (read amp ~1.9x,
+ * profiles/r02_clickbench_k4.json) and grid-depth-insensitive (not latency-bound).
+ * K5 instead walks the INPUT in row order (coalesced reads), places each row's bytes
+ * into a partition-major LDS byte image, and flushes per-partition byte runs with
+ * shift-aligned 8 B stores (coalesced writes) — the K3-P precompute discipline applied
+ * to bytes:
+ *   - rounds of R5 = 1024 rows (16 waves x one 64-row group); bcounts[seg][p] = bytes
+ *     of segment seg for partition p (k5_count), scanned to GLOBAL byte bases (the
+ *     shared dd_launch_scan_deep + part_offsets fold), per-round image bases
+ *     roffB[r][p..P] incl. total (k5_roff);
+ *   - eligibility: ONE var column, max string length <= DD_K5_MAXLEN (the LDS image is
+ *     R5 * maxlen-bounded; k5_maxlen computes it at create), var bytes < 2^31; longer
+ *     strings keep the K4 gather (they are line-dense already).
+ * Byte-granular LDS writes use head-bytes / aligned-dword / tail-bytes splits (LDS
+ * requires natural alignment, unlike gfx950 global unaligned 8 B ops). Stability and
+ * agreement with K4c's slot-side offsets hold by construction: both are prefix sums of
+ * the same lengths in the same partition-major, row-stable order. */
+
+__global__ void k5_maxlen(const int32_t *offsets, int64_t n, uint32_t *out_max) {
+    const int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    uint32_t len = 0;
+    if (i < n) len = (uint32_t)(offsets[i + 1] - offsets[i]);
+#pragma unroll
+    for (int d = 32; d >= 1; d >>= 1) {
+        uint32_t o = (uint32_t)__shfl_down((int)len, d);
+        if (o > len) len = o;
+    }
+    if ((threadIdx.x % WAVE) == 0 && len) atomicMax(out_max, len);
+}
+
+/* per-64-row-group byte histograms: seg = one wave's group; counts[seg][p] (u32) */
+__global__ __launch_bounds__(BLOCK_THREADS) void k5_count(
+    int64_t n, uint32_t nparts, const uint32_t *pid, const int32_t *offsets,
+    uint32_t *bcounts /* [nseg5][P] */, uint32_t *partials /* [nranges] fused L1 */,
+    int nranges, int64_t nseg5) {
+    extern __shared__ __attribute__((aligned(16))) char smem[];
+    const int wid = threadIdx.x / WAVE;
+    const int lane = threadIdx.x % WAVE;
+    const int64_t seg = (int64_t)blockIdx.x * WAVES_PER_BLOCK + wid;
+    if (seg >= nseg5) return;
+    uint32_t *hist = (uint32_t *)smem + (size_t)wid * nparts;
+    for (uint32_t p = lane; p < nparts; p += WAVE) hist[p] = 0;
+    const int64_t row = seg * WAVE + lane;
+    if (row < n) {
+        const uint32_t len = (uint32_t)(offsets[row + 1] - offsets[row]);
+        if (len) atomicAdd(&hist[pid[row]], len);
+    }
+    /* fair-division range of this segment (k_scan_partial mapping) */
+    const int64_t rw = ((seg + 1) * (int64_t)nranges + nseg5 - 1) / nseg5 - 1;
+    uint32_t *prow = partials + (size_t)rw * nparts;
+    for (uint32_t p = lane; p < nparts; p += WAVE) {
+        const uint32_t h = hist[p];
+        bcounts[(size_t)seg * nparts + p] = h;
+        if (h) atomicAdd(&prow[p], h);
+    }
+}
+
+/* per-round image byte bases incl. total: roffB[r][p] for p in [0,P], stride P+1 */
+__global__ __launch_bounds__(BLOCK_THREADS) void k5_roff(
+    const uint32_t *bcounts /* GLOBAL byte bases after the scan fold */,
+    const uint64_t *part_boffsets, int64_t nrounds, int wpb, uint32_t nparts,
+    uint32_t *roffB /* [nrounds][P+1] */) {
+    const int wid = threadIdx.x / WAVE;
+    const int lane = threadIdx.x % WAVE;
+    const int64_t r = (int64_t)blockIdx.x * WAVES_PER_BLOCK + wid;
+    if (r >= nrounds) return;
+    const uint32_t *rb = bcounts + (size_t)r * wpb * nparts;
+    const uint32_t *re = (r + 1 < nrounds) ? bcounts + (size_t)(r + 1) * wpb * nparts : nullptr;
+    uint32_t *orow = roffB + (size_t)r * (nparts + 1);
+    uint32_t carry = 0;
+    for (uint32_t p0 = 0; p0 < nparts; p0 += WAVE) {
+        const uint32_t p = p0 + lane;
+        uint32_t c = 0;
+        if (p < nparts) c = (re ? re[p] : (uint32_t)part_boffsets[p + 1]) - rb[p];
+        uint32_t v = c;
+#pragma unroll
+        for (int d = 1; d < WAVE; d <<= 1) {
+            uint32_t u = (uint32_t)__shfl_up((int)v, d);
+            if (lane >= d) v += u;
+        }
+        if (p < nparts) orow[p] = carry + v - c; /* exclusive */
+        carry += (uint32_t)__shfl((int)v, WAVE - 1);
+    }
+    if (lane == 0) orow[nparts] = carry; /* round byte total */
+}
+
+/* byte copy helpers: LDS is natural-alignment-only; global unaligned 8B is native */
+__device__ __forceinline__ void k5_bytes_to_lds(char *dst, const uint8_t *src,
+                                                uint32_t len) {
+    uint32_t b = 0;
+    /* head: bytes until dst is 4-aligned */
+    while (b < len && ((uintptr_t)(dst + b) & 3)) {
+        dst[b] = (char)src[b];
+        b++;
+    }
+    for (; b + 4 <= len; b += 4) {
+        uint32_t t;
+        __builtin_memcpy(&t, src + b, 4); /* unaligned global read: native */
+        *(uint32_t *)(dst + b) = t;
+    }
+    for (; b < len; b++) dst[b] = (char)src[b];
+}
+
+template <int WPB5>
+__global__ __launch_bounds__(WPB5 * WAVE) void k5_scatter(
+    int64_t n, uint32_t nparts, int nbits, const uint32_t *pid, const int32_t *offsets,
+    const uint8_t *in_bytes, const uint32_t *gbaseB /* [nseg5][P] global byte bases */,
+    const uint32_t *roffB /* [nrounds][P+1] */, int64_t nrounds, uint8_t *out_bytes) {
+    constexpr int R5 = WPB5 * WAVE;
+    extern __shared__ __attribute__((aligned(16))) char smem[];
+    /* carve: per-wave gbase u32[P] | shared roff u32[P+1] | image */
+    char *ws = smem;
+    uint32_t *gb_all = (uint32_t *)ws;
+    ws += sizeof(uint32_t) * WPB5 * nparts;
+    uint32_t *roff = (uint32_t *)ws;
+    ws += sizeof(uint32_t) * (nparts + 1);
+    char *const img = ws;
+
+    const int tid = threadIdx.x;
+    const int wid = tid / WAVE;
+    const int lane = tid % WAVE;
+    uint32_t *gb = gb_all + (size_t)wid * nparts;
+    uint32_t *gb0 = gb_all;
+    const uint64_t lt = ((uint64_t)1 << lane) - 1;
+    __shared__ uint32_t lens_s[WPB5][WAVE];
+    __shared__ uint32_t srcb_s[WPB5][WAVE];
+
+    const int64_t r = blockIdx.x;
+    if (r >= nrounds) return;
+    const int64_t seg = r * WPB5 + wid;
+    const int64_t row = seg * WAVE + lane;
+
+    /* base rows -> LDS (wave-private write; cross-wave gb0 reads ordered by barrier) */
+    {
+        const uint32_t *grow = gbaseB + (size_t)seg * nparts;
+        for (uint32_t p = lane; p < nparts; p += WAVE) gb[p] = grow[p];
+        const uint32_t *rrow = roffB + (size_t)r * (nparts + 1);
+        for (uint32_t p = tid; p <= nparts; p += WPB5 * WAVE) roff[p] = rrow[p];
+    }
+    __syncthreads();
+
+    /* place: per-row byte range into the partition-major image */
+    uint32_t len = 0, sb = 0, pidv = 0;
+    if (row < n) {
+        const int32_t o0 = offsets[row];
+        len = (uint32_t)(offsets[row + 1] - o0);
+        sb = (uint32_t)o0;
+        pidv = pid[row];
+    }
+    lens_s[wid][lane] = len;
+    srcb_s[wid][lane] = sb;
+    /* group byte prefix among same-partition lanes (v1 k_scatter idiom) */
+    const uint64_t act = __ballot(row < n);
+    if (row < n) {
+        uint64_t eq = dd_eq_mask(pidv, act, nbits);
+        uint32_t bpre = 0, bsum = 0;
+        uint64_t m = eq;
+        while (m) {
+            int j = __ffsll((unsigned long long)m) - 1;
+            uint32_t lj = lens_s[wid][j];
+            if (j < lane) bpre += lj;
+            bsum += lj;
+            m &= m - 1;
+        }
+        (void)bsum;
+        if (len) {
+            const uint32_t gdst = gb[pidv] + bpre;       /* global byte pos */
+            const uint32_t slot = roff[pidv] + (gdst - gb0[pidv]); /* image byte pos */
+            k5_bytes_to_lds(img + slot, in_bytes + sb, len);
+        }
+    }
+    __syncthreads();
+
+    /* flush: per-partition byte runs, lanes striding 8 B; LDS reads are shift-aligned
+     * dword pairs, global stores unaligned 8 B (native). Partitions are wave-strided
+     * for balance. */
+    const uint32_t total = roff[nparts];
+    (void)total;
+    for (uint32_t p = wid; p < nparts; p += WPB5) {
+        const uint32_t i0 = roff[p];
+        const uint32_t cnt = roff[p + 1] - i0;
+        if (!cnt) continue;
+        /* global destination of this run: round base for p = gb0[p] */
+        uint8_t *gdst = out_bytes + gb0[p];
+        uint32_t b = lane * 8;
+        for (; b + 8 <= cnt; b += WAVE * 8) {
+            /* read 8 unaligned LDS bytes as two aligned dwords + funnel shift */
+            const uint32_t src = i0 + b;
+            const uint32_t a0 = (uint32_t)(src & ~3u);
+            const uint32_t sh = (src & 3u) * 8;
+            uint32_t w0 = *(const uint32_t *)(img + a0);
+            uint32_t w1 = *(const uint32_t *)(img + a0 + 4);
+            uint32_t w2 = *(const uint32_t *)(img + a0 + 8);
+            uint64_t v;
+            if (sh == 0) {
+                v = (uint64_t)w0 | ((uint64_t)w1 << 32);
+            } else {
+                uint32_t lo = (w0 >> sh) | (w1 << (32 - sh));
+                uint32_t hi = (w1 >> sh) | (w2 << (32 - sh));
+                v = (uint64_t)lo | ((uint64_t)hi << 32);
+            }
+            __builtin_memcpy(gdst + b, &v, 8); /* unaligned global store: native */
+        }
+        /* tail < 8 B: after the strided loop exactly one lane has b < cnt */
+        if (b < cnt) {
+            for (uint32_t t = b; t < cnt; t++) gdst[t] = (uint8_t)img[i0 + t];
+        }
+    }
+}
 /* ================= K4: var-width bytes for the staged path =================
  * The staged scatter (v2) handles var columns' LENGTHS and a ROWID permutation as
  * synthetic fixed u32 columns (DD_KDT_VARLEN / DD_KDT_ROWID, set up by dd_host.cpp).
@@ -1804,7 +2016,7 @@ hipError_t dd_launch_var_bytes(const uint32_t *lens, const uint32_t *src_row,
                                uint64_t *out_off, uint8_t *out_bytes,
                                const uint64_t *part_offsets, uint32_t nparts,
                                uint64_t *part_boffsets, uint32_t *k4w_meta,
-                               uint32_t *k4w_order, hipStream_t s) {
+                               uint32_t *k4w_order, int skip_copy, hipStream_t s) {
     const int wavegrid = K4_RANGES * WAVE / 256;
     hipLaunchKernelGGL(k4_len_partials, dim3(wavegrid), dim3(256), 0, s, lens, n, partials);
     hipLaunchKernelGGL(k4_scan_partials, dim3(1), dim3(256), 0, s, partials, K4_RANGES,
@@ -1834,6 +2046,57 @@ hipError_t dd_launch_var_bytes(const uint32_t *lens, const uint32_t *src_row,
     (void)total_bytes;
     hipLaunchKernelGGL(k4_part_boffsets, dim3((nparts + 256) / 256 + 1), dim3(256), 0, s,
                        out_off, part_offsets, nparts, n, total_bytes, part_boffsets);
+    return hipGetLastError();
+}
+
+hipError_t dd_launch_k5_maxlen(const int32_t *offsets, int64_t n, uint32_t *out_max,
+                               hipStream_t s) {
+    int blocks = (int)((n + 255) / 256);
+    if (blocks < 1) blocks = 1;
+    hipLaunchKernelGGL(k5_maxlen, dim3(blocks), dim3(256), 0, s, offsets, n, out_max);
+    return hipGetLastError();
+}
+
+hipError_t dd_launch_k5_count(int64_t n, uint32_t nparts, const uint32_t *pid,
+                              const int32_t *offsets, uint32_t *bcounts,
+                              uint32_t *partials, int nranges, int64_t nseg5,
+                              size_t lds_bytes, hipStream_t s) {
+    int64_t blocks = nseg5 / WAVES_PER_BLOCK;
+    if (blocks < 1) blocks = 1;
+    if (lds_bytes > 65536) {
+        hipError_t e = hipFuncSetAttribute((const void *)k5_count,
+                                           hipFuncAttributeMaxDynamicSharedMemorySize,
+                                           (int)lds_bytes);
+        if (e != hipSuccess) return e;
+    }
+    hipLaunchKernelGGL(k5_count, dim3((unsigned)blocks), dim3(BLOCK_THREADS), lds_bytes,
+                       s, n, nparts, pid, offsets, bcounts, partials, nranges, nseg5);
+    return hipGetLastError();
+}
+
+hipError_t dd_launch_k5_roff(const uint32_t *bcounts, const uint64_t *part_boffsets,
+                             int64_t nrounds, int wpb, uint32_t nparts, uint32_t *roffB,
+                             hipStream_t s) {
+    dim3 grid((unsigned)((nrounds + WAVES_PER_BLOCK - 1) / WAVES_PER_BLOCK));
+    hipLaunchKernelGGL(k5_roff, grid, dim3(BLOCK_THREADS), 0, s, bcounts, part_boffsets,
+                       nrounds, wpb, nparts, roffB);
+    return hipGetLastError();
+}
+
+hipError_t dd_launch_k5_scatter(int64_t n, uint32_t nparts, int nbits, const uint32_t *pid,
+                                const int32_t *offsets, const uint8_t *in_bytes,
+                                const uint32_t *gbaseB, const uint32_t *roffB,
+                                int64_t nrounds, uint8_t *out_bytes, size_t lds_bytes,
+                                hipStream_t s) {
+    if (lds_bytes > 65536) {
+        hipError_t e = hipFuncSetAttribute((const void *)k5_scatter<16>,
+                                           hipFuncAttributeMaxDynamicSharedMemorySize,
+                                           (int)lds_bytes);
+        if (e != hipSuccess) return e;
+    }
+    hipLaunchKernelGGL((k5_scatter<16>), dim3((unsigned)nrounds), dim3(16 * WAVE),
+                       lds_bytes, s, n, nparts, nbits, pid, offsets, in_bytes, gbaseB,
+                       roffB, nrounds, out_bytes);
     return hipGetLastError();
 }
 
