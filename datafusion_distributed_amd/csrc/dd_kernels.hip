@@ -2030,17 +2030,20 @@ hipError_t dd_launch_var_bytes(const uint32_t *lens, const uint32_t *src_row,
      * and as a balanced window-bucketed group order (randomized group-level write
      * order, 10 ms): the write side loses more than the read side gains.
      * DD_K4_BLOCKS overrides the grid cap (latency-hiding depth experiment). */
-    int copy_blocks = (int)((n + 255) / 256);
-    int cap = 8192;
-    if (const char *e = getenv("DD_K4_BLOCKS")) {
-        int v = atoi(e);
-        if (v >= 256 && v <= 65536) cap = v;
+    if (!skip_copy) { /* skip_copy: K5 (k5_scatter) writes the bytes instead */
+        int copy_blocks = (int)((n + 255) / 256);
+        int cap = 8192;
+        if (const char *e = getenv("DD_K4_BLOCKS")) {
+            int v = atoi(e);
+            if (v >= 256 && v <= 65536) cap = v;
+        }
+        if (copy_blocks > cap) copy_blocks = cap; /* latency-bound: deep
+                                                     oversubscription hides the random
+                                                     string reads */
+        if (copy_blocks < 1) copy_blocks = 1;
+        hipLaunchKernelGGL(k4_copy, dim3(copy_blocks), dim3(256), 0, s, src_row, out_off,
+                           in_offsets, in_bytes, n, out_bytes);
     }
-    if (copy_blocks > cap) copy_blocks = cap; /* latency-bound: deep oversubscription
-                                                 hides the random string reads */
-    if (copy_blocks < 1) copy_blocks = 1;
-    hipLaunchKernelGGL(k4_copy, dim3(copy_blocks), dim3(256), 0, s, src_row, out_off,
-                       in_offsets, in_bytes, n, out_bytes);
     (void)k4w_meta;
     (void)k4w_order;
     (void)total_bytes;
