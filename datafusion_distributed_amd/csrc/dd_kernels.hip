@@ -1683,9 +1683,16 @@ __device__ __forceinline__ void k5_bytes_to_lds(char *dst, const uint8_t *src,
         dst[b] = (char)src[b];
         b++;
     }
+    /* 8-B unaligned global reads (native on gfx950), two aligned dword LDS stores */
+    for (; b + 8 <= len; b += 8) {
+        uint64_t t;
+        __builtin_memcpy(&t, src + b, 8);
+        *(uint32_t *)(dst + b) = (uint32_t)t;
+        *(uint32_t *)(dst + b + 4) = (uint32_t)(t >> 32);
+    }
     for (; b + 4 <= len; b += 4) {
         uint32_t t;
-        __builtin_memcpy(&t, src + b, 4); /* unaligned global read: native */
+        __builtin_memcpy(&t, src + b, 4);
         *(uint32_t *)(dst + b) = t;
     }
     for (; b < len; b++) dst[b] = (char)src[b];
