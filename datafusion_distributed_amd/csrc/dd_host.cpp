@@ -597,12 +597,14 @@ extern "C" dd_status dd_partitioner_create(const dd_batch_desc *batch, const int
             kc.out_data = p->out_lengths[ci];
             ka.n_cols++;
         }
-        dd_kcol &kr = ka.cols[ka.n_cols];
-        memset(&kr, 0, sizeof(kr));
-        kr.dtype = DD_KDT_ROWID;
-        kr.elem = 4;
-        kr.out_data = p->src_row;
-        ka.n_cols++;
+        if (!p->k5) { /* ROWID (src_row) only feeds the K4 gather; K5 replaces it */
+            dd_kcol &kr = ka.cols[ka.n_cols];
+            memset(&kr, 0, sizeof(kr));
+            kr.dtype = DD_KDT_ROWID;
+            kr.elem = 4;
+            kr.out_data = p->src_row;
+            ka.n_cols++;
+        }
     }
 
     for (auto &e : p->ev)
