@@ -202,17 +202,22 @@ struct FB {
 struct FbTable {
     const uint8_t *base; /* buffer start */
     size_t len;
-    size_t pos; /* table position */
+    size_t pos; /* table position; every accessor bounds-checks against len (the wire
+                   may carry hostile/corrupt bytes — reads never leave the metadata
+                   buffer; absent/invalid fields read as defaults) */
 
-    bool valid() const { return pos != 0; }
+    bool valid() const { return pos != 0 && pos + 4 <= len; }
     uint16_t vt_entry(int id) const {
+        if (!valid()) return 0;
         int32_t so;
         memcpy(&so, base + pos, 4);
+        if (so < 0 ? pos + (size_t)(-so) > len : (size_t)so > pos) return 0;
         size_t vt = pos - so;
         if (vt + 4 > len) return 0;
         uint16_t vt_len;
         memcpy(&vt_len, base + vt, 2);
-        size_t slot = 4 + 2 * id;
+        if (vt_len < 4 || vt + vt_len > len) return 0;
+        size_t slot = 4 + 2 * (size_t)id;
         if (slot + 2 > vt_len) return 0;
         uint16_t off;
         memcpy(&off, base + vt + slot, 2);
@@ -220,33 +225,40 @@ struct FbTable {
     }
     template <typename T> T scalar(int id, T dflt) const {
         uint16_t o = vt_entry(id);
-        if (!o) return dflt;
+        if (!o || pos + o + sizeof(T) > len) return dflt;
         T v;
         memcpy(&v, base + pos + o, sizeof(T));
         return v;
     }
-    size_t ref(int id) const { /* position of referenced object; 0 = absent */
+    size_t ref(int id) const { /* position of referenced object; 0 = absent/invalid */
         uint16_t o = vt_entry(id);
-        if (!o) return 0;
+        if (!o || pos + o + 4 > len) return 0;
         uint32_t rel;
         memcpy(&rel, base + pos + o, 4);
-        return pos + o + rel;
+        size_t t = pos + o + rel;
+        return (t + 4 <= len) ? t : 0;
     }
     FbTable table(int id) const { return FbTable{base, len, ref(id)}; }
 };
 
+
 struct FbVector {
     const uint8_t *base;
     size_t pos; /* 0 = absent */
+    size_t len = SIZE_MAX; /* metadata buffer bound (when provided) */
     uint32_t count() const {
-        if (!pos) return 0;
+        if (!pos || pos + 4 > len) return 0;
         uint32_t n;
         memcpy(&n, base + pos, 4);
+        /* clamp: a hostile count cannot push element reads past the buffer */
+        if (len != SIZE_MAX && pos + 4 + (size_t)n * 4 > len && n > (len - pos) / 4)
+            n = (uint32_t)((len - pos - 4) / 4);
         return n;
     }
     size_t elem(size_t i, size_t elem_size) const { return pos + 4 + i * elem_size; }
     size_t ref_elem(size_t i) const {
         size_t e = elem(i, 4);
+        if (e + 4 > len) return 0;
         uint32_t rel;
         memcpy(&rel, base + e, 4);
         return e + rel;
@@ -637,14 +649,14 @@ extern "C" dd_status dd_ipc_reader_create(const uint8_t *data, int64_t len,
         memcpy(&rootrel, flat, 4);
         FbTable msg{flat, (size_t)meta_len, rootrel};
         uint8_t hdr_type = msg.scalar<uint8_t>(1, 0);
-        int64_t body_len = msg.scalar<int64_t>(3, 0);
+        const int64_t body_len = msg.scalar<int64_t>(3, 0);
         const uint8_t *body = data + meta_at + meta_len;
         if (meta_at + meta_len + (size_t)body_len > (size_t)len)
             return dd_set_error(DD_ERR_INVALID, "truncated IPC body");
 
         if (hdr_type == MSG_SCHEMA) {
             FbTable schema = msg.table(2);
-            FbVector fields{flat, schema.ref(1)};
+            FbVector fields{flat, schema.ref(1), (size_t)meta_len};
             for (uint32_t i = 0; i < fields.count(); i++) {
                 FbTable f{flat, (size_t)meta_len, fields.ref_elem(i)};
                 if (f.ref(4) != 0)
@@ -668,8 +680,8 @@ extern "C" dd_status dd_ipc_reader_create(const uint8_t *data, int64_t len,
                 return dd_set_error(DD_ERR_INVALID, "RecordBatch before Schema");
             FbTable rb = msg.table(2);
             int64_t n_rows = rb.scalar<int64_t>(0, 0);
-            FbVector nodes{flat, rb.ref(1)};
-            FbVector buffers{flat, rb.ref(2)};
+            FbVector nodes{flat, rb.ref(1), (size_t)meta_len};
+            FbVector buffers{flat, rb.ref(2), (size_t)meta_len};
             int compressed = 0;
             FbTable comp = rb.table(3);
             if (comp.valid()) {
@@ -686,32 +698,43 @@ extern "C" dd_status dd_ipc_reader_create(const uint8_t *data, int64_t len,
                 if (bi >= buffers.count()) return false;
                 int64_t off, blen;
                 size_t e = buffers.elem(bi++, 16);
+                if (e + 16 > (size_t)meta_len) return false;
                 memcpy(&off, flat + e, 8);
                 memcpy(&blen, flat + e + 8, 8);
+                if (off < 0 || blen < 0 || off + blen > body_len) return false;
                 return decompress_buf(body + off, (size_t)blen, compressed, out);
             };
             for (size_t c = 0; c < r->cols.size(); c++) {
                 int64_t nc = 0;
                 if (c < nodes.count()) {
                     size_t e = nodes.elem(c, 16);
-                    memcpy(&nc, flat + e + 8, 8);
+                    if (e + 16 <= (size_t)meta_len) memcpy(&nc, flat + e + 8, 8);
                 }
                 bt.null_count.push_back(nc);
                 std::vector<uint8_t> vbits, dat, off;
                 if (!next_buf(vbits))
                     return dd_set_error(DD_ERR_INVALID, "missing validity buffer");
                 std::vector<uint8_t> vu8;
-                if (!vbits.empty() && nc > 0) unpack_bits(vbits.data(), n_rows, vu8);
+                if (!vbits.empty() && nc > 0) {
+                    if ((int64_t)vbits.size() * 8 < n_rows)
+                        return dd_set_error(DD_ERR_INVALID, "validity bitmap too short");
+                    unpack_bits(vbits.data(), n_rows, vu8);
+                }
                 bt.validity.push_back(std::move(vu8));
                 if (r->cols[c].dtype == DD_DT_UTF8) {
                     if (!next_buf(off) || !next_buf(dat))
                         return dd_set_error(DD_ERR_INVALID, "missing utf8 buffers");
+                    if ((int64_t)off.size() < (n_rows + 1) * 4)
+                        return dd_set_error(DD_ERR_INVALID, "utf8 offsets too short");
                 } else if (r->cols[c].dtype == DD_DT_BOOL) {
                     std::vector<uint8_t> bits;
                     if (!next_buf(bits))
                         return dd_set_error(DD_ERR_INVALID, "missing bool buffer");
-                    if (!bits.empty()) unpack_bits(bits.data(), n_rows, dat);
-                    else dat.assign((size_t)n_rows, 0);
+                    if (!bits.empty()) {
+                        if ((int64_t)bits.size() * 8 < n_rows)
+                            return dd_set_error(DD_ERR_INVALID, "bool bitmap too short");
+                        unpack_bits(bits.data(), n_rows, dat);
+                    } else dat.assign((size_t)n_rows, 0);
                 } else {
                     if (!next_buf(dat))
                         return dd_set_error(DD_ERR_INVALID, "missing data buffer");
