@@ -293,6 +293,43 @@ __global__ __launch_bounds__(BT) void k_abl(
         const bool more = r + 1 < r1;
         if (more) preload(r + 1);
 
+        if (D == 8) { /* paired flush: lane covers slots (2i, 2i+1); same-run pairs
+                          (~97% at run ~32 slots) fuse into one 16B store per column */
+#pragma unroll
+            for (int u = 0; u < GMAX / 2; u++) {
+                const int i = 2 * (tid + u * BT);
+                const int ic = (i < round_rows) ? i : (round_rows - 1);
+                const int ic2 = (i + 1 < round_rows) ? i + 1 : (round_rows - 1);
+                const uint64_t d0 = dstg[ic];
+                const uint64_t d1 = dstg[ic2];
+                char *stage = stage0;
+                if (d1 == d0 + 1 && (d0 & 1) == 0) {
+                    { uint64_t v2[2] = {((const uint64_t *)stage)[ic], ((const uint64_t *)stage)[ic2]};
+                      __builtin_memcpy(o0 + d0, v2, 16); }
+                    stage += (size_t)R * 8;
+                    { uint64_t v2[2] = {((const uint64_t *)stage)[ic], ((const uint64_t *)stage)[ic2]};
+                      __builtin_memcpy(o1 + d0, v2, 16); }
+                    stage += (size_t)R * 8;
+                    { uint64_t v2[2] = {((const uint64_t *)stage)[ic], ((const uint64_t *)stage)[ic2]};
+                      __builtin_memcpy(o2 + d0, v2, 16); }
+                    stage += (size_t)R * 8;
+                    { uint32_t v2[2] = {((const uint32_t *)stage)[ic], ((const uint32_t *)stage)[ic2]};
+                      __builtin_memcpy(o3 + d0, v2, 8); }
+                } else {
+                    o0[d0] = ((const uint64_t *)stage)[ic];
+                    o0[d1] = ((const uint64_t *)stage)[ic2];
+                    stage += (size_t)R * 8;
+                    o1[d0] = ((const uint64_t *)stage)[ic];
+                    o1[d1] = ((const uint64_t *)stage)[ic2];
+                    stage += (size_t)R * 8;
+                    o2[d0] = ((const uint64_t *)stage)[ic];
+                    o2[d1] = ((const uint64_t *)stage)[ic2];
+                    stage += (size_t)R * 8;
+                    o3[d0] = ((const uint32_t *)stage)[ic];
+                    o3[d1] = ((const uint32_t *)stage)[ic2];
+                }
+            }
+        } else
         for (int rep = 0; rep < (D == 6 ? 2 : 1); rep++) {
             /* rep 1 (D==6) writes a SECOND buffer set: same pattern, no WAW hazard */
             uint64_t *q0 = rep ? ob0 : o0, *q1 = rep ? ob1 : o1, *q2 = rep ? ob2 : o2;
@@ -463,6 +500,21 @@ int main() {
     float t5 = run(std::integral_constant<int, 5>{}, names[5]);
     float t6 = run(std::integral_constant<int, 6>{}, names[6]);
     float t7 = run(std::integral_constant<int, 7>{}, "nt_stores(negative)");
+    float t8 = run(std::integral_constant<int, 8>{}, "paired_16B_flush");
+    {
+        /* verify the paired variant too (it ran last into o0) */
+        std::vector<uint64_t> out0(n);
+        HC(hipMemcpy(out0.data(), d_o0, n * 8, hipMemcpyDeviceToHost));
+        std::vector<uint64_t> cursor(P);
+        for (uint32_t p = 0; p < P; p++) cursor[p] = ptot[p];
+        int bad = 0;
+        for (int64_t i = 0; i < n && bad < 3; i++) {
+            uint64_t d = cursor[pid[i]]++;
+            if (out0[d] != hv[i]) bad++;
+        }
+        printf(bad ? "PAIRED VERIFY FAILED\n" : "paired verify ok\n");
+        printf("paired delta vs full: %.3f ms\n", t8 - full);
+    }
     printf("marginal ms: pid=%.3f cols=%.3f bases=%.3f rank=%.3f place=%.3f flush=%.3f "
            "nt_delta=%.3f\n",
            t1 - full, t2 - full, t3 - full, t4 - full, t5 - full, t6 - full, t7 - full);
