@@ -1532,12 +1532,10 @@ __global__ __launch_bounds__(BLOCK_THREADS) void k_hash_count_seg(
         }
 #pragma unroll
         for (int u = 0; u < 4; u++) {
-            uint64_t act = __ballot(actu[u]);
-            if (actu[u]) {
-                uint64_t eq = dd_eq_mask(pidu[u], act, nbits);
-                int leader = __ffsll((unsigned long long)eq) - 1;
-                if (lane == leader) hist[pidu[u]] += (uint32_t)__popcll((unsigned long long)eq);
-            }
+            /* LDS atomicAdd counting: measured 30% FASTER than the ballot-multisplit
+             * leader-add on this shape (tools/ablate_k1.cpp: 0.143 vs 0.202 ms) — the
+             * log2(P) ballot chain costs more than uniform-key LDS atomic contention */
+            if (actu[u]) atomicAdd(&hist[pidu[u]], 1u);
         }
     }
     /* range of this segment under k_scan_partial's fair-division mapping
