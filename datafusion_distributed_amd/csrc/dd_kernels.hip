@@ -80,11 +80,8 @@ __global__ __launch_bounds__(BLOCK_THREADS) void k_hash_count(
             pid = fine >> a.pid_shift;
             pid_out[row] = pid;
         }
-        uint64_t act = __ballot(active);
         if (active) {
-            uint64_t eq = dd_eq_mask(pid, act, nbits);
-            int leader = __ffsll((unsigned long long)eq) - 1;
-            if (lane == leader) hist[pid] += (uint32_t)__popcll((unsigned long long)eq);
+            atomicAdd(&hist[pid], 1u); /* LDS atomic counting (tools/ablate_k1.cpp) */
             /* var-col byte sums: LDS atomic add (order-free) */
             for (int v = 0; v < a.n_var; v++) {
                 const dd_kcol &c = a.cols[a.var_idx[v]];
@@ -344,13 +341,9 @@ __global__ __launch_bounds__(BLOCK_THREADS) void k_hash_count_tile(
         }
 #pragma unroll
         for (int u = 0; u < 2; u++) {
-            uint64_t act = __ballot(actu[u]);
-            if (actu[u]) {
-                uint64_t eq = dd_eq_mask(pidu[u], act, nbits);
-                int leader = __ffsll((unsigned long long)eq) - 1;
-                if (lane == leader)
-                    myhist[pidu[u]] += (uint32_t)__popcll((unsigned long long)eq);
-            }
+            /* LDS atomicAdd counting: measured faster than ballot-multisplit
+             * (tools/ablate_k1.cpp) */
+            if (actu[u]) atomicAdd(&myhist[pidu[u]], 1u);
         }
     }
     __syncthreads();
