@@ -572,6 +572,9 @@ bool decompress_buf(const uint8_t *p, size_t n, int compressed, std::vector<uint
         out.assign(p + 8, p + n);
         return true;
     }
+    /* hostile-length guard: lz4 frames expand at most ~255x (+ header slack); a claimed
+     * size beyond that is corrupt — refuse instead of attempting a huge allocation */
+    if (ulen < 0 || (uint64_t)ulen > (uint64_t)(n - 8) * 256 + 65536) return false;
     out.resize((size_t)ulen);
     if (ulen == 0) return true;
     void *dctx = nullptr;
@@ -642,7 +645,7 @@ extern "C" dd_status dd_ipc_reader_create(const uint8_t *data, int64_t len,
             meta_at = pos + 4;
         }
         if (meta_len == 0) break; /* EOS */
-        if (meta_at + (size_t)meta_len > (size_t)len)
+        if (meta_len < 8 || meta_at + (size_t)meta_len > (size_t)len)
             return dd_set_error(DD_ERR_INVALID, "truncated IPC message");
         const uint8_t *flat = data + meta_at;
         uint32_t rootrel;
@@ -651,7 +654,7 @@ extern "C" dd_status dd_ipc_reader_create(const uint8_t *data, int64_t len,
         uint8_t hdr_type = msg.scalar<uint8_t>(1, 0);
         const int64_t body_len = msg.scalar<int64_t>(3, 0);
         const uint8_t *body = data + meta_at + meta_len;
-        if (meta_at + meta_len + (size_t)body_len > (size_t)len)
+        if (body_len < 0 || meta_at + meta_len + (size_t)body_len > (size_t)len)
             return dd_set_error(DD_ERR_INVALID, "truncated IPC body");
 
         if (hdr_type == MSG_SCHEMA) {
@@ -667,10 +670,11 @@ extern "C" dd_status dd_ipc_reader_create(const uint8_t *data, int64_t len,
                 if (!dt) return dd_set_error(DD_ERR_UNSUPPORTED, "unsupported wire type");
                 std::string name;
                 size_t nm = f.ref(0);
-                if (nm) {
+                if (nm && nm + 4 <= (size_t)meta_len) {
                     uint32_t nl;
                     memcpy(&nl, flat + nm, 4);
-                    name.assign((const char *)flat + nm + 4, nl);
+                    if (nm + 4 + (size_t)nl <= (size_t)meta_len) /* hostile-length guard */
+                        name.assign((const char *)flat + nm + 4, nl);
                 }
                 r->cols.push_back({dt, name, (int32_t)f.scalar<uint8_t>(1, 0)});
             }
@@ -680,6 +684,10 @@ extern "C" dd_status dd_ipc_reader_create(const uint8_t *data, int64_t len,
                 return dd_set_error(DD_ERR_INVALID, "RecordBatch before Schema");
             FbTable rb = msg.table(2);
             int64_t n_rows = rb.scalar<int64_t>(0, 0);
+            /* hostile row-count guard: rows must be representable by the stream's own
+             * buffers (validity/bool checks below bound further) */
+            if (n_rows < 0 || (uint64_t)n_rows > (uint64_t)len * 256 + 65536)
+                return dd_set_error(DD_ERR_INVALID, "implausible row count");
             FbVector nodes{flat, rb.ref(1), (size_t)meta_len};
             FbVector buffers{flat, rb.ref(2), (size_t)meta_len};
             int compressed = 0;
@@ -730,11 +738,9 @@ extern "C" dd_status dd_ipc_reader_create(const uint8_t *data, int64_t len,
                     std::vector<uint8_t> bits;
                     if (!next_buf(bits))
                         return dd_set_error(DD_ERR_INVALID, "missing bool buffer");
-                    if (!bits.empty()) {
-                        if ((int64_t)bits.size() * 8 < n_rows)
-                            return dd_set_error(DD_ERR_INVALID, "bool bitmap too short");
-                        unpack_bits(bits.data(), n_rows, dat);
-                    } else dat.assign((size_t)n_rows, 0);
+                    if ((int64_t)bits.size() * 8 < n_rows)
+                        return dd_set_error(DD_ERR_INVALID, "bool bitmap too short");
+                    unpack_bits(bits.data(), n_rows, dat);
                 } else {
                     if (!next_buf(dat))
                         return dd_set_error(DD_ERR_INVALID, "missing data buffer");

@@ -248,3 +248,36 @@ def test_wire_py_roundtrips():
     blob0 = write_stream([], [(7, {}), (3, {})], True)
     back0 = wire.decode_batches(blob0)
     assert [b.num_rows for b in back0] == [7, 3]
+
+
+def test_reader_survives_hostile_bytes():
+    """The wire may carry corrupt/hostile bytes: every mutation of a valid stream (and
+    pure noise) must return an error or a truncated-but-safe result — never crash or
+    read out of bounds (the reader bounds-checks all flatbuffer/body accesses)."""
+    rng = np.random.default_rng(11)
+    ours, _ = make_pa_batch(rng, 200)
+    blob = bytearray(write_stream(SCHEMA, [(200, ours)], True))
+    L = api.lib()
+    for trial in range(300):
+        mutated = bytearray(blob)
+        for _ in range(rng.integers(1, 8)):
+            mutated[rng.integers(0, len(mutated))] = rng.integers(0, 256)
+        r = ctypes.c_void_p(0)
+        st = L.dd_ipc_reader_create(bytes(mutated), ctypes.c_int64(len(mutated)),
+                                    ctypes.byref(r))
+        if st == 0 and r.value:
+            # decodable-enough stream: accessors must stay in bounds too
+            nb = L.dd_ipc_reader_n_batches(r)
+            nf = L.dd_ipc_reader_n_fields(r)
+            for b in range(min(nb, 4)):
+                for c in range(min(nf, 8)):
+                    a = IpcArray()
+                    L.dd_ipc_reader_batch_col(r, b, c, ctypes.byref(a))
+            L.dd_ipc_reader_destroy(r)
+    for trial in range(100):  # pure noise
+        noise = bytes(rng.integers(0, 256, rng.integers(8, 400), dtype=np.int64)
+                      .astype(np.uint8))
+        r = ctypes.c_void_p(0)
+        st = L.dd_ipc_reader_create(noise, ctypes.c_int64(len(noise)), ctypes.byref(r))
+        if st == 0 and r.value:
+            L.dd_ipc_reader_destroy(r)
