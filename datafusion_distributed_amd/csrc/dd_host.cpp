@@ -150,6 +150,7 @@ struct dd_partitioner {
     uint32_t *k4w_meta = nullptr;        /* staged-var: window hist/base (k4_copy_ord) */
     uint32_t *k4w_order = nullptr;       /* staged-var: window-bucketed group order */
     int k5 = 0;                          /* LDS-staged var-byte scatter (small strings) */
+    int k5_wpb = 16;                     /* waves/block (8 = 2 blocks/CU overlap) */
     int64_t k5_nrounds = 0, k5_nseg = 0;
     uint32_t k5_maxlen = 0;
     size_t lds_k5 = 0, lds_k5c = 0;
@@ -528,13 +529,19 @@ extern "C" dd_status dd_partitioner_create(const dd_batch_desc *batch, const int
                                hipMemcpy(&maxlen, mdev, 4, hipMemcpyDeviceToHost) ==
                                    hipSuccess;
                     (void)hipFree(mdev);
-                    const int64_t R5 = 16 * 64;
+                    int wpb5 = 16;
+                    if (const char *e = getenv("DD_K5_WPB")) { /* experiment knob */
+                        int v = atoi(e);
+                        if (v == 8 || v == 16) wpb5 = v;
+                    }
+                    const int64_t R5 = (int64_t)wpb5 * 64;
                     const size_t img = (size_t)R5 * maxlen + 8;
                     const size_t lds5 =
-                        (size_t)16 * P * 4 + (size_t)(P + 1) * 4 + img;
+                        (size_t)wpb5 * P * 4 + (size_t)(P + 1) * 4 + img;
                     if (mok && maxlen > 0 && maxlen <= 128 && lds5 <= 163840) {
+                        p->k5_wpb = wpb5;
                         p->k5_nrounds = (n + R5 - 1) / R5;
-                        p->k5_nseg = p->k5_nrounds * 16;
+                        p->k5_nseg = p->k5_nrounds * wpb5;
                         p->k5_maxlen = maxlen;
                         p->lds_k5 = lds5;
                         p->lds_k5c = (size_t)WAVES_PER_BLOCK_H * P * 4;
@@ -672,8 +679,8 @@ extern "C" dd_status dd_partitioner_run_phase1(dd_partitioner *p, void *stream) 
         HIP_TRY(dd_launch_scan_deep(p->k5_bcounts, p->k5_nseg, p->nparts, 2048, 64,
                                     p->k5_partials, p->k5_partials2, p->part_boffsets,
                                     1, s));
-        HIP_TRY(dd_launch_k5_roff(p->k5_bcounts, p->part_boffsets, p->k5_nrounds, 16,
-                                  p->nparts, p->k5_roffB, s));
+        HIP_TRY(dd_launch_k5_roff(p->k5_bcounts, p->part_boffsets, p->k5_nrounds,
+                                  p->k5_wpb, p->nparts, p->k5_roffB, s));
     }
     if (!p->staged) {
         for (int v = 0; v < p->ka.n_var; v++) {
@@ -713,7 +720,7 @@ extern "C" dd_status dd_partitioner_run_phase2(dd_partitioner *p, void *stream) 
                 HIP_TRY(dd_launch_k5_scatter(
                     p->ka.n_rows, p->nparts, p->nbits, p->pid, cd.offsets,
                     (const uint8_t *)cd.data, p->k5_bcounts, p->k5_roffB, p->k5_nrounds,
-                    (uint8_t *)p->out_data[ci], p->lds_k5, s));
+                    (uint8_t *)p->out_data[ci], p->k5_wpb, p->lds_k5, s));
             }
         }
     } else {

@@ -83,8 +83,8 @@ hipError_t dd_launch_k5_roff(const uint32_t *bcounts, const uint64_t *part_boffs
 hipError_t dd_launch_k5_scatter(int64_t n, uint32_t nparts, int nbits, const uint32_t *pid,
                                 const int32_t *offsets, const uint8_t *in_bytes,
                                 const uint32_t *gbaseB, const uint32_t *roffB,
-                                int64_t nrounds, uint8_t *out_bytes, size_t lds_bytes,
-                                hipStream_t s);
+                                int64_t nrounds, uint8_t *out_bytes, int wpb5,
+                                size_t lds_bytes, hipStream_t s);
 hipError_t dd_launch_hash_count(const dd_kargs *a, int64_t nchunks, int64_t chunk_rows,
                                 uint32_t nparts, int nbits, uint32_t *pid_out,
                                 uint32_t *counts, uint32_t *bcounts, size_t lds_bytes,

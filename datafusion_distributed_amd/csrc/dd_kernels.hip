@@ -2087,18 +2087,26 @@ hipError_t dd_launch_k5_roff(const uint32_t *bcounts, const uint64_t *part_boffs
 hipError_t dd_launch_k5_scatter(int64_t n, uint32_t nparts, int nbits, const uint32_t *pid,
                                 const int32_t *offsets, const uint8_t *in_bytes,
                                 const uint32_t *gbaseB, const uint32_t *roffB,
-                                int64_t nrounds, uint8_t *out_bytes, size_t lds_bytes,
-                                hipStream_t s) {
-    if (lds_bytes > 65536) {
-        hipError_t e = hipFuncSetAttribute((const void *)k5_scatter<16>,
-                                           hipFuncAttributeMaxDynamicSharedMemorySize,
-                                           (int)lds_bytes);
-        if (e != hipSuccess) return e;
+                                int64_t nrounds, uint8_t *out_bytes, int wpb5,
+                                size_t lds_bytes, hipStream_t s) {
+#define DD_K5S(W)                                                                            \
+    if (wpb5 == W) {                                                                         \
+        if (lds_bytes > 65536) {                                                             \
+            hipError_t e = hipFuncSetAttribute((const void *)k5_scatter<W>,                  \
+                                               hipFuncAttributeMaxDynamicSharedMemorySize,   \
+                                               (int)lds_bytes);                              \
+            if (e != hipSuccess) return e;                                                   \
+        }                                                                                    \
+        hipLaunchKernelGGL((k5_scatter<W>), dim3((unsigned)nrounds), dim3(W * WAVE),         \
+                           lds_bytes, s, n, nparts, nbits, pid, offsets, in_bytes, gbaseB,   \
+                           roffB, nrounds, out_bytes);                                       \
+        return hipGetLastError();                                                            \
     }
-    hipLaunchKernelGGL((k5_scatter<16>), dim3((unsigned)nrounds), dim3(16 * WAVE),
-                       lds_bytes, s, n, nparts, nbits, pid, offsets, in_bytes, gbaseB,
-                       roffB, nrounds, out_bytes);
-    return hipGetLastError();
+    DD_K5S(16)
+    DD_K5S(8) /* half-size blocks: ~62 KB LDS -> 2 blocks/CU, so one block's flush
+                 overlaps the co-resident block's place loads */
+#undef DD_K5S
+    return hipErrorInvalidValue;
 }
 
 hipError_t dd_launch_dict_hashes(const uint8_t *bytes, const int32_t *offsets, int64_t n,
