@@ -219,3 +219,28 @@ def test_decode_rejects_non_column_hash_expr():
     blob[idx] = 0x12
     st, _, _ = decode_partitioning(bytes(blob))
     assert st == 6  # DD_ERR_UNSUPPORTED
+
+
+def test_proto_decoder_survives_hostile_bytes():
+    """Mutated and random blobs must decode to an error or bounded values — never
+    crash or read out of bounds (the wire reader in dd_proto.cpp checks every
+    varint/length against the buffer end)."""
+    rng = __import__("numpy").random.default_rng(13)
+    np = __import__("numpy")
+    qid = uuid.uuid4()
+    req = M["ddtest.ExecuteTaskRequest"]()
+    req.task_key.CopyFrom(make_task_key(qid, 1, 2))
+    req.target_partition_start = 5
+    req.target_partition_end = 9
+    req.repartition.partitioning = make_partitioning([("a", 1), ("b", 2)], 64)
+    base = bytearray(req.SerializeToString())
+    for _ in range(500):
+        m = bytearray(base)
+        for _ in range(rng.integers(1, 6)):
+            m[rng.integers(0, len(m))] = rng.integers(0, 256)
+        decode_execute(bytes(m))  # any status; must not crash
+    for _ in range(200):
+        noise = bytes(rng.integers(0, 256, rng.integers(0, 120), dtype=np.int64)
+                      .astype(np.uint8))
+        decode_execute(noise)
+        decode_partitioning(noise)
