@@ -382,7 +382,7 @@ extern "C" dd_status dd_partitioner_create(const dd_batch_desc *batch, const int
                 pgmax = 2;
             else if (Pn <= 256 && pre_is({8, 8, 8, 4}))
                 pgmax = 4;
-            else if (Pn <= 512 && pre_is({8, 8, 8, 4}))
+            else if (Pn <= 512 && (pre_is({8, 8, 8, 4}) || pre_is({8, 8, 8, 4, 4})))
                 pgmax = 2;
         }
         if (pgmax > 0) {

@@ -2260,8 +2260,9 @@ hipError_t dd_launch_scatter_pre(const dd_kargs *a, int64_t nblocks, int64_t nro
     /* 129..256 tier (NBG=4, NBI=2): headline + multikey */
     DD_PRE(4, 4, 2, 8, 8, 8, 4)
     DD_PRE(4, 4, 2, 8, 8, 8, 4, 4)
-    /* 257..512 tier (NBG=8, NBI=4, G=2 for LDS): headline shape */
+    /* 257..512 tier (NBG=8, NBI=4, G=2 for LDS): headline + multikey shapes */
     DD_PRE(2, 8, 4, 8, 8, 8, 4)
+    DD_PRE(2, 8, 4, 8, 8, 8, 4, 4)
 #undef DD_PRE
 #undef DD_PREW
     return hipErrorInvalidValue;
