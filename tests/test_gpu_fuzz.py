@@ -171,6 +171,8 @@ def test_pre_ab(monkeypatch):
           random_col(rng, n, "f64", 0), random_col(rng, n, "i32", 0),
           random_col(rng, n, "i32", 0)]
     check_against_oracle(mk, [0, 4], 128)
+    check_against_oracle(mk, [0, 4], 512)  # multikey P<=512 tier (G2, NBG8)
+    check_against_oracle(mk, [0, 4], 300)
     q1 = [random_col(rng, n, "u8", 0), random_col(rng, n, "bool", 0),
           random_col(rng, n, "f64", 0), random_col(rng, n, "f64", 0),
           random_col(rng, n, "f64", 0), random_col(rng, n, "f64", 0),
