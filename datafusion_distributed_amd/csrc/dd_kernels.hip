@@ -1981,18 +1981,18 @@ __global__ void k4_part_boffsets(const uint64_t *out_off, const uint64_t *part_o
 
 /* one spec-path launch, fully typed; shared by the RHASH and pid variants (C++ linkage:
  * templates cannot live inside the extern "C" block below) */
-template <int G, int C, int N, bool RH>
+template <int G, int C, int N, bool RH, bool HV = false>
 static hipError_t dd_launch_spec1(const dd_kargs *a, dim3 grid, int64_t tile_rows,
                                   uint32_t nparts, int nbits, const uint32_t *pid_in,
                                   const uint32_t *tile_off, const uint64_t *part_offsets,
                                   size_t lds_bytes, hipStream_t s) {
     if (lds_bytes > 65536) {
         hipError_t e =
-            hipFuncSetAttribute((const void *)k_scatter_staged<G, 16, C, false, N, RH>,
+            hipFuncSetAttribute((const void *)k_scatter_staged<G, 16, C, HV, N, RH>,
                                 hipFuncAttributeMaxDynamicSharedMemorySize, (int)lds_bytes);
         if (e != hipSuccess) return e;
     }
-    hipLaunchKernelGGL((k_scatter_staged<G, 16, C, false, N, RH>), grid, dim3(16 * WAVE),
+    hipLaunchKernelGGL((k_scatter_staged<G, 16, C, HV, N, RH>), grid, dim3(16 * WAVE),
                        lds_bytes, s, *a, tile_rows, nparts, nbits, pid_in, tile_off,
                        part_offsets);
     return hipGetLastError();
@@ -2396,6 +2396,34 @@ hipError_t dd_launch_scatter_staged(const dd_kargs *a, int64_t nblocks, int64_t 
     DD_SPEC8(4, 7)
     DD_SPEC8(4, 8)
 #undef DD_SPEC8
+    /* NC-specialized staged-var: compile-time column-count unroll for var batches with
+     * no validity (the original var columns ride as elem-0 no-op entries; the synthetic
+     * VARLEN/ROWID columns are handled by the HASVAR branch). The generic runtime-column
+     * loop measured 8-15% slower on the spec path in round 1 — same reasoning here. */
+    bool can_specv = hasvar && wpb == 16 && a->n_cols <= 8;
+    for (int c = 0; c < a->n_cols && can_specv; c++)
+        if (a->cols[c].valid) can_specv = false;
+#define DD_SPECV(G, N)                                                                       \
+    if (can_specv && gmax == G && a->n_cols == N) {                                          \
+        return dd_launch_spec1<G, 8, N, false, true>(a, grid, tile_rows, nparts, nbits,      \
+                                                     pid_in, tile_off, part_offsets,         \
+                                                     lds_bytes, s);                          \
+    }
+    DD_SPECV(2, 2)
+    DD_SPECV(2, 3)
+    DD_SPECV(2, 4)
+    DD_SPECV(2, 5)
+    DD_SPECV(2, 6)
+    DD_SPECV(2, 7)
+    DD_SPECV(2, 8)
+    DD_SPECV(4, 2)
+    DD_SPECV(4, 3)
+    DD_SPECV(4, 4)
+    DD_SPECV(4, 5)
+    DD_SPECV(4, 6)
+    DD_SPECV(4, 7)
+    DD_SPECV(4, 8)
+#undef DD_SPECV
     /* rhash batches MUST take a spec path above (no pid array exists to read) — fail
      * loudly rather than fall through to a pid-consuming variant */
     if (a->rhash) return hipErrorInvalidValue;
