@@ -146,6 +146,8 @@ struct dd_partitioner {
     uint64_t *part_boffsets = nullptr;   /* [nvar][P+1] */
     uint16_t *imgb = nullptr;            /* pre: [nrounds][sP2] round image bases (roff) */
     uint32_t *partials2 = nullptr;       /* pre: second-level scan partials [64][P] */
+    uint16_t *counts16 = nullptr;        /* pre: [nseg][P] u16 segment counts */
+    uint32_t *gbase = nullptr;           /* pre: [nseg][P] u32 global slot bases */
     uint32_t *src_row = nullptr;         /* staged-var: permutation out[slot] = input row */
     uint32_t *k4w_meta = nullptr;        /* staged-var: window hist/base (k4_copy_ord) */
     uint32_t *k4w_order = nullptr;       /* staged-var: window-bucketed group order */
@@ -154,7 +156,8 @@ struct dd_partitioner {
     int64_t k5_nrounds = 0, k5_nseg = 0;
     uint32_t k5_maxlen = 0;
     size_t lds_k5 = 0, lds_k5c = 0;
-    uint32_t *k5_bcounts = nullptr;      /* [nseg5][P] */
+    uint16_t *k5_bcounts = nullptr;      /* [nseg5][P] u16 byte counts */
+    uint32_t *k5_gbase = nullptr;        /* [nseg5][P] u32 global byte bases */
     uint32_t *k5_partials = nullptr;     /* [2048][P] */
     uint32_t *k5_partials2 = nullptr;    /* [64][P] */
     uint32_t *k5_roffB = nullptr;        /* [nrounds5][P+1] */
@@ -181,7 +184,10 @@ struct dd_partitioner {
         (void)hipFree(src_row);
         (void)hipFree(k4w_meta);
         (void)hipFree(k4w_order);
+        (void)hipFree(counts16);
+        (void)hipFree(gbase);
         (void)hipFree(k5_bcounts);
+        (void)hipFree(k5_gbase);
         (void)hipFree(k5_partials);
         (void)hipFree(k5_partials2);
         (void)hipFree(k5_roffB);
@@ -497,13 +503,15 @@ extern "C" dd_status dd_partitioner_create(const dd_batch_desc *batch, const int
     };
     bool ok = (ka.rhash ? true
                         : halloc((void **)&p->pid, (size_t)n * (ka.pid8 ? 1 : 4))) &&
-              halloc((void **)&p->counts, (size_t)nchunks * P * 4) &&
+              halloc((void **)&p->counts, p->pre ? 4 : (size_t)nchunks * P * 4) &&
               halloc((void **)&p->partials,
                      (size_t)(p->pre ? p->pre_nranges : DD_SCAN_RANGES) * P * 4) &&
               halloc((void **)&p->part_offsets, (size_t)(P + 1) * 8);
     if (ok && p->pre)
         ok = halloc((void **)&p->imgb, (size_t)p->nrounds * p->sP2 * 2) &&
-             halloc((void **)&p->partials2, (size_t)64 * P * 4);
+             halloc((void **)&p->partials2, (size_t)64 * P * 4) &&
+             halloc((void **)&p->counts16, (size_t)p->nseg_pad * P * 2) &&
+             halloc((void **)&p->gbase, (size_t)p->nseg_pad * P * 4);
     if (ok && nvar > 0) {
         ok = halloc((void **)&p->part_boffsets, (size_t)nvar * (P + 1) * 8);
         if (p->staged) {
@@ -546,6 +554,8 @@ extern "C" dd_status dd_partitioner_create(const dd_batch_desc *batch, const int
                         p->lds_k5 = lds5;
                         p->lds_k5c = (size_t)WAVES_PER_BLOCK_H * P * 4;
                         ok = halloc((void **)&p->k5_bcounts,
+                                    (size_t)p->k5_nseg * P * 2) &&
+                             halloc((void **)&p->k5_gbase,
                                     (size_t)p->k5_nseg * P * 4) &&
                              halloc((void **)&p->k5_partials, (size_t)2048 * P * 4) &&
                              halloc((void **)&p->k5_partials2, (size_t)64 * P * 4) &&
@@ -649,7 +659,7 @@ extern "C" dd_status dd_partitioner_run_phase1(dd_partitioner *p, void *stream) 
     if (p->pre) {
         HIP_TRY(hipMemsetAsync(p->partials, 0, (size_t)p->pre_nranges * p->nparts * 4, s));
         HIP_TRY(dd_launch_hash_count_seg(&p->ka, p->nseg_pad, p->chunk_rows, p->nparts,
-                                         p->nbits, p->pid, p->counts, p->partials,
+                                         p->nbits, p->pid, p->counts16, p->partials,
                                          p->pre_nranges, p->lds_k1, s));
     } else if (p->staged) {
         HIP_TRY(dd_launch_hash_count_tile(&p->ka, p->nchunks, p->chunk_rows, p->nparts,
@@ -660,14 +670,15 @@ extern "C" dd_status dd_partitioner_run_phase1(dd_partitioner *p, void *stream) 
     }
     HIP_TRY(hipEventRecord(p->ev[1], s));
     if (p->pre) {
-        HIP_TRY(dd_launch_scan_deep(p->counts, p->nchunks, p->nparts, p->pre_nranges, 64,
-                                    p->partials, p->partials2, p->part_offsets, 1, s));
+        HIP_TRY(dd_launch_scan_deep(p->counts16, p->nchunks, p->nparts, p->pre_nranges,
+                                    64, p->partials, p->partials2, p->part_offsets,
+                                    p->gbase, s));
     } else {
         HIP_TRY(dd_launch_scan(p->counts, p->nchunks, p->nparts, DD_SCAN_RANGES,
                                p->partials, p->part_offsets, 0, s));
     }
     if (p->pre) {
-        HIP_TRY(dd_launch_round_layout(p->counts, p->part_offsets, p->nrounds, p->wpb,
+        HIP_TRY(dd_launch_round_layout(p->gbase, p->part_offsets, p->nrounds, p->wpb,
                                        p->nparts, p->sP2, p->imgb, s));
     }
     if (p->k5) { /* byte-base precompute for the K5 var scatter */
@@ -678,8 +689,8 @@ extern "C" dd_status dd_partitioner_run_phase1(dd_partitioner *p, void *stream) 
                                    p->k5_partials, 2048, p->k5_nseg, p->lds_k5c, s));
         HIP_TRY(dd_launch_scan_deep(p->k5_bcounts, p->k5_nseg, p->nparts, 2048, 64,
                                     p->k5_partials, p->k5_partials2, p->part_boffsets,
-                                    1, s));
-        HIP_TRY(dd_launch_k5_roff(p->k5_bcounts, p->part_boffsets, p->k5_nrounds,
+                                    p->k5_gbase, s));
+        HIP_TRY(dd_launch_k5_roff(p->k5_gbase, p->part_boffsets, p->k5_nrounds,
                                   p->k5_wpb, p->nparts, p->k5_roffB, s));
     }
     if (!p->staged) {
@@ -701,7 +712,7 @@ extern "C" dd_status dd_partitioner_run_phase2(dd_partitioner *p, void *stream) 
     if (p->pre) {
         const int64_t nblocks = (p->nrounds + p->rpb - 1) / p->rpb;
         HIP_TRY(dd_launch_scatter_pre(&p->ka, nblocks, p->nrounds, p->rpb, p->nparts,
-                                      p->nbits, p->pid, p->counts, p->imgb, p->sP2,
+                                      p->nbits, p->pid, p->gbase, p->imgb, p->sP2,
                                       p->gmax, p->wpb, p->lds_k3, s));
     } else if (p->staged) {
         HIP_TRY(dd_launch_scatter_staged(&p->ka, p->nchunks, p->chunk_rows, p->nparts,
@@ -719,7 +730,7 @@ extern "C" dd_status dd_partitioner_run_phase2(dd_partitioner *p, void *stream) 
             if (p->k5) {
                 HIP_TRY(dd_launch_k5_scatter(
                     p->ka.n_rows, p->nparts, p->nbits, p->pid, cd.offsets,
-                    (const uint8_t *)cd.data, p->k5_bcounts, p->k5_roffB, p->k5_nrounds,
+                    (const uint8_t *)cd.data, p->k5_gbase, p->k5_roffB, p->k5_nrounds,
                     (uint8_t *)p->out_data[ci], p->k5_wpb, p->lds_k5, s));
             }
         }

@@ -74,7 +74,7 @@ hipError_t dd_launch_dict_hashes(const uint8_t *bytes, const int32_t *offsets, i
 hipError_t dd_launch_k5_maxlen(const int32_t *offsets, int64_t n, uint32_t *out_max,
                                hipStream_t s);
 hipError_t dd_launch_k5_count(int64_t n, uint32_t nparts, const uint32_t *pid,
-                              const int32_t *offsets, uint32_t *bcounts,
+                              const int32_t *offsets, uint16_t *bcounts,
                               uint32_t *partials, int nranges, int64_t nseg5,
                               size_t lds_bytes, hipStream_t s);
 hipError_t dd_launch_k5_roff(const uint32_t *bcounts, const uint64_t *part_boffsets,
@@ -92,12 +92,12 @@ hipError_t dd_launch_hash_count(const dd_kargs *a, int64_t nchunks, int64_t chun
 hipError_t dd_launch_scan(uint32_t *counts, int64_t nchunks, uint32_t nparts, int nranges,
                           uint32_t *partials, uint64_t *part_offsets, int fold_global,
                           hipStream_t s);
-hipError_t dd_launch_scan_deep(uint32_t *counts, int64_t nchunks, uint32_t nparts,
+hipError_t dd_launch_scan_deep(uint16_t *counts16, int64_t nchunks, uint32_t nparts,
                                int nr1, int nr2, uint32_t *partials, uint32_t *partials2,
-                               uint64_t *part_offsets, int fold_global, hipStream_t s);
+                               uint64_t *part_offsets, uint32_t *gbase, hipStream_t s);
 hipError_t dd_launch_hash_count_seg(const dd_kargs *a, int64_t nseg, int64_t seg_rows,
                                     uint32_t nparts, int nbits, uint32_t *pid_out,
-                                    uint32_t *counts, uint32_t *partials, int nranges,
+                                    uint16_t *counts, uint32_t *partials, int nranges,
                                     size_t lds_bytes, hipStream_t s);
 hipError_t dd_launch_round_layout(const uint32_t *counts, const uint64_t *part_offsets,
                                   int64_t nrounds, int wpb, uint32_t nparts, uint32_t sP2,

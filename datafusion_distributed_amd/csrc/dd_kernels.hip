@@ -1481,7 +1481,8 @@ __global__ __launch_bounds__(W * WAVE) void k_scatter_pre(
  * full re-read of counts. a.pid8=1 stores pids as u8. */
 __global__ __launch_bounds__(BLOCK_THREADS) void k_hash_count_seg(
     dd_kargs a, int64_t nseg, int64_t seg_rows, uint32_t nparts, int nbits,
-    uint32_t *pid_out, uint32_t *counts /* [nseg][P] */,
+    uint32_t *pid_out, uint16_t *counts /* [nseg][P]; seg counts <= SEG <= 1024: u16
+                                           halves the count-stream traffic */,
     uint32_t *partials /* [nranges][P], pre-zeroed */, int nranges) {
     extern __shared__ __attribute__((aligned(16))) char smem[];
     const int wid = threadIdx.x / WAVE;
@@ -1537,7 +1538,7 @@ __global__ __launch_bounds__(BLOCK_THREADS) void k_hash_count_seg(
     uint32_t *prow = partials + (size_t)rw * nparts;
     for (uint32_t p = lane; p < nparts; p += WAVE) {
         const uint32_t h = hist[p];
-        counts[(size_t)seg * nparts + p] = h;
+        counts[(size_t)seg * nparts + p] = (uint16_t)h;
         if (h) atomicAdd(&prow[p], h);
     }
     } /* segment loop */
@@ -1612,8 +1613,8 @@ __global__ void k5_maxlen(const int32_t *offsets, int64_t n, uint32_t *out_max) 
 /* per-64-row-group byte histograms: seg = one wave's group; counts[seg][p] (u32) */
 __global__ __launch_bounds__(BLOCK_THREADS) void k5_count(
     int64_t n, uint32_t nparts, const uint32_t *pid, const int32_t *offsets,
-    uint32_t *bcounts /* [nseg5][P] */, uint32_t *partials /* [nranges] fused L1 */,
-    int nranges, int64_t nseg5) {
+    uint16_t *bcounts /* [nseg5][P]; 64-row-seg bytes <= 64*maxlen(128) fits u16 */,
+    uint32_t *partials /* [nranges] fused L1 */, int nranges, int64_t nseg5) {
     extern __shared__ __attribute__((aligned(16))) char smem[];
     const int wid = threadIdx.x / WAVE;
     const int lane = threadIdx.x % WAVE;
@@ -1631,7 +1632,7 @@ __global__ __launch_bounds__(BLOCK_THREADS) void k5_count(
     uint32_t *prow = partials + (size_t)rw * nparts;
     for (uint32_t p = lane; p < nparts; p += WAVE) {
         const uint32_t h = hist[p];
-        bcounts[(size_t)seg * nparts + p] = h;
+        bcounts[(size_t)seg * nparts + p] = (uint16_t)h;
         if (h) atomicAdd(&prow[p], h);
     }
 }
@@ -2059,7 +2060,7 @@ hipError_t dd_launch_k5_maxlen(const int32_t *offsets, int64_t n, uint32_t *out_
 }
 
 hipError_t dd_launch_k5_count(int64_t n, uint32_t nparts, const uint32_t *pid,
-                              const int32_t *offsets, uint32_t *bcounts,
+                              const int32_t *offsets, uint16_t *bcounts,
                               uint32_t *partials, int nranges, int64_t nseg5,
                               size_t lds_bytes, hipStream_t s) {
     int64_t blocks = nseg5 / WAVES_PER_BLOCK;
@@ -2150,31 +2151,46 @@ hipError_t dd_launch_scan(uint32_t *counts, int64_t nchunks, uint32_t nparts, in
     return hipGetLastError();
 }
 
+__global__ void k_scan_rewrite16(const uint16_t *counts16, int64_t nchunks,
+                                 uint32_t nparts, int nranges, const uint32_t *partials,
+                                 const uint64_t *fold_offsets, uint32_t *gbase) {
+    const int64_t tid = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    const int64_t total = (int64_t)nranges * nparts;
+    if (tid >= total) return;
+    const int r = (int)(tid / nparts);
+    const uint32_t p = (uint32_t)(tid % nparts);
+    const int64_t c0 = nchunks * r / nranges, c1 = nchunks * (r + 1) / nranges;
+    uint32_t run = partials[(size_t)r * nparts + p] + (uint32_t)fold_offsets[p];
+    for (int64_t c = c0; c < c1; c++) {
+        gbase[c * nparts + p] = run;
+        run += counts16[c * nparts + p];
+    }
+}
+
 /* two-level scan for segment-granular counts (pre path): k_scan_combine walking 2048
  * ranges serially in one block measured 0.6 ms (latency-bound); recurse instead:
  * counts -> partials[nr1] -> partials2[nr2] -> combine(nr2) -> rewrite back up. */
-hipError_t dd_launch_scan_deep(uint32_t *counts, int64_t nchunks, uint32_t nparts,
+hipError_t dd_launch_scan_deep(uint16_t *counts16, int64_t nchunks, uint32_t nparts,
                                int nr1, int nr2, uint32_t *partials, uint32_t *partials2,
-                               uint64_t *part_offsets, int fold_global, hipStream_t s) {
+                               uint64_t *part_offsets, uint32_t *gbase, hipStream_t s) {
     int threads = 256;
     int b1 = (int)(((int64_t)nr1 * nparts + threads - 1) / threads);
     int b2 = (int)(((int64_t)nr2 * nparts + threads - 1) / threads);
-    /* level-1 partials were fused into k_hash_count_seg (atomicAdd per wave) */
+    /* level-1 partials were fused into k_hash_count_seg / k5_count (atomicAdd) */
     hipLaunchKernelGGL(k_scan_partial, dim3(b2), dim3(threads), 0, s, partials, (int64_t)nr1,
                        nparts, nr2, partials2);
     hipLaunchKernelGGL(k_scan_combine, dim3(1), dim3(256), 0, s, partials2, nr2, nparts,
                        part_offsets);
     hipLaunchKernelGGL(k_scan_rewrite, dim3(b2), dim3(threads), 0, s, partials, (int64_t)nr1,
                        nparts, nr2, partials2, nullptr);
-    hipLaunchKernelGGL(k_scan_rewrite, dim3(b1), dim3(threads), 0, s, counts, nchunks,
-                       nparts, nr1, partials,
-                       fold_global ? (const uint64_t *)part_offsets : nullptr);
+    hipLaunchKernelGGL(k_scan_rewrite16, dim3(b1), dim3(threads), 0, s, counts16, nchunks,
+                       nparts, nr1, partials, (const uint64_t *)part_offsets, gbase);
     return hipGetLastError();
 }
 
 hipError_t dd_launch_hash_count_seg(const dd_kargs *a, int64_t nseg, int64_t seg_rows,
                                     uint32_t nparts, int nbits, uint32_t *pid_out,
-                                    uint32_t *counts, uint32_t *partials, int nranges,
+                                    uint16_t *counts, uint32_t *partials, int nranges,
                                     size_t lds_bytes, hipStream_t s) {
     /* full grid (one segment per wave): a capped grid with consecutive-segment runs
      * per wave measured K1 0.26 -> 0.40 ms — the dense all-waves-in-one-row-window
