@@ -281,3 +281,21 @@ def test_reader_survives_hostile_bytes():
         st = L.dd_ipc_reader_create(noise, ctypes.c_int64(len(noise)), ctypes.byref(r))
         if st == 0 and r.value:
             L.dd_ipc_reader_destroy(r)
+
+
+def test_zstd_compression_rejected_cleanly():
+    """The reference supports lz4|zstd|none (distributed_config.rs:36-38); this tier
+    implements lz4 (the default) — a zstd stream must fail loudly, not misdecode."""
+    if not pa.Codec.is_available("zstd"):
+        pytest.skip("no zstd in this pyarrow")
+    rng = np.random.default_rng(12)
+    _, pab = make_pa_batch(rng, 100)
+    sink = io.BytesIO()
+    with pa.ipc.new_stream(sink, pab.schema,
+                           options=pa.ipc.IpcWriteOptions(compression="zstd")) as w:
+        w.write_batch(pab)
+    L = api.lib()
+    r = ctypes.c_void_p(0)
+    blob = sink.getvalue()
+    st = L.dd_ipc_reader_create(blob, ctypes.c_int64(len(blob)), ctypes.byref(r))
+    assert st == 6  # DD_ERR_UNSUPPORTED
