@@ -392,24 +392,47 @@ extern "C" dd_status dd_execute_task_proto(const uint8_t *execute_task_pb, int64
     }
     if (lo > hi || hi > nparts)
         return dd_set_err_invalid("partition range outside the decoded partitioning");
+    /* lazy head insert (task_data.rs:104-116). The partitioner is CREATED outside the
+     * cache mutex (device allocations take milliseconds; holding the global lock would
+     * serialize unrelated tasks) and installed under the lock; a losing racer destroys
+     * its copy and reuses the winner's. */
+    std::shared_ptr<dd_partitioner> fresh;
     {
         std::lock_guard<std::mutex> g(g_ptasks_mu);
-        if (!t->part) { /* lazy head insert (task_data.rs:104-116) */
-            dd_partitioner *p = nullptr;
-            st = dd_partitioner_create(&t->batch, key_cols, n_keys, nparts, &p);
-            if (st != DD_OK) return st;
-            t->part = std::shared_ptr<dd_partitioner>(p, dd_partitioner_destroy);
+        if (t->part) {
+            if (t->nparts != nparts || t->n_keys != n_keys ||
+                memcmp(t->key_cols, key_cols, sizeof(int32_t) * n_keys) != 0)
+                return dd_set_err_invalid(
+                    "producer_head changed between executes of one task (the reference "
+                    "builds it once per TaskData)");
+            *out = t->part.get();
+            return DD_OK;
+        }
+    }
+    {
+        dd_partitioner *p = nullptr;
+        st = dd_partitioner_create(&t->batch, key_cols, n_keys, nparts, &p);
+        if (st != DD_OK) return st;
+        fresh = std::shared_ptr<dd_partitioner>(p, dd_partitioner_destroy);
+    }
+    {
+        std::lock_guard<std::mutex> g(g_ptasks_mu);
+        if (!t->part) { /* we win: install + run (launches are async enqueues) */
+            t->part = fresh;
             memcpy(t->key_cols, key_cols, sizeof(key_cols));
             t->n_keys = n_keys;
             t->nparts = nparts;
-            st = dd_partitioner_run(p, stream);
-            if (st != DD_OK) return st;
+            st = dd_partitioner_run(fresh.get(), stream);
+            if (st != DD_OK) {
+                t->part.reset(); /* leave the task re-executable */
+                return st;
+            }
         } else if (t->nparts != nparts || t->n_keys != n_keys ||
                    memcmp(t->key_cols, key_cols, sizeof(int32_t) * n_keys) != 0) {
             return dd_set_err_invalid(
                 "producer_head changed between executes of one task (the reference "
                 "builds it once per TaskData)");
-        }
+        } /* else: lost the race; `fresh` is destroyed on scope exit */
     }
     *out = t->part.get();
     return DD_OK;
