@@ -299,3 +299,19 @@ def test_zstd_compression_rejected_cleanly():
     blob = sink.getvalue()
     st = L.dd_ipc_reader_create(blob, ctypes.c_int64(len(blob)), ctypes.byref(r))
     assert st == 6  # DD_ERR_UNSUPPORTED
+
+
+def test_wire_py_sliced_and_null_bool():
+    """Sliced RecordBatches (array offset != 0) are rebased before encoding, and
+    null-bearing bool columns survive — both read back by pyarrow value-equal."""
+    from datafusion_distributed_amd import wire
+
+    b = pa.record_batch([pa.array(np.arange(100, dtype=np.int64)),
+                         pa.array([f"s{i}" for i in range(100)])], names=["x", "s"])
+    sl = b.slice(10, 50)
+    got = pa.ipc.open_stream(io.BytesIO(wire.encode_batches([sl], use_lz4=True))).read_all()
+    assert got.equals(pa.Table.from_batches([sl]))
+
+    bb = pa.record_batch([pa.array([True, None, False] * 30)], names=["b"])
+    got2 = pa.ipc.open_stream(io.BytesIO(wire.encode_batches([bb]))).read_all()
+    assert got2.equals(pa.Table.from_batches([bb]))
